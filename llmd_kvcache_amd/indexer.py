@@ -1,0 +1,123 @@
+"""KV-cache Indexer orchestrator: the read path.
+
+Parity with reference pkg/kvcache/indexer.go:
+ - Config aggregates prefix-store / token-processor / index / scorer /
+   tokenization-pool configs (:36-43) with a defaults constructor (:47-60);
+ - get_pod_scores (:132-166): tokenize (blocking) -> block keys -> index
+   lookup (filtered by the candidate pod set; empty set = all pods) ->
+   longest-prefix score; empty key list returns no scores;
+ - run() starts the tokenization pool workers (:116-118);
+ - kv_block_index() exposes the index for the events write path (:121-123).
+
+MI355X note: with a GpuIndex backend the lookup+score steps run fused in a
+single HIP kernel (ops/csrc/kvidx_hip.hip); with the sharded parallel index
+(parallel/sharded.py) per-shard partial hit masks are merged over RCCL.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Sequence
+
+from .kvblock.index import Index, IndexConfig, new_index
+from .kvblock.keys import Key
+from .kvblock.token_processor import ChunkedTokenDatabase, TokenProcessorConfig
+from .scorer import (
+    KVBlockScorerConfig,
+    KVCacheBackendConfig,
+    default_kv_cache_backend_configs,
+    new_kv_block_scorer,
+)
+from .tokenization.pool import TokenizationConfig, TokenizationPool
+from .tokenization.prefixstore import LRUStoreConfig, LRUTokenStore
+
+
+@dataclass
+class Config:
+    prefix_store: LRUStoreConfig = field(default_factory=LRUStoreConfig)
+    token_processor: TokenProcessorConfig = field(
+        default_factory=TokenProcessorConfig
+    )
+    kv_block_index: IndexConfig = field(default_factory=IndexConfig.default)
+    scorer: KVBlockScorerConfig = field(default_factory=KVBlockScorerConfig)
+    tokenizers_pool: TokenizationConfig = field(
+        default_factory=TokenizationConfig
+    )
+    backend_configs: List[KVCacheBackendConfig] = field(
+        default_factory=default_kv_cache_backend_configs
+    )
+
+
+class Indexer:
+    def __init__(
+        self,
+        config: Optional[Config] = None,
+        tokenization_pool: Optional[TokenizationPool] = None,
+        kv_block_index: Optional[Index] = None,
+    ):
+        self.config = config or Config()
+
+        self.tokens_indexer = LRUTokenStore(self.config.prefix_store)
+        self.tokens_processor = ChunkedTokenDatabase(self.config.token_processor)
+        self._kv_block_index = kv_block_index or new_index(
+            self.config.kv_block_index
+        )
+        # scorer backend configs are overridden by top-level backend configs
+        # (indexer.go:96-97)
+        self.config.scorer.backend_configs = self.config.backend_configs
+        self.kv_block_scorer = new_kv_block_scorer(self.config.scorer)
+        self.tokenizers_pool = tokenization_pool or TokenizationPool(
+            self.config.tokenizers_pool, self.tokens_indexer
+        )
+
+    def run(self) -> None:
+        self.tokenizers_pool.run()
+
+    def shutdown(self) -> None:
+        self.tokenizers_pool.shutdown()
+
+    def kv_block_index(self) -> Index:
+        return self._kv_block_index
+
+    def get_pod_scores(
+        self,
+        render_req,
+        prompt: str,
+        model_name: str,
+        pod_identifiers: Sequence[str],
+    ) -> Dict[str, float]:
+        # 1. tokenize prompt (blocking on the pool)
+        tokens = self.tokenizers_pool.tokenize(render_req, prompt, model_name)
+
+        # 2. block keys
+        block_keys = self.tokens_processor.tokens_to_kv_block_keys(
+            None, tokens, model_name
+        )
+        if not block_keys:
+            return {}
+
+        # 3. index lookup (empty filter set = all pods)
+        key_to_pods = self._kv_block_index.lookup(
+            block_keys, set(pod_identifiers)
+        )
+
+        # 4. score
+        return self.kv_block_scorer.score(block_keys, key_to_pods)
+
+    def score_tokens(
+        self,
+        tokens: Sequence[int],
+        model_name: str,
+        pod_identifiers: Sequence[str],
+    ) -> Dict[str, float]:
+        """Pre-tokenized scoring entry point (MI355X-native addition for
+        callers that already hold token ids; skips the tokenization pool)."""
+        block_keys = self.tokens_processor.tokens_to_kv_block_keys(
+            None, tokens, model_name
+        )
+        if not block_keys:
+            return {}
+        key_to_pods = self._kv_block_index.lookup(
+            block_keys, set(pod_identifiers)
+        )
+        return self.kv_block_scorer.score(block_keys, key_to_pods)

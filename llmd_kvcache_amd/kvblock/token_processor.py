@@ -1,0 +1,89 @@
+"""Tokens -> chained KV-block keys (vLLM-compatible content addressing).
+
+Behavioral parity with reference pkg/kvcache/kvblock/token_processor.go:
+ - tokens are chunked into BlockSize-token chunks (default 16, the vLLM
+   default; token_processor.go:31), partial tail chunks dropped (:126-138);
+ - hash chain h_i = FNV-64a(canonical-CBOR([h_{i-1}, chunk, null]))
+   (:94-112); root = FNV-64a(hash_seed) (:81-90), which must be aligned with
+   the vLLM fleet's PYTHONHASHSEED-derived seed;
+ - a chain may be continued from an explicit parent key (:141-162), which the
+   event write path uses to stitch chains (kvevents/pool.go:279-296).
+
+Fast paths: the C++ extension (ops._C.tokens_to_chunk_hashes) computes the
+same chain natively; the HIP kernel (ops on gfx950) batches many prompts per
+launch, one lane per prompt chain.  This Python implementation is the golden
+reference for both and the fallback when the extension is unavailable on
+CPU-only hosts.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import List, Optional, Sequence
+
+from ..utils import hashing
+from .keys import Key
+
+DEFAULT_BLOCK_SIZE = 16
+
+
+@dataclass
+class TokenProcessorConfig:
+    block_size: int = DEFAULT_BLOCK_SIZE
+    # Aligned with vLLM's PYTHONHASHSEED; see token_processor.go:36-40.
+    hash_seed: str = ""
+    _init_hash: Optional[int] = field(default=None, repr=False)
+
+    def init_hash(self) -> int:
+        if self._init_hash is None:
+            self._init_hash = hashing.init_hash(self.hash_seed)
+        return self._init_hash
+
+
+class ChunkedTokenDatabase:
+    """Converts token sequences to chained block keys."""
+
+    def __init__(self, config: Optional[TokenProcessorConfig] = None):
+        self.config = config or TokenProcessorConfig()
+        self._native = None
+        try:  # optional C++ fast path
+            from ..ops import cpu_ext
+
+            self._native = cpu_ext.maybe_load()
+        except Exception:  # pragma: no cover - ops package always importable
+            self._native = None
+
+    @property
+    def block_size(self) -> int:
+        return self.config.block_size
+
+    def chunk_hashes(self, parent_hash: int, tokens: Sequence[int]) -> List[int]:
+        """Full-chunk chain hashes starting from parent_hash."""
+        bs = self.config.block_size
+        n_chunks = len(tokens) // bs
+        if n_chunks == 0:
+            return []
+        if self._native is not None:
+            return self._native.tokens_to_chunk_hashes(
+                list(tokens[: n_chunks * bs]), parent_hash, bs
+            )
+        hashes = []
+        h = parent_hash
+        for i in range(n_chunks):
+            h = hashing.chunk_hash(h, tokens[i * bs : (i + 1) * bs])
+            hashes.append(h)
+        return hashes
+
+    def tokens_to_kv_block_keys(
+        self,
+        parent_key: Optional[Key],
+        tokens: Sequence[int],
+        model_name: str,
+    ) -> List[Key]:
+        """Parity with TokensToKVBlockKeys (token_processor.go:141-162)."""
+        parent_hash = (
+            parent_key.chunk_hash if parent_key is not None else self.config.init_hash()
+        )
+        return [
+            Key(model_name, h) for h in self.chunk_hashes(parent_hash, tokens)
+        ]

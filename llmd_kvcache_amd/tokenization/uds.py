@@ -1,0 +1,122 @@
+"""UDS tokenizer client: HTTP over a Unix domain socket.
+
+Parity with reference pkg/tokenization/uds_tokenizer.go:
+ - POST /tokenize with {"prompt", "model"} -> {"input_ids",
+   "offset_mapping"} (:108-130);
+ - POST /chat-template (:133-157);
+ - 5 s timeout, 2 retries with exponential backoff + jitter (:59-69,
+   164-223).
+
+The server side is services/uds_tokenizer (same-shape aiohttp sidecar).
+"""
+
+from __future__ import annotations
+
+import http.client
+import json
+import random
+import socket
+import time
+from dataclasses import dataclass
+from typing import List, Tuple
+
+from .tokenizer import Offset, TokenizationError, Tokenizer
+
+DEFAULT_SOCKET_PATH = "/tmp/tokenizer/tokenizer-uds.socket"
+
+
+@dataclass
+class UdsTokenizerConfig:
+    socket_path: str = DEFAULT_SOCKET_PATH
+    timeout_s: float = 5.0
+    max_retries: int = 2
+    backoff_base_s: float = 0.1
+
+    def is_enabled(self) -> bool:
+        return bool(self.socket_path)
+
+
+class _UnixHTTPConnection(http.client.HTTPConnection):
+    def __init__(self, socket_path: str, timeout: float):
+        super().__init__("localhost", timeout=timeout)
+        self._socket_path = socket_path
+
+    def connect(self) -> None:
+        sock = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        sock.settimeout(self.timeout)
+        sock.connect(self._socket_path)
+        self.sock = sock
+
+
+class UdsTokenizer(Tokenizer):
+    def __init__(self, config: UdsTokenizerConfig):
+        self.config = config
+
+    @property
+    def type(self) -> str:
+        return "uds"
+
+    def _post(self, path: str, payload: dict) -> dict:
+        body = json.dumps(payload).encode("utf-8")
+        last_err: Exception = TokenizationError("uds request failed")
+        for attempt in range(self.config.max_retries + 1):
+            try:
+                conn = _UnixHTTPConnection(
+                    self.config.socket_path, self.config.timeout_s
+                )
+                try:
+                    conn.request(
+                        "POST",
+                        path,
+                        body=body,
+                        headers={"Content-Type": "application/json"},
+                    )
+                    resp = conn.getresponse()
+                    data = resp.read()
+                    if resp.status != 200:
+                        raise TokenizationError(
+                            f"uds {path} returned {resp.status}: {data[:200]!r}"
+                        )
+                    return json.loads(data)
+                finally:
+                    conn.close()
+            except Exception as e:
+                last_err = e
+                if attempt < self.config.max_retries:
+                    backoff = self.config.backoff_base_s * (2**attempt)
+                    time.sleep(backoff * (1 + random.random() * 0.5))
+        raise TokenizationError(f"uds request failed after retries: {last_err}")
+
+    def encode(
+        self, prompt: str, model_name: str, add_special_tokens: bool = True
+    ) -> Tuple[List[int], List[Offset]]:
+        result = self._post(
+            "/tokenize",
+            {
+                "prompt": prompt,
+                "model": model_name,
+                "add_special_tokens": add_special_tokens,
+            },
+        )
+        tokens = [int(t) for t in result.get("input_ids", [])]
+        offsets = [tuple(o) for o in result.get("offset_mapping", [])]
+        return tokens, offsets
+
+    def render_chat_template(self, req) -> str:
+        payload = {
+            "messages": req.conversations[0] if req.conversations else [],
+            "chat_template": req.chat_template,
+            "tools": req.tools,
+            "documents": req.documents,
+            "add_generation_prompt": req.add_generation_prompt,
+            "continue_final_message": req.continue_final_message,
+            "chat_template_kwargs": req.chat_template_kwargs,
+        }
+        result = self._post("/chat-template", payload)
+        rendered = result.get("rendered")
+        if rendered is None:
+            chats = result.get("rendered_chats") or []
+            rendered = chats[0] if chats else None
+        if rendered is None:
+            raise TokenizationError("uds chat-template response missing 'rendered'")
+        return rendered

@@ -1,0 +1,98 @@
+"""Token processor (chunked chained hashing) behavior tests.
+
+Mirrors reference semantics from pkg/kvcache/kvblock/token_processor.go:
+chunking with dropped partial tails, chain continuation from a parent key,
+seed handling.
+"""
+
+from llmd_kvcache_amd.kvblock.keys import Key
+from llmd_kvcache_amd.kvblock.token_processor import (
+    ChunkedTokenDatabase,
+    TokenProcessorConfig,
+)
+from llmd_kvcache_amd.utils import hashing
+
+
+def db(block_size=4, seed=""):
+    return ChunkedTokenDatabase(
+        TokenProcessorConfig(block_size=block_size, hash_seed=seed)
+    )
+
+
+class TestChunking:
+    def test_partial_tail_dropped(self):
+        d = db(block_size=4)
+        keys = d.tokens_to_kv_block_keys(None, [1, 2, 3, 4, 5, 6], "m")
+        assert len(keys) == 1
+
+    def test_too_few_tokens_returns_empty(self):
+        d = db(block_size=4)
+        assert d.tokens_to_kv_block_keys(None, [1, 2, 3], "m") == []
+
+    def test_exact_multiple(self):
+        d = db(block_size=4)
+        keys = d.tokens_to_kv_block_keys(None, list(range(8)), "m")
+        assert len(keys) == 2
+
+
+class TestChain:
+    def test_chain_links(self):
+        d = db(block_size=2)
+        keys = d.tokens_to_kv_block_keys(None, [1, 2, 3, 4], "m")
+        root = hashing.init_hash("")
+        h1 = hashing.chunk_hash(root, [1, 2])
+        h2 = hashing.chunk_hash(h1, [3, 4])
+        assert [k.chunk_hash for k in keys] == [h1, h2]
+        assert all(k.model_name == "m" for k in keys)
+
+    def test_parent_key_continuation(self):
+        d = db(block_size=2)
+        full = d.tokens_to_kv_block_keys(None, [1, 2, 3, 4], "m")
+        head = d.tokens_to_kv_block_keys(None, [1, 2], "m")
+        tail = d.tokens_to_kv_block_keys(head[-1], [3, 4], "m")
+        assert head + tail == full
+
+    def test_seed_changes_chain(self):
+        d0 = db(block_size=2, seed="")
+        d1 = db(block_size=2, seed="42")
+        k0 = d0.tokens_to_kv_block_keys(None, [1, 2], "m")
+        k1 = d1.tokens_to_kv_block_keys(None, [1, 2], "m")
+        assert k0 != k1
+        assert k1[0].chunk_hash == hashing.chunk_hash(
+            hashing.fnv1a_64(b"42"), [1, 2]
+        )
+
+    def test_vllm_default_block_size(self):
+        d = ChunkedTokenDatabase()
+        assert d.block_size == 16
+
+    def test_deterministic(self):
+        d = db(block_size=16)
+        tokens = list(range(64))
+        a = d.tokens_to_kv_block_keys(None, tokens, "m")
+        b = d.tokens_to_kv_block_keys(None, tokens, "m")
+        assert a == b
+
+
+class TestNativeParity:
+    """If the C++ extension is built, its chain must match Python exactly."""
+
+    def test_native_matches_python(self):
+        from llmd_kvcache_amd.ops import cpu_ext
+
+        mod = cpu_ext.maybe_load()
+        if mod is None:
+            import pytest
+
+            pytest.skip("native extension not built")
+        import random
+
+        rng = random.Random(7)
+        tokens = [rng.randrange(0, 2**32) for _ in range(160)]
+        py = []
+        h = hashing.init_hash("")
+        for i in range(10):
+            h = hashing.chunk_hash(h, tokens[i * 16 : (i + 1) * 16])
+            py.append(h)
+        native = mod.tokens_to_chunk_hashes(tokens, hashing.init_hash(""), 16)
+        assert list(native) == py
