@@ -1,0 +1,200 @@
+"""ZMTP transport tests: frame codec bytes (spec golden values) and
+PUB->SUB end-to-end over real TCP sockets, including the full subscriber ->
+pool -> index path (the reference's offline example topology)."""
+
+import socket
+import struct
+import threading
+import time
+
+import pytest
+
+from llmd_kvcache_amd.kvevents import zmtp
+from llmd_kvcache_amd.kvevents.zmtp import PubSocket, SubSocket
+
+
+class TestFraming:
+    def test_greeting_layout(self):
+        g = zmtp._greeting()
+        assert len(g) == 64
+        assert g[0] == 0xFF and g[9] == 0x7F
+        assert g[10:12] == bytes([3, 0])
+        assert g[12:32].rstrip(b"\x00") == b"NULL"
+
+    def test_short_frame_bytes(self):
+        a, b = socket.socketpair()
+        try:
+            zmtp._send_frame(a, b"hello", more=True)
+            data = b.recv(100)
+            assert data == bytes([zmtp.FLAG_MORE, 5]) + b"hello"
+        finally:
+            a.close()
+            b.close()
+
+    def test_long_frame_bytes(self):
+        a, b = socket.socketpair()
+        try:
+            body = b"x" * 300
+            zmtp._send_frame(a, body)
+            data = b""
+            while len(data) < 309:
+                data += b.recv(4096)
+            assert data[0] == zmtp.FLAG_LONG
+            assert struct.unpack(">Q", data[1:9])[0] == 300
+            assert data[9:] == body
+        finally:
+            a.close()
+            b.close()
+
+    def test_ready_command_parse(self):
+        body = zmtp._ready_command("PUB")
+        name, props = zmtp._parse_command(body)
+        assert name == "READY"
+        assert props["Socket-Type"] == b"PUB"
+
+    def test_parse_endpoint(self):
+        assert zmtp.parse_endpoint("tcp://*:5557") == ("0.0.0.0", 5557)
+        assert zmtp.parse_endpoint("tcp://127.0.0.1:1234") == ("127.0.0.1", 1234)
+        with pytest.raises(ValueError):
+            zmtp.parse_endpoint("ipc:///tmp/sock")
+
+
+class TestPubSub:
+    def test_pub_connect_sub_bind_roundtrip(self):
+        received = []
+        done = threading.Event()
+
+        def on_message(parts):
+            received.append(parts)
+            done.set()
+
+        sub = SubSocket(on_message)
+        sub.subscribe(b"kv@")
+        sub.bind("tcp://127.0.0.1:0")
+        pub = PubSocket()
+        try:
+            pub.connect(f"tcp://127.0.0.1:{sub.port}")
+            assert pub.wait_for_subscriber(5.0)
+            pub.send_multipart(
+                [b"kv@pod-a@model", struct.pack(">Q", 7), b"payload"]
+            )
+            assert done.wait(5.0)
+            assert received[0] == [
+                b"kv@pod-a@model",
+                struct.pack(">Q", 7),
+                b"payload",
+            ]
+        finally:
+            pub.close()
+            sub.close()
+
+    def test_topic_prefix_filtering(self):
+        received = []
+
+        def on_message(parts):
+            received.append(parts)
+
+        sub = SubSocket(on_message)
+        sub.subscribe(b"kv@")
+        sub.bind("tcp://127.0.0.1:0")
+        pub = PubSocket()
+        try:
+            pub.connect(f"tcp://127.0.0.1:{sub.port}")
+            assert pub.wait_for_subscriber(5.0)
+            pub.send_multipart([b"other@topic", b"x"])
+            pub.send_multipart([b"kv@pod@m", b"y"])
+            deadline = time.monotonic() + 5.0
+            while not received and time.monotonic() < deadline:
+                time.sleep(0.01)
+            time.sleep(0.1)  # let any stray frame arrive
+            assert len(received) == 1
+            assert received[0][0] == b"kv@pod@m"
+        finally:
+            pub.close()
+            sub.close()
+
+    def test_multiple_publishers(self):
+        received = []
+        lock = threading.Lock()
+
+        def on_message(parts):
+            with lock:
+                received.append(parts[0])
+
+        sub = SubSocket(on_message)
+        sub.subscribe(b"kv@")
+        sub.bind("tcp://127.0.0.1:0")
+        pubs = [PubSocket() for _ in range(3)]
+        try:
+            for i, pub in enumerate(pubs):
+                pub.connect(f"tcp://127.0.0.1:{sub.port}")
+                assert pub.wait_for_subscriber(5.0)
+            for i, pub in enumerate(pubs):
+                pub.send_multipart([f"kv@pod-{i}@m".encode(), b"p"])
+            deadline = time.monotonic() + 5.0
+            while time.monotonic() < deadline:
+                with lock:
+                    if len(received) == 3:
+                        break
+                time.sleep(0.01)
+            with lock:
+                assert sorted(received) == [
+                    b"kv@pod-0@m",
+                    b"kv@pod-1@m",
+                    b"kv@pod-2@m",
+                ]
+        finally:
+            for pub in pubs:
+                pub.close()
+            sub.close()
+
+
+class TestSubscriberToIndex:
+    def test_full_write_path_over_tcp(self):
+        """PUB (vLLM-sim) -> ZMTP -> subscriber -> sharded pool -> index."""
+        from llmd_kvcache_amd.kvblock import InMemoryIndex, InMemoryIndexConfig
+        from llmd_kvcache_amd.kvblock.keys import PodEntry
+        from llmd_kvcache_amd.kvblock.token_processor import (
+            ChunkedTokenDatabase,
+            TokenProcessorConfig,
+        )
+        from llmd_kvcache_amd.kvevents.events import BlockStored, EventBatch
+        from llmd_kvcache_amd.kvevents.pool import EventsConfig, EventsPool
+
+        index = InMemoryIndex(InMemoryIndexConfig(size=1000, pod_cache_size=10))
+        tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=4))
+        pool = EventsPool(
+            EventsConfig(zmq_endpoint="tcp://127.0.0.1:0", concurrency=2),
+            index,
+            tp,
+        )
+        pool.start(with_subscriber=True)
+        pub = PubSocket()
+        try:
+            deadline = time.monotonic() + 5.0
+            while pool._subscriber.port is None and time.monotonic() < deadline:
+                time.sleep(0.02)
+            assert pool._subscriber.port is not None
+            pub.connect(f"tcp://127.0.0.1:{pool._subscriber.port}")
+            assert pub.wait_for_subscriber(5.0)
+
+            batch = EventBatch(
+                ts=time.time(),
+                events=[BlockStored([100], None, [1, 2, 3, 4], 4)],
+            )
+            pub.send_multipart(
+                [b"kv@pod-a@m", struct.pack(">Q", 1), batch.encode()]
+            )
+
+            keys = tp.tokens_to_kv_block_keys(None, [1, 2, 3, 4], "m")
+            deadline = time.monotonic() + 5.0
+            result = {}
+            while time.monotonic() < deadline:
+                result = index.lookup(keys, set())
+                if keys[0] in result:
+                    break
+                time.sleep(0.02)
+            assert result.get(keys[0]) == [PodEntry("pod-a", "gpu")]
+        finally:
+            pub.close()
+            pool.shutdown()
