@@ -1,0 +1,505 @@
+"""Table-backed index: HBM3E-resident on GPU, same layout on CPU.
+
+This is the MI355X-native replacement for the reference's Go-map/LRU
+in-memory index (pkg/kvcache/kvblock/in_memory.go): an open-addressing
+hash table (ops/csrc/kvidx_common.h) probed by wave-cooperative HIP
+kernels on gfx950, or by the bit-identical C++ reference on CPU.
+
+Two concrete backends:
+ - ``NativeIndex``  - the table on CPU tensors (fast CPU mode; also the
+   golden reference for GPU differential tests);
+ - ``GpuIndex``     - the table in HBM on ``cuda:N``; inserts/evicts/
+   lookups are batched kernel launches; the read path can bypass the
+   generic ``lookup`` entirely via ``fused_scores`` (one kernel does
+   probe + longest-prefix scoring).
+
+String identities (pod, model, tier) are interned host-side into dense
+ids (Registry); the table stores only ids.
+"""
+
+from __future__ import annotations
+
+import threading
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Sequence, Set, Tuple
+
+import torch
+
+from ..scorer import default_kv_cache_backend_configs
+from .index import Index
+from .keys import Key, PodEntry
+
+MAX_TIERS = 4
+DEFAULT_CAPACITY = 1 << 21  # 2M slots; ~1M keys at 0.5 load factor
+DEFAULT_PODS_PER_KEY = 10  # in_memory.go:34
+MAX_PODS = 4096
+
+
+def _to_i64(h: int) -> int:
+    """uint64 -> int64 bit reinterpretation for torch tensors."""
+    return h - (1 << 64) if h >= (1 << 63) else h
+
+
+def _to_u64(v: int) -> int:
+    return v & 0xFFFFFFFFFFFFFFFF
+
+
+class Registry:
+    """Interns pod/model/tier strings to dense ids (thread-safe)."""
+
+    def __init__(self, tier_names: Optional[Sequence[str]] = None):
+        self._lock = threading.Lock()
+        self.pod_to_id: Dict[str, int] = {}
+        self.id_to_pod: List[str] = []
+        self.model_to_id: Dict[str, int] = {}
+        self.id_to_model: List[str] = []
+        tiers = list(tier_names or [b.name for b in default_kv_cache_backend_configs()])
+        self.tier_to_id: Dict[str, int] = {}
+        self.id_to_tier: List[str] = []
+        for t in tiers:
+            self._intern_tier(t)
+
+    def _intern_tier(self, tier: str) -> int:
+        tid = self.tier_to_id.get(tier)
+        if tid is None:
+            if len(self.id_to_tier) >= MAX_TIERS:
+                tid = MAX_TIERS - 1  # overflow tiers share the last slot
+                self.tier_to_id[tier] = tid
+                return tid
+            tid = len(self.id_to_tier)
+            self.tier_to_id[tier] = tid
+            self.id_to_tier.append(tier)
+        return tid
+
+    def pod_id(self, pod: str) -> int:
+        with self._lock:
+            pid = self.pod_to_id.get(pod)
+            if pid is None:
+                if len(self.id_to_pod) >= MAX_PODS:
+                    raise ValueError(f"pod registry full ({MAX_PODS})")
+                pid = len(self.id_to_pod)
+                self.pod_to_id[pod] = pid
+                self.id_to_pod.append(pod)
+            return pid
+
+    def model_id(self, model: str) -> int:
+        with self._lock:
+            mid = self.model_to_id.get(model)
+            if mid is None:
+                if len(self.id_to_model) >= 0xFFFF:
+                    raise ValueError("model registry full")
+                mid = len(self.id_to_model)
+                self.model_to_id[model] = mid
+                self.id_to_model.append(model)
+            return mid
+
+    def tier_id(self, tier: str) -> int:
+        with self._lock:
+            return self._intern_tier(tier)
+
+    @property
+    def num_pods(self) -> int:
+        with self._lock:
+            return len(self.id_to_pod)
+
+
+@dataclass
+class TableIndexConfig:
+    capacity: int = DEFAULT_CAPACITY
+    pods_per_key: int = DEFAULT_PODS_PER_KEY
+    device: str = "cpu"
+    tier_names: List[str] = field(
+        default_factory=lambda: [b.name for b in default_kv_cache_backend_configs()]
+    )
+
+
+@dataclass
+class GpuIndexConfig(TableIndexConfig):
+    device: str = "cuda:0"
+
+
+class KvTable:
+    """Tensor bundle + op dispatch for one table instance."""
+
+    def __init__(self, cfg: TableIndexConfig):
+        cap = cfg.capacity
+        if cap & (cap - 1):
+            raise ValueError("capacity must be a power of two")
+        self.cfg = cfg
+        self.device = torch.device(cfg.device)
+        self.is_cuda = self.device.type == "cuda"
+        from ..ops import cpu_ext
+
+        self.ops = cpu_ext.require()
+        if self.is_cuda and not self.ops.HAS_HIP:
+            raise RuntimeError(
+                "native extension built without HIP support but a GPU table "
+                "was requested - rebuild with ROCm torch"
+            )
+        opts = dict(device=self.device)
+        self.keys = torch.zeros(cap, dtype=torch.int64, **opts)
+        self.meta = torch.zeros(cap, dtype=torch.int32, **opts)
+        self.stamp = torch.zeros(cap, dtype=torch.int32, **opts)
+        self.pods = torch.zeros(cap * cfg.pods_per_key, dtype=torch.int32, **opts)
+        self.e_keys = torch.zeros(cap, dtype=torch.int64, **opts)
+        self.e_meta = torch.zeros(cap, dtype=torch.int32, **opts)
+        self.e_vals = torch.zeros(cap, dtype=torch.int64, **opts)
+        self._epoch = 0
+        self._epoch_lock = threading.Lock()
+
+    def next_epoch(self) -> int:
+        with self._epoch_lock:
+            self._epoch = (self._epoch + 1) & 0x7FFFFFFF
+            return self._epoch
+
+    def _t(self):
+        return (
+            self.keys,
+            self.meta,
+            self.stamp,
+            self.pods,
+            self.e_keys,
+            self.e_meta,
+            self.e_vals,
+            self.cfg.pods_per_key,
+        )
+
+    # -- raw ops (tensors in the table's device) -----------------------
+    def insert(self, engine_hashes, request_hashes, model_id, pod_entries):
+        fn = self.ops.gpu_insert if self.is_cuda else self.ops.cpu_insert
+        fn(*self._t(), engine_hashes, request_hashes, model_id, pod_entries,
+           self.next_epoch())
+
+    def evict(self, engine_hashes, model_id, pod_entries):
+        fn = self.ops.gpu_evict if self.is_cuda else self.ops.cpu_evict
+        fn(*self._t(), engine_hashes, model_id, pod_entries)
+
+    def lookup(self, request_hashes, model_id, filter_words, num_pods):
+        fn = self.ops.gpu_lookup if self.is_cuda else self.ops.cpu_lookup
+        return fn(*self._t(), request_hashes, model_id, filter_words,
+                  num_pods, self.next_epoch())
+
+    def fused_score(self, hashes, counts_or_offsets, model_id, filter_words,
+                    weights, num_pods, max_k=None):
+        if self.is_cuda:
+            return self.ops.gpu_fused_score(
+                *self._t(), hashes, counts_or_offsets, model_id, filter_words,
+                weights, num_pods, self.next_epoch(),
+                max_k if max_k is not None else 512)
+        return self.ops.cpu_fused_score(
+            *self._t(), hashes, counts_or_offsets, model_id, filter_words,
+            weights, num_pods, self.next_epoch())
+
+    def get_request_keys(self, engine_hashes, model_id):
+        fn = (self.ops.gpu_get_request_keys if self.is_cuda
+              else self.ops.cpu_get_request_keys)
+        return fn(*self._t(), engine_hashes, model_id)
+
+
+class TableIndex(Index):
+    """Index contract over a KvTable (CPU or GPU device)."""
+
+    def __init__(self, cfg: Optional[TableIndexConfig] = None,
+                 registry: Optional[Registry] = None):
+        self.cfg = cfg or TableIndexConfig()
+        self.table = KvTable(self.cfg)
+        self.registry = registry or Registry(self.cfg.tier_names)
+        self._write_lock = threading.Lock()  # CPU ops are single-writer
+        self.device = self.table.device
+
+    # -- helpers -------------------------------------------------------
+    def _hashes_tensor(self, keys: Sequence[Key]):
+        return torch.tensor([_to_i64(k.chunk_hash) for k in keys],
+                            dtype=torch.int64, device=self.device)
+
+    def _entries_tensor(self, entries: Sequence[PodEntry]):
+        vals = []
+        for e in entries:
+            pid = self.registry.pod_id(e.pod_identifier)
+            tid = self.registry.tier_id(e.device_tier)
+            vals.append((tid << 24) | (pid + 1))
+        return torch.tensor(vals, dtype=torch.int32, device=self.device)
+
+    def _filter_tensor(self, pod_identifier_set: Set[str], num_pods: int):
+        if not pod_identifier_set:
+            return torch.zeros(0, dtype=torch.int64, device=self.device)
+        W = (num_pods + 63) // 64
+        words = [0] * W
+        for pod in pod_identifier_set:
+            pid = self.registry.pod_to_id.get(pod)
+            if pid is not None and pid < num_pods:
+                words[pid // 64] |= 1 << (pid % 64)
+        return torch.tensor([_to_i64(w) for w in words], dtype=torch.int64,
+                            device=self.device)
+
+    def _num_pods_padded(self) -> int:
+        return max(64, (self.registry.num_pods + 63) // 64 * 64)
+
+    # -- Index contract ------------------------------------------------
+    def lookup(self, request_keys: Sequence[Key],
+               pod_identifier_set: Set[str]) -> Dict[Key, List[PodEntry]]:
+        if not request_keys:
+            raise ValueError("no request keys provided for lookup")
+        model_id = self.registry.model_id(request_keys[0].model_name)
+        num_pods = self._num_pods_padded()
+        hashes = self._hashes_tensor(request_keys)
+        filt = self._filter_tensor(pod_identifier_set, num_pods)
+        found, masks = self.table.lookup(hashes, model_id, filt, num_pods)
+        found = found.cpu().tolist()
+        masks = masks.cpu()
+        W = masks.shape[2]
+
+        result: Dict[Key, List[PodEntry]] = {}
+        for i, key in enumerate(request_keys):
+            f = found[i]
+            if f == 0:
+                continue  # absent: skip, keep walking (in_memory.go:141)
+            if f == 2:
+                break  # present-but-empty: chain cut (in_memory.go:118-121)
+            entries: List[PodEntry] = []
+            for t in range(MAX_TIERS):
+                tier_name = (self.registry.id_to_tier[t]
+                             if t < len(self.registry.id_to_tier) else None)
+                if tier_name is None:
+                    continue
+                for w in range(W):
+                    bits = _to_u64(int(masks[i, t, w]))
+                    while bits:
+                        b = (bits & -bits).bit_length() - 1
+                        bits &= bits - 1
+                        pid = w * 64 + b
+                        if pid < len(self.registry.id_to_pod):
+                            entries.append(PodEntry(
+                                self.registry.id_to_pod[pid], tier_name))
+            if entries:
+                result[key] = entries
+        return result
+
+    def add(self, engine_keys: Sequence[Key], request_keys: Sequence[Key],
+            entries: Sequence[PodEntry]) -> None:
+        if not engine_keys or not request_keys or not entries:
+            raise ValueError("no keys or entries provided for adding to index")
+        if len(engine_keys) != len(request_keys):
+            raise ValueError("mismatch between engine keys and request keys length")
+        model_id = self.registry.model_id(request_keys[0].model_name)
+        eh = self._hashes_tensor(engine_keys)
+        rh = self._hashes_tensor(request_keys)
+        pe = self._entries_tensor(entries)
+        with self._write_lock:
+            self.table.insert(eh, rh, model_id, pe)
+
+    def evict(self, engine_key: Key, entries: Sequence[PodEntry]) -> None:
+        if not entries:
+            raise ValueError("no entries provided for eviction from index")
+        model_id = self.registry.model_id(engine_key.model_name)
+        eh = self._hashes_tensor([engine_key])
+        pe = self._entries_tensor(entries)
+        with self._write_lock:
+            self.table.evict(eh, model_id, pe)
+
+    def get_request_key(self, engine_key: Key) -> Optional[Key]:
+        model_id = self.registry.model_id(engine_key.model_name)
+        eh = self._hashes_tensor([engine_key])
+        found, out = self.table.get_request_keys(eh, model_id)
+        if int(found[0]) == 0:
+            return None
+        return Key(engine_key.model_name, _to_u64(int(out[0])))
+
+    # -- fast paths ----------------------------------------------------
+    def fused_scores(self, hashes: torch.Tensor, counts_or_offsets: torch.Tensor,
+                     model_name: str, pod_identifier_set: Set[str],
+                     weights: Optional[torch.Tensor] = None,
+                     max_k: Optional[int] = None) -> torch.Tensor:
+        """Batched probe + longest-prefix score, one call.
+
+        hashes: int64 flat request hashes; counts (CPU table) or offsets
+        [B+1] int32 (GPU table). Returns float32 [B, num_pods] scores in
+        registry pod-id order."""
+        model_id = self.registry.model_id(model_name)
+        num_pods = self._num_pods_padded()
+        filt = self._filter_tensor(pod_identifier_set, num_pods)
+        if weights is None:
+            weights = self.tier_weights()
+        with self._write_lock if not self.table.is_cuda else _nullcontext():
+            return self.table.fused_score(hashes, counts_or_offsets, model_id,
+                                          filt, weights, num_pods, max_k)
+
+    def tier_weights(self, weight_map: Optional[Dict[str, float]] = None
+                     ) -> torch.Tensor:
+        """Per-tier-id weight vector; unknown tiers weigh 1.0
+        (kvblock_scorer.go:93-99)."""
+        if weight_map is None:
+            weight_map = {b.name: b.weight
+                          for b in default_kv_cache_backend_configs()}
+        w = [1.0] * MAX_TIERS
+        for i, name in enumerate(self.registry.id_to_tier):
+            w[i] = weight_map.get(name, 1.0)
+        return torch.tensor(w, dtype=torch.float32, device=self.device)
+
+    def scores_to_map(self, scores: torch.Tensor) -> List[Dict[str, float]]:
+        """float [B, num_pods] -> per-prompt {pod: score} (nonzero only)."""
+        out: List[Dict[str, float]] = []
+        sc = scores.cpu()
+        nz = sc != 0
+        for b in range(sc.shape[0]):
+            row: Dict[str, float] = {}
+            for pid in torch.nonzero(nz[b]).flatten().tolist():
+                if pid < len(self.registry.id_to_pod):
+                    row[self.registry.id_to_pod[pid]] = float(sc[b, pid])
+            out.append(row)
+        return out
+
+
+class _nullcontext:
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        return False
+
+
+class NativeIndex(TableIndex):
+    """CPU-tensor table (fast CPU backend + GPU-parity reference)."""
+
+    def __init__(self, cfg: Optional[TableIndexConfig] = None, **kw):
+        cfg = cfg or TableIndexConfig()
+        cfg.device = "cpu"
+        super().__init__(cfg, **kw)
+
+
+class GpuIndex(TableIndex):
+    """HBM3E-resident table probed by gfx950 HIP kernels.
+
+    Fails loudly if the HIP extension is missing or no GPU is visible -
+    there is no silent CPU fallback on a GPU host."""
+
+    def __init__(self, cfg: Optional[GpuIndexConfig] = None, **kw):
+        cfg = cfg or GpuIndexConfig()
+        if not torch.cuda.is_available():
+            raise RuntimeError(
+                "GpuIndex requires a visible AMD GPU (torch.cuda unavailable)"
+            )
+        super().__init__(cfg, **kw)
+        if not self.table.is_cuda:
+            raise RuntimeError("GpuIndexConfig.device must be a cuda device")
+
+    def apply_event_batches(self, batches: List[Tuple[str, str, list]],
+                            token_processor=None) -> None:
+        """Apply decoded KV event batches fully on-device.
+
+        batches: list of (pod_identifier, model_name, [events]) in arrival
+        order; per-pod ordering is preserved by grouping (one wave per pod
+        group processes its events serially - kvevents/pool.go:132-144).
+        """
+        from ..kvevents.events import (AllBlocksCleared, BlockRemoved,
+                                       BlockStored, get_hash_as_uint64)
+
+        if token_processor is None:
+            from .token_processor import ChunkedTokenDatabase
+
+            token_processor = ChunkedTokenDatabase()
+        block_size = token_processor.block_size
+        init_hash = token_processor.config.init_hash()
+
+        tokens: List[int] = []
+        tok_off = [0]
+        ehashes: List[int] = []
+        eh_off = [0]
+        parents: List[int] = []
+        has_parent: List[int] = []
+        ev_type: List[int] = []
+        pod_entry: List[int] = []
+        grp_off = [0]
+        model_ids: Set[int] = set()
+        model_id = 0
+
+        by_pod: Dict[Tuple[str, str], list] = {}
+        order: List[Tuple[str, str]] = []
+        for pod, model, events in batches:
+            k = (pod, model)
+            if k not in by_pod:
+                by_pod[k] = []
+                order.append(k)
+            by_pod[k].extend(events)
+
+        for pod, model in order:
+            model_id = self.registry.model_id(model)
+            model_ids.add(model_id)
+            pod_id = self.registry.pod_id(pod)
+            n_events_in_group = 0
+            for ev in by_pod[(pod, model)]:
+                if isinstance(ev, BlockStored):
+                    tier = self.registry.tier_id(
+                        ev.medium.lower() if ev.medium else "gpu")
+                    hs = []
+                    for raw in ev.block_hashes:
+                        try:
+                            hs.append(get_hash_as_uint64(raw))
+                        except Exception:
+                            continue
+                    if not hs:
+                        continue
+                    ehashes.extend(_to_i64(h) for h in hs)
+                    eh_off.append(len(ehashes))
+                    tokens.extend(int(t) for t in ev.token_ids)
+                    tok_off.append(len(tokens))
+                    if ev.parent_block_hash is not None:
+                        try:
+                            parents.append(
+                                _to_i64(get_hash_as_uint64(ev.parent_block_hash)))
+                            has_parent.append(1)
+                        except Exception:
+                            parents.append(0)
+                            has_parent.append(0)
+                    else:
+                        parents.append(0)
+                        has_parent.append(0)
+                    ev_type.append(0)
+                    pod_entry.append((tier << 24) | (pod_id + 1))
+                    n_events_in_group += 1
+                elif isinstance(ev, BlockRemoved):
+                    tier = self.registry.tier_id(
+                        ev.medium.lower() if ev.medium else "gpu")
+                    hs = []
+                    for raw in ev.block_hashes:
+                        try:
+                            hs.append(get_hash_as_uint64(raw))
+                        except Exception:
+                            continue
+                    if not hs:
+                        continue
+                    ehashes.extend(_to_i64(h) for h in hs)
+                    eh_off.append(len(ehashes))
+                    tok_off.append(len(tokens))
+                    parents.append(0)
+                    has_parent.append(0)
+                    ev_type.append(1)
+                    pod_entry.append((tier << 24) | (pod_id + 1))
+                    n_events_in_group += 1
+                elif isinstance(ev, AllBlocksCleared):
+                    continue
+            if n_events_in_group:
+                grp_off.append(grp_off[-1] + n_events_in_group)
+
+        if len(grp_off) == 1:
+            return
+        if len(model_ids) > 1:
+            raise ValueError(
+                "apply_event_batches supports one model per call; split by model")
+
+        d = self.device
+        i32 = torch.int32
+        self.table.ops.gpu_apply_events(
+            *self.table._t(),
+            torch.tensor(tokens, dtype=torch.int64, device=d),
+            torch.tensor(tok_off, dtype=i32, device=d),
+            torch.tensor(ehashes, dtype=torch.int64, device=d),
+            torch.tensor(eh_off, dtype=i32, device=d),
+            torch.tensor(parents, dtype=torch.int64, device=d),
+            torch.tensor(has_parent, dtype=torch.uint8, device=d),
+            torch.tensor(ev_type, dtype=torch.uint8, device=d),
+            torch.tensor(pod_entry, dtype=i32, device=d),
+            torch.tensor(grp_off, dtype=i32, device=d),
+            model_id, _to_i64(init_hash), block_size,
+            self.table.next_epoch(),
+        )

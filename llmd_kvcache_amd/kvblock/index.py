@@ -49,6 +49,7 @@ class Index:
 class IndexConfig:
     in_memory: Optional["InMemoryIndexConfig"] = None
     gpu: Optional["GpuIndexConfig"] = None
+    native: Optional["TableIndexConfig"] = None
     cost_aware: Optional["CostAwareMemoryIndexConfig"] = None
     valkey: Optional["RedisIndexConfig"] = None
     redis: Optional["RedisIndexConfig"] = None
@@ -76,6 +77,10 @@ def new_index(cfg: Optional[IndexConfig] = None) -> Index:
         from .gpu_index import GpuIndex
 
         idx = GpuIndex(cfg.gpu)
+    elif cfg.native is not None:
+        from .gpu_index import NativeIndex
+
+        idx = NativeIndex(cfg.native)
     elif cfg.cost_aware is not None:
         from .cost_aware import CostAwareMemoryIndex
 

@@ -40,8 +40,15 @@ def _fake_redis_factory():
     )
 
 
+def _native_factory():
+    from llmd_kvcache_amd.kvblock.gpu_index import NativeIndex, TableIndexConfig
+
+    return NativeIndex(TableIndexConfig(capacity=1 << 14, pods_per_key=10))
+
+
 FACTORIES = {
     "in_memory": _in_memory_factory,
+    "native": _native_factory,
     "cost_aware": _cost_aware_factory,
     "redis": _fake_redis_factory,
 }
