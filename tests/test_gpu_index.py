@@ -1,0 +1,268 @@
+"""GPU (gfx950) kernel tests: the HBM-resident table vs the bit-identical
+CPU reference ops, plus the fused lookup+score and hash-chain kernels vs
+the pure-Python golden implementations.  All tests require an MI355X."""
+
+import random
+
+import pytest
+
+torch = pytest.importorskip("torch")
+
+pytestmark = pytest.mark.gpu
+
+from llmd_kvcache_amd.kvblock.gpu_index import (  # noqa: E402
+    GpuIndex,
+    GpuIndexConfig,
+    NativeIndex,
+    TableIndexConfig,
+    _to_i64,
+    _to_u64,
+)
+from llmd_kvcache_amd.kvblock.keys import Key, PodEntry  # noqa: E402
+from llmd_kvcache_amd.kvblock.token_processor import (  # noqa: E402
+    ChunkedTokenDatabase,
+    TokenProcessorConfig,
+)
+from llmd_kvcache_amd.kvevents.events import BlockRemoved, BlockStored  # noqa: E402
+from llmd_kvcache_amd.scorer import new_kv_block_scorer  # noqa: E402
+from llmd_kvcache_amd.utils import hashing  # noqa: E402
+
+MODEL = "m"
+
+
+@pytest.fixture
+def gpu_index():
+    return GpuIndex(GpuIndexConfig(capacity=1 << 14, pods_per_key=10))
+
+
+def test_native_extension_is_loaded():
+    """The HIP extension must actually be the in-tree .so (no silent
+    eager fallback on a GPU box)."""
+    from llmd_kvcache_amd.ops import cpu_ext
+
+    mod = cpu_ext.require()
+    assert mod.HAS_HIP
+    assert torch.cuda.is_available()
+
+
+class TestGpuTableOps:
+    def test_basic_add_lookup_evict(self, gpu_index):
+        keys = [Key(MODEL, h) for h in (1, 2, 3)]
+        gpu_index.add(keys, keys, [PodEntry("pod-a", "gpu")])
+        result = gpu_index.lookup(keys, set())
+        assert set(result.keys()) == set(keys)
+        for k in keys:
+            assert result[k] == [PodEntry("pod-a", "gpu")]
+        assert gpu_index.get_request_key(Key(MODEL, 1)) == Key(MODEL, 1)
+        gpu_index.evict(Key(MODEL, 2), [PodEntry("pod-a", "gpu")])
+        result = gpu_index.lookup(keys, set())
+        assert Key(MODEL, 2) not in result
+
+    def test_dual_keys(self, gpu_index):
+        ek = [Key(MODEL, 100)]
+        rk = [Key(MODEL, 200)]
+        gpu_index.add(ek, rk, [PodEntry("pod-a", "gpu")])
+        assert gpu_index.get_request_key(Key(MODEL, 100)) == Key(MODEL, 200)
+        assert gpu_index.get_request_key(Key(MODEL, 999)) is None
+        gpu_index.evict(Key(MODEL, 100), [PodEntry("pod-a", "gpu")])
+        assert Key(MODEL, 200) not in gpu_index.lookup(rk + ek, set())
+
+    @pytest.mark.parametrize("seed", [1, 2])
+    def test_differential_vs_cpu_table(self, seed):
+        from tests.test_native_index import random_workload, run_op
+
+        rng = random.Random(seed)
+        cpu = NativeIndex(TableIndexConfig(capacity=1 << 12, pods_per_key=10))
+        gpu = GpuIndex(GpuIndexConfig(capacity=1 << 12, pods_per_key=10))
+        for op in random_workload(rng, n_ops=200):
+            r_cpu = run_op(cpu, op)
+            r_gpu = run_op(gpu, op)
+            if op[0] == "lookup" and r_cpu != "error":
+                assert set(r_cpu.keys()) == set(r_gpu.keys()), op
+                for k in r_cpu:
+                    assert set(r_cpu[k]) == set(r_gpu[k]), (op, k)
+
+
+class TestFusedScoreKernel:
+    @pytest.mark.parametrize("seed", [21, 22])
+    def test_matches_python_scorer(self, seed):
+        rng = random.Random(seed)
+        gpu = GpuIndex(GpuIndexConfig(capacity=1 << 14, pods_per_key=10))
+        scorer = new_kv_block_scorer()
+        for _ in range(200):
+            start = rng.randrange(150)
+            n = rng.randrange(1, 10)
+            keys = [Key(MODEL, 7000 + start + i) for i in range(n)]
+            pod = f"pod-{rng.randrange(32)}"
+            tier = rng.choice(["gpu", "cpu"])
+            gpu.add(keys, keys, [PodEntry(pod, tier)])
+
+        prompts = []
+        for _ in range(50):
+            start = rng.randrange(150)
+            n = rng.randrange(1, 32)
+            prompts.append([Key(MODEL, 7000 + start + i) for i in range(n)])
+
+        flat = [_to_i64(k.chunk_hash) for p in prompts for k in p]
+        offsets = [0]
+        for p in prompts:
+            offsets.append(offsets[-1] + len(p))
+        hashes = torch.tensor(flat, dtype=torch.int64, device="cuda")
+        offs = torch.tensor(offsets, dtype=torch.int32, device="cuda")
+        scores = gpu.fused_scores(hashes, offs, MODEL, set(),
+                                  max_k=max(len(p) for p in prompts))
+        maps = gpu.scores_to_map(scores)
+
+        for i, prompt in enumerate(prompts):
+            key_to_pods = gpu.lookup(prompt, set())
+            expected = {p: s for p, s in
+                        scorer.score(prompt, key_to_pods).items() if s != 0}
+            assert maps[i].keys() == expected.keys(), (i, maps[i], expected)
+            for p in expected:
+                assert maps[i][p] == pytest.approx(expected[p])
+
+    def test_pod_filter(self):
+        gpu = GpuIndex(GpuIndexConfig(capacity=1 << 12, pods_per_key=10))
+        keys = [Key(MODEL, 1), Key(MODEL, 2)]
+        gpu.add(keys, keys, [PodEntry("pod-a", "gpu"), PodEntry("pod-b", "gpu")])
+        hashes = torch.tensor([_to_i64(k.chunk_hash) for k in keys],
+                              dtype=torch.int64, device="cuda")
+        offs = torch.tensor([0, 2], dtype=torch.int32, device="cuda")
+        maps = gpu.scores_to_map(
+            gpu.fused_scores(hashes, offs, MODEL, {"pod-b"}, max_k=2))
+        assert maps[0] == {"pod-b": 2.0}
+
+
+class TestHashChainKernel:
+    @pytest.mark.parametrize("block_size", [4, 16])
+    def test_matches_python(self, block_size):
+        from llmd_kvcache_amd.ops import cpu_ext
+
+        mod = cpu_ext.require()
+        rng = random.Random(5)
+        B = 67  # not a multiple of 64: tail-wave coverage
+        toks, off = [], [0]
+        for _ in range(B):
+            n = rng.randrange(1, 12) * block_size
+            toks.extend(rng.randrange(0, 2**32) for _ in range(n))
+            off.append(len(toks))
+        parents = [hashing.init_hash("")] * B
+        tokens_t = torch.tensor(toks, dtype=torch.int64, device="cuda")
+        off_t = torch.tensor(off, dtype=torch.int64, device="cuda")
+        par_t = torch.tensor([_to_i64(p) for p in parents],
+                             dtype=torch.int64, device="cuda")
+        out, chunk_off = mod.gpu_hash_chain(tokens_t, off_t, par_t, block_size)
+        out = out.cpu().tolist()
+        chunk_off = chunk_off.cpu().tolist()
+        for b in range(B):
+            h = parents[b]
+            chunk_toks = toks[off[b]: off[b + 1]]
+            for c in range((off[b + 1] - off[b]) // block_size):
+                h = hashing.chunk_hash(
+                    h, chunk_toks[c * block_size:(c + 1) * block_size])
+                assert _to_u64(out[chunk_off[b] + c]) == h, (b, c)
+
+
+class TestApplyEventsKernel:
+    def test_stored_then_removed_in_order(self):
+        gpu = GpuIndex(GpuIndexConfig(capacity=1 << 12, pods_per_key=10))
+        tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=4))
+        ev1 = BlockStored([100, 200], None, list(range(8)), 4)
+        ev2 = BlockRemoved([100])
+        gpu.apply_event_batches([("pod-a", MODEL, [ev1, ev2])], tp)
+        torch.cuda.synchronize()
+        req = tp.tokens_to_kv_block_keys(None, list(range(8)), MODEL)
+        result = gpu.lookup(req, set())
+        assert req[0] not in result  # removed
+        assert result.get(req[1]) == [PodEntry("pod-a", "gpu")]
+
+    def test_parent_chain_stitching_on_device(self):
+        gpu = GpuIndex(GpuIndexConfig(capacity=1 << 12, pods_per_key=10))
+        tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=4))
+        t1, t2 = [1, 2, 3, 4], [5, 6, 7, 8]
+        gpu.apply_event_batches(
+            [("pod-a", MODEL, [BlockStored([100], None, t1, 4)])], tp)
+        gpu.apply_event_batches(
+            [("pod-a", MODEL, [BlockStored([200], 100, t2, 4)])], tp)
+        torch.cuda.synchronize()
+        full = tp.tokens_to_kv_block_keys(None, t1 + t2, MODEL)
+        result = gpu.lookup(full, set())
+        assert set(result.keys()) == set(full)
+
+    def test_matches_cpu_pool_digest(self):
+        """GPU on-device event application == CPU EventsPool.digest_events."""
+        from llmd_kvcache_amd.kvevents.pool import EventsConfig, EventsPool
+
+        rng = random.Random(9)
+        tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=4))
+        cpu_idx = NativeIndex(TableIndexConfig(capacity=1 << 12, pods_per_key=10))
+        gpu_idx = GpuIndex(GpuIndexConfig(capacity=1 << 12, pods_per_key=10))
+        pool = EventsPool(EventsConfig(concurrency=1), cpu_idx, tp)
+
+        batches = []
+        next_hash = 1000
+        chains = {}  # pod -> last engine hash
+        for _ in range(50):
+            pod = f"pod-{rng.randrange(4)}"
+            n_blocks = rng.randrange(1, 5)
+            toks = [rng.randrange(0, 1 << 31) for _ in range(n_blocks * 4)]
+            hs = list(range(next_hash, next_hash + n_blocks))
+            next_hash += n_blocks
+            parent = chains.get(pod)
+            ev = BlockStored(hs, parent, toks, 4)
+            chains[pod] = hs[-1]
+            batches.append((pod, MODEL, [ev]))
+
+        for pod, model, events in batches:
+            pool.digest_events(pod, model, events)
+        # apply to GPU in per-pod groups (same order)
+        gpu_idx.apply_event_batches(batches, tp)
+        torch.cuda.synchronize()
+
+        # compare a sample of request-key lookups
+        for pod, model, events in batches:
+            ev = events[0]
+            req = tp.tokens_to_kv_block_keys(None, ev.token_ids, model)
+            # per-event chains may differ (parent stitching), compare via
+            # engine->request mapping instead
+            for h in ev.block_hashes:
+                ck = Key(model, h)
+                assert (cpu_idx.get_request_key(ck) ==
+                        gpu_idx.get_request_key(ck)), ck
+        # spot-check pod visibility through engine-derived request keys
+        for h in range(1000, next_hash):
+            ck = Key(MODEL, h)
+            rk = cpu_idx.get_request_key(ck)
+            if rk is None:
+                continue
+            c = cpu_idx.lookup([rk], set())
+            g = gpu_idx.lookup([rk], set())
+            assert set(c.get(rk, [])) == set(g.get(rk, [])), rk
+
+
+class TestConcurrentGpuInserts:
+    def test_many_duplicate_inserts_converge(self):
+        """Thousands of threads inserting the same keys concurrently must
+        produce one consistent slot per key (lock-free claim correctness)."""
+        gpu = GpuIndex(GpuIndexConfig(capacity=1 << 14, pods_per_key=10))
+        keys = [Key(MODEL, 42)]
+        eh = torch.full((5000,), 42, dtype=torch.int64, device="cuda")
+        rh = torch.full((5000,), 43, dtype=torch.int64, device="cuda")
+        pe = gpu._entries_tensor([PodEntry("pod-a", "gpu")])
+        gpu.table.insert(eh, rh, gpu.registry.model_id(MODEL), pe)
+        torch.cuda.synchronize()
+        result = gpu.lookup([Key(MODEL, 43)], set())
+        assert result[Key(MODEL, 43)] == [PodEntry("pod-a", "gpu")]
+        assert gpu.get_request_key(Key(MODEL, 42)) == Key(MODEL, 43)
+
+    def test_concurrent_distinct_inserts(self):
+        gpu = GpuIndex(GpuIndexConfig(capacity=1 << 16, pods_per_key=10))
+        n = 20000
+        eh = torch.arange(1, n + 1, dtype=torch.int64, device="cuda")
+        rh = eh + 1_000_000
+        pe = gpu._entries_tensor([PodEntry("pod-a", "gpu")])
+        gpu.table.insert(eh, rh, gpu.registry.model_id(MODEL), pe)
+        torch.cuda.synchronize()
+        sample = [Key(MODEL, 1_000_001 + i) for i in range(0, n, 977)]
+        result = gpu.lookup(sample, set())
+        assert set(result.keys()) == set(sample)
