@@ -108,6 +108,11 @@ class TableIndexConfig:
     capacity: int = DEFAULT_CAPACITY
     pods_per_key: int = DEFAULT_PODS_PER_KEY
     device: str = "cpu"
+    # sharding: the main table stores only keys with
+    # chunk_hash % num_shards == shard_id; the engine map is replicated
+    # (parallel/sharded.py merges per-shard masks over RCCL)
+    shard_id: int = 0
+    num_shards: int = 1
     tier_names: List[str] = field(
         default_factory=lambda: [b.name for b in default_kv_cache_backend_configs()]
     )
@@ -168,7 +173,7 @@ class KvTable:
     def insert(self, engine_hashes, request_hashes, model_id, pod_entries):
         fn = self.ops.gpu_insert if self.is_cuda else self.ops.cpu_insert
         fn(*self._t(), engine_hashes, request_hashes, model_id, pod_entries,
-           self.next_epoch())
+           self.next_epoch(), self.cfg.shard_id, self.cfg.num_shards)
 
     def evict(self, engine_hashes, model_id, pod_entries):
         fn = self.ops.gpu_evict if self.is_cuda else self.ops.cpu_evict
@@ -501,5 +506,5 @@ class GpuIndex(TableIndex):
             torch.tensor(pod_entry, dtype=i32, device=d),
             torch.tensor(grp_off, dtype=i32, device=d),
             model_id, _to_i64(init_hash), block_size,
-            self.table.next_epoch(),
+            self.table.next_epoch(), self.cfg.shard_id, self.cfg.num_shards,
         )

@@ -14,7 +14,8 @@ std::vector<at::Tensor> hash_chain_batch(at::Tensor, at::Tensor, at::Tensor,
                                          int64_t);
 void cpu_insert(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                 at::Tensor, at::Tensor, int64_t, at::Tensor, at::Tensor,
-                int64_t, at::Tensor, int64_t);
+                int64_t, at::Tensor, int64_t, int64_t, int64_t);
+at::Tensor cpu_score_from_masks(at::Tensor, at::Tensor, at::Tensor, int64_t);
 void cpu_evict(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                at::Tensor, at::Tensor, int64_t, at::Tensor, int64_t,
                at::Tensor);
@@ -36,7 +37,7 @@ std::vector<at::Tensor> cpu_get_request_keys(at::Tensor, at::Tensor,
 // hip_ops.hip
 void gpu_insert(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                 at::Tensor, at::Tensor, int64_t, at::Tensor, at::Tensor,
-                int64_t, at::Tensor, int64_t);
+                int64_t, at::Tensor, int64_t, int64_t, int64_t);
 void gpu_evict(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                at::Tensor, at::Tensor, int64_t, at::Tensor, int64_t,
                at::Tensor);
@@ -60,7 +61,7 @@ void gpu_apply_events(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, int64_t, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, at::Tensor, int64_t,
-                      int64_t, int64_t, int64_t);
+                      int64_t, int64_t, int64_t, int64_t, int64_t);
 #endif
 
 }  // namespace kvidx
@@ -73,6 +74,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cpu_evict", &kvidx::cpu_evict);
   m.def("cpu_lookup", &kvidx::cpu_lookup);
   m.def("cpu_fused_score", &kvidx::cpu_fused_score);
+  m.def("cpu_score_from_masks", &kvidx::cpu_score_from_masks);
   m.def("cpu_get_request_keys", &kvidx::cpu_get_request_keys);
 #ifdef KVIDX_WITH_HIP
   m.attr("HAS_HIP") = true;

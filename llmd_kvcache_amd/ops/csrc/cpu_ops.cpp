@@ -241,11 +241,16 @@ std::vector<at::Tensor> hash_chain_batch(at::Tensor tokens, at::Tensor offsets,
   return {out, chunk_off};
 }
 
+// shard_id/num_shards: the main table stores only keys it owns
+// (request_hash % num_shards == shard_id); the engine->request map is
+// replicated on every shard so parent-chain stitching never needs a
+// cross-rank lookup (see parallel/sharded.py).
 void cpu_insert(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
                 at::Tensor pods, at::Tensor e_keys, at::Tensor e_meta,
                 at::Tensor e_vals, int64_t pods_per_key,
                 at::Tensor engine_hashes, at::Tensor request_hashes,
-                int64_t model_id, at::Tensor pod_entries, int64_t epoch) {
+                int64_t model_id, at::Tensor pod_entries, int64_t epoch,
+                int64_t shard_id, int64_t num_shards) {
   auto v = make_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
                      (int)pods_per_key);
   auto eh = engine_hashes.contiguous();
@@ -259,10 +264,57 @@ void cpu_insert(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
   TORCH_CHECK(rh.numel() == n, "engine/request key length mismatch");
   for (int64_t i = 0; i < n; ++i) {
     emap_put(v, ehp[i], (uint32_t)model_id, remap_hash(rhp[i]), (int32_t)epoch);
+    if (num_shards > 1 && (int64_t)(rhp[i] % (uint64_t)num_shards) != shard_id)
+      continue;
     int64_t slot = table_put(v, rhp[i], (uint32_t)model_id, (int32_t)epoch);
     for (int64_t j = 0; j < m; ++j)
       pod_set_add(v, slot, pep[j], (int32_t)epoch);
   }
+}
+
+// Walk precomputed (possibly all-reduce-merged) masks -> scores; CPU twin
+// of k_score_from_masks for gloo-backed multi-process tests.
+at::Tensor cpu_score_from_masks(at::Tensor masks, at::Tensor offsets,
+                                at::Tensor weights, int64_t num_pods) {
+  auto m = masks.contiguous();
+  auto off = offsets.contiguous();
+  const uint64_t* mp = reinterpret_cast<uint64_t*>(m.data_ptr<int64_t>());
+  const int32_t* op = off.data_ptr<int32_t>();
+  const float* wts = weights.data_ptr<float>();
+  int64_t B = off.numel() - 1;
+  int64_t W = (num_pods + 63) / 64;
+  TORCH_CHECK(W <= 64, "num_pods > 4096 not supported");
+  auto scores = at::zeros({B, num_pods}, at::kFloat);
+  float* sp = scores.data_ptr<float>();
+  at::parallel_for(0, B, 1, [&](int64_t begin, int64_t end) {
+    for (int64_t b = begin; b < end; ++b) {
+      uint64_t active[64];
+      float* out = sp + b * num_pods;
+      int64_t K = op[b + 1] - op[b];
+      for (int64_t k = 0; k < K; ++k) {
+        const uint64_t* mk = mp + (size_t)(op[b] + k) * MAX_TIERS * W;
+        bool any = false;
+        for (int64_t w = 0; w < W; ++w) {
+          uint64_t cur = 0;
+          for (int t = 0; t < MAX_TIERS; ++t) cur |= mk[t * W + w];
+          uint64_t act = (k == 0) ? cur : (active[w] & cur);
+          active[w] = act;
+          if (act) any = true;
+          uint64_t bits = act;
+          while (bits) {
+            int bit = __builtin_ctzll(bits);
+            bits &= bits - 1;
+            float wmax = 0.f;
+            for (int t = 0; t < MAX_TIERS; ++t)
+              if ((mk[t * W + w] >> bit) & 1) wmax = std::max(wmax, wts[t]);
+            out[w * 64 + bit] += wmax;
+          }
+        }
+        if (!any) break;
+      }
+    }
+  });
+  return scores;
 }
 
 void cpu_evict(at::Tensor keys, at::Tensor meta, at::Tensor stamp,

@@ -230,10 +230,13 @@ DEV int dev_probe_collect(const DevTable& v, uint64_t h, uint32_t model,
 __global__ void k_insert(DevTable v, const uint64_t* __restrict__ eh,
                          const uint64_t* __restrict__ rh, int64_t n,
                          uint32_t model, const uint32_t* __restrict__ entries,
-                         int n_entries, int32_t epoch) {
+                         int n_entries, int32_t epoch, int shard_id,
+                         int num_shards) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   dev_emap_put(v, eh[i], model, remap_hash(rh[i]));
+  if (num_shards > 1 && (int)(rh[i] % (uint64_t)num_shards) != shard_id)
+    return;  // engine map replicated; main table sharded by ownership
   int64_t slot = dev_table_put(v, rh[i], model, epoch);
   for (int j = 0; j < n_entries; ++j)
     dev_pod_set_add(v, slot, entries[j], epoch);
@@ -441,7 +444,7 @@ __global__ void k_apply_events(
     const uint32_t* __restrict__ pod_entry,  // [E]
     const int32_t* __restrict__ grp_off,     // [G+1] event groups by pod
     int64_t G, uint32_t model, uint64_t init_hash, int block_size,
-    int32_t epoch,
+    int32_t epoch, int shard_id, int num_shards,
     uint64_t* __restrict__ req_scratch) {    // [total engine hashes]
   const int g = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const int lane = threadIdx.x & 63;
@@ -500,6 +503,9 @@ __global__ void k_apply_events(
     const int n_ins = min(n_chunks, nh);
     for (int i = lane; i < n_ins; i += 64) {
       dev_emap_put(v, eh[i], model, remap_hash(req[i]));
+      if (num_shards > 1 &&
+          (int)(remap_hash(req[i]) % (uint64_t)num_shards) != shard_id)
+        continue;
       int64_t slot = dev_table_put(v, req[i], model, epoch);
       dev_pod_set_add(v, slot, pod_entry[e], epoch);
     }
@@ -538,7 +544,8 @@ void gpu_insert(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
                 at::Tensor pods, at::Tensor e_keys, at::Tensor e_meta,
                 at::Tensor e_vals, int64_t pods_per_key,
                 at::Tensor engine_hashes, at::Tensor request_hashes,
-                int64_t model_id, at::Tensor pod_entries, int64_t epoch) {
+                int64_t model_id, at::Tensor pod_entries, int64_t epoch,
+                int64_t shard_id, int64_t num_shards) {
   auto v = dev_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
                     pods_per_key);
   int64_t n = engine_hashes.numel();
@@ -550,7 +557,8 @@ void gpu_insert(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
                      (uint32_t)model_id,
                      reinterpret_cast<const uint32_t*>(
                          pod_entries.data_ptr<int32_t>()),
-                     (int)pod_entries.numel(), (int32_t)epoch);
+                     (int)pod_entries.numel(), (int32_t)epoch,
+                     (int)shard_id, (int)num_shards);
 }
 
 void gpu_evict(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
@@ -695,7 +703,7 @@ void gpu_apply_events(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
                       at::Tensor ev_type, at::Tensor pod_entry,
                       at::Tensor grp_off, int64_t model_id,
                       int64_t init_hash_bits, int64_t block_size,
-                      int64_t epoch) {
+                      int64_t epoch, int64_t shard_id, int64_t num_shards) {
   auto v = dev_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
                     pods_per_key);
   int64_t G = grp_off.numel() - 1;
@@ -712,6 +720,7 @@ void gpu_apply_events(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
       reinterpret_cast<const uint32_t*>(pod_entry.data_ptr<int32_t>()),
       grp_off.data_ptr<int32_t>(), G, (uint32_t)model_id,
       (uint64_t)init_hash_bits, (int)block_size, (int32_t)epoch,
+      (int)shard_id, (int)num_shards,
       reinterpret_cast<uint64_t*>(req_scratch.data_ptr<int64_t>()));
 }
 

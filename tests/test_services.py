@@ -1,0 +1,167 @@
+"""Service-surface tests: proto3 codec golden bytes, gRPC round trip,
+HTTP endpoints - the reference's L6 API layer (api/indexer.proto,
+examples/kv_events/online/main.go:269-365)."""
+
+import json
+import threading
+import urllib.request
+
+import pytest
+
+from llmd_kvcache_amd.indexer import Config, Indexer
+from llmd_kvcache_amd.kvblock import InMemoryIndex, InMemoryIndexConfig
+from llmd_kvcache_amd.kvblock.keys import Key, PodEntry
+from llmd_kvcache_amd.service import proto
+from llmd_kvcache_amd.tokenization.pool import TokenizationPool
+from llmd_kvcache_amd.tokenization.tokenizer import Tokenizer
+
+
+class FixedTokenizer(Tokenizer):
+    """Deterministic tokenizer: 1 token per 4 chars."""
+
+    @property
+    def type(self):
+        return "fixed"
+
+    def encode(self, prompt, model_name, add_special_tokens=True):
+        tokens, offsets = [], []
+        for i, start in enumerate(range(0, len(prompt) - 3, 4)):
+            tokens.append(100 + (hash(prompt[start : start + 4]) % 1000))
+            offsets.append((start, start + 4))
+        return tokens, offsets
+
+    def render_chat_template(self, req):
+        return json.dumps(req.conversations)
+
+
+def make_indexer():
+    cfg = Config()
+    cfg.token_processor.block_size = 4
+    index = InMemoryIndex(InMemoryIndexConfig(size=10_000, pod_cache_size=10))
+    pool = TokenizationPool(cfg.tokenizers_pool, tokenizer=FixedTokenizer())
+    idx = Indexer(cfg, tokenization_pool=pool, kv_block_index=index)
+    return idx, index
+
+
+class TestProtoCodec:
+    def test_request_golden_bytes(self):
+        req = proto.GetPodScoresRequest(
+            prompt="hi", model_name="m", pod_identifiers=["a", "b"]
+        )
+        # field1 "hi", field2 "m", field3 "a", field3 "b"
+        assert req.encode() == (
+            b"\x0a\x02hi" b"\x12\x01m" b"\x1a\x01a" b"\x1a\x01b"
+        )
+        back = proto.GetPodScoresRequest.decode(req.encode())
+        assert back == req
+
+    def test_response_roundtrip(self):
+        resp = proto.GetPodScoresResponse(
+            scores=[proto.PodScore("pod-a", 3.5), proto.PodScore("pod-b", 0.5)]
+        )
+        back = proto.GetPodScoresResponse.decode(resp.encode())
+        assert back == resp
+
+    def test_double_encoding(self):
+        ps = proto.PodScore("p", 1.0)
+        # field 2, wire type 1 (fixed64), little-endian IEEE754 1.0
+        assert ps.encode()[-9:] == b"\x11\x00\x00\x00\x00\x00\x00\xf0?"
+
+    def test_empty_request(self):
+        assert proto.GetPodScoresRequest.decode(b"") == proto.GetPodScoresRequest()
+
+    def test_varint_boundaries(self):
+        long_prompt = "x" * 300  # forces 2-byte varint length
+        req = proto.GetPodScoresRequest(prompt=long_prompt)
+        assert proto.GetPodScoresRequest.decode(req.encode()).prompt == long_prompt
+
+
+class TestGrpcService:
+    def test_get_pod_scores_end_to_end(self):
+        from llmd_kvcache_amd.service.grpc_server import (
+            IndexerClient,
+            serve,
+        )
+
+        idx, index = make_indexer()
+        # seed the index through the write-path types
+        prompt = "abcdefghijklmnopqrstuvwxyz" * 10
+        tokens = FixedTokenizer().encode(prompt, "m")[0]
+        keys = idx.tokens_processor.tokens_to_kv_block_keys(None, tokens, "m")
+        index.add(keys, keys, [PodEntry("pod-a", "gpu")])
+
+        server = serve(idx, address="127.0.0.1:0")
+        try:
+            client = IndexerClient(f"127.0.0.1:{server._kvidx_port}")
+            scores = client.get_pod_scores(prompt, "m")
+            assert scores.get("pod-a", 0) > 0
+            # unknown prompt scores empty
+            assert client.get_pod_scores("zzzz" * 50, "m") == {}
+            client.close()
+        finally:
+            server.stop(None)
+
+    def test_pod_filter(self):
+        from llmd_kvcache_amd.service.grpc_server import IndexerClient, serve
+
+        idx, index = make_indexer()
+        prompt = "abcdefgh" * 20
+        tokens = FixedTokenizer().encode(prompt, "m")[0]
+        keys = idx.tokens_processor.tokens_to_kv_block_keys(None, tokens, "m")
+        index.add(keys, keys, [PodEntry("pod-a", "gpu"), PodEntry("pod-b", "gpu")])
+        server = serve(idx, address="127.0.0.1:0")
+        try:
+            client = IndexerClient(f"127.0.0.1:{server._kvidx_port}")
+            scores = client.get_pod_scores(prompt, "m", ["pod-b"])
+            assert "pod-a" not in scores and scores.get("pod-b", 0) > 0
+            client.close()
+        finally:
+            server.stop(None)
+
+
+class TestHttpService:
+    def test_score_completions_and_metrics(self):
+        from llmd_kvcache_amd.service.http_server import HttpService
+
+        idx, index = make_indexer()
+        prompt = "hello world, this is a test prompt!" * 8
+        tokens = FixedTokenizer().encode(prompt, "m")[0]
+        keys = idx.tokens_processor.tokens_to_kv_block_keys(None, tokens, "m")
+        index.add(keys, keys, [PodEntry("pod-x", "gpu")])
+
+        svc = HttpService(idx, host="127.0.0.1", port=0)
+        svc.start()
+        try:
+            body = json.dumps({"prompt": prompt, "model": "m"}).encode()
+            req = urllib.request.Request(
+                f"http://127.0.0.1:{svc.port}/score_completions",
+                data=body,
+                headers={"Content-Type": "application/json"},
+            )
+            with urllib.request.urlopen(req, timeout=5) as resp:
+                scores = json.loads(resp.read())
+            assert scores.get("pod-x", 0) > 0
+
+            # missing prompt -> 400
+            req = urllib.request.Request(
+                f"http://127.0.0.1:{svc.port}/score_completions",
+                data=json.dumps({"model": "m"}).encode(),
+                headers={"Content-Type": "application/json"},
+            )
+            with pytest.raises(urllib.error.HTTPError) as ei:
+                urllib.request.urlopen(req, timeout=5)
+            assert ei.value.code == 400
+
+            # /metrics exposition
+            with urllib.request.urlopen(
+                f"http://127.0.0.1:{svc.port}/metrics", timeout=5
+            ) as resp:
+                assert resp.status == 200
+
+            # /health
+            with urllib.request.urlopen(
+                f"http://127.0.0.1:{svc.port}/health", timeout=5
+            ) as resp:
+                assert json.loads(resp.read())["status"] == "ok"
+        finally:
+            svc.stop()

@@ -1,0 +1,182 @@
+"""Multi-process sharded-index tests over gloo (world_size 2, CPU) -
+the distributed path the driver's 8-GPU RCCL bench exercises, minus the
+device type.  Spawned with torch.multiprocessing; rendezvous on
+127.0.0.1 with a free port."""
+
+import os
+import random
+import socket
+
+import pytest
+
+torch = pytest.importorskip("torch")
+import torch.distributed as dist  # noqa: E402
+import torch.multiprocessing as mp  # noqa: E402
+
+from llmd_kvcache_amd.ops import cpu_ext  # noqa: E402
+
+pytestmark = pytest.mark.skipif(
+    cpu_ext.maybe_load() is None, reason="native extension not built"
+)
+
+MODEL = "m"
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _worker(rank, world_size, port, fn_name, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world_size)
+        result = globals()[fn_name](rank, world_size)
+        q.put((rank, "ok", result))
+    except Exception as e:  # pragma: no cover
+        import traceback
+
+        q.put((rank, "err", f"{e}\n{traceback.format_exc()}"))
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def run_distributed(fn_name, world_size=2):
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [
+        ctx.Process(target=_worker, args=(r, world_size, port, fn_name, q))
+        for r in range(world_size)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in procs:
+        rank, status, payload = q.get()
+        assert status == "ok", f"rank {rank} failed: {payload}"
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=30)
+    return results
+
+
+# ---- distributed bodies (run inside workers) -------------------------
+
+
+def _body_basic_sharded_score(rank, world_size):
+    from llmd_kvcache_amd.kvblock.gpu_index import TableIndexConfig
+    from llmd_kvcache_amd.kvblock.keys import Key, PodEntry
+    from llmd_kvcache_amd.parallel.sharded import ShardedIndex
+
+    idx = ShardedIndex(TableIndexConfig(capacity=1 << 12, pods_per_key=10))
+    # identical replicated writes on every rank
+    keys = [Key(MODEL, 100 + i) for i in range(8)]
+    idx.add(keys, keys, [PodEntry("pod-a", "gpu")])
+    idx.add(keys[:4], keys[:4], [PodEntry("pod-b", "cpu")])
+
+    # ownership actually sharded: local table holds only owned keys
+    owned = sum(
+        1 for k in keys
+        if len(idx.local.lookup([k, Key(MODEL, 999)], set()).get(k, []))
+    )
+    assert owned < len(keys) or world_size == 1
+
+    scores = idx.score_keys(keys, set())
+    return (rank, owned, scores)
+
+
+def test_sharded_score_merges_across_ranks():
+    results = run_distributed("_body_basic_sharded_score")
+    # both ranks computed identical merged scores
+    s0 = results[0][2]
+    s1 = results[1][2]
+    assert s0 == s1
+    assert s0["pod-a"] == pytest.approx(8.0)
+    assert s0["pod-b"] == pytest.approx(4 * 0.8)
+    # keys were distributed: neither rank owned everything
+    assert results[0][1] + results[1][1] == 8
+
+
+def _body_matches_single_process(rank, world_size):
+    from llmd_kvcache_amd.kvblock.gpu_index import (
+        NativeIndex,
+        TableIndexConfig,
+        _to_i64,
+    )
+    from llmd_kvcache_amd.kvblock.keys import Key, PodEntry
+    from llmd_kvcache_amd.parallel.sharded import ShardedIndex
+    from llmd_kvcache_amd.scorer import new_kv_block_scorer
+
+    rng = random.Random(77)  # same seed on every rank: replicated stream
+    sharded = ShardedIndex(TableIndexConfig(capacity=1 << 12, pods_per_key=10))
+    single = NativeIndex(TableIndexConfig(capacity=1 << 12, pods_per_key=10))
+    scorer = new_kv_block_scorer()
+
+    for _ in range(100):
+        start = rng.randrange(60)
+        n = rng.randrange(1, 8)
+        keys = [Key(MODEL, 3000 + start + i) for i in range(n)]
+        pod = f"pod-{rng.randrange(10)}"
+        tier = rng.choice(["gpu", "cpu"])
+        sharded.add(keys, keys, [PodEntry(pod, tier)])
+        single.add(keys, keys, [PodEntry(pod, tier)])
+
+    mismatches = []
+    for _ in range(30):
+        start = rng.randrange(60)
+        n = rng.randrange(1, 12)
+        prompt = [Key(MODEL, 3000 + start + i) for i in range(n)]
+        got = sharded.score_keys(prompt, set())
+        expected = {
+            p: s
+            for p, s in scorer.score(prompt, single.lookup(prompt, set())).items()
+            if s != 0
+        }
+        if set(got) != set(expected) or any(
+            abs(got[p] - expected[p]) > 1e-4 for p in expected
+        ):
+            mismatches.append((prompt, got, expected))
+    return mismatches
+
+
+def test_sharded_matches_single_process_scoring():
+    results = run_distributed("_body_matches_single_process")
+    for rank, mismatches in results.items():
+        assert mismatches == [], f"rank {rank}: {mismatches[:3]}"
+
+
+def _body_batch_scores(rank, world_size):
+    from llmd_kvcache_amd.kvblock.gpu_index import TableIndexConfig, _to_i64
+    from llmd_kvcache_amd.kvblock.keys import Key, PodEntry
+    from llmd_kvcache_amd.parallel.sharded import ShardedIndex
+
+    idx = ShardedIndex(TableIndexConfig(capacity=1 << 12, pods_per_key=10))
+    keys = [Key(MODEL, 500 + i) for i in range(6)]
+    idx.add(keys, keys, [PodEntry("pod-a", "gpu")])
+
+    prompts = [keys[:3], keys, [Key(MODEL, 9999)]]
+    flat = [_to_i64(k.chunk_hash) for p in prompts for k in p]
+    offsets = [0]
+    for p in prompts:
+        offsets.append(offsets[-1] + len(p))
+    scores = idx.sharded_scores(
+        torch.tensor(flat, dtype=torch.int64),
+        torch.tensor(offsets, dtype=torch.int32),
+        MODEL,
+        set(),
+    )
+    return idx.local.scores_to_map(scores)
+
+
+def test_sharded_batch_scores():
+    results = run_distributed("_body_batch_scores")
+    for rank, maps in results.items():
+        assert maps[0] == {"pod-a": 3.0}
+        assert maps[1] == {"pod-a": 6.0}
+        assert maps[2] == {}
