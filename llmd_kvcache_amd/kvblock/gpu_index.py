@@ -179,10 +179,14 @@ class KvTable:
         fn = self.ops.gpu_evict if self.is_cuda else self.ops.cpu_evict
         fn(*self._t(), engine_hashes, model_id, pod_entries)
 
-    def lookup(self, request_hashes, model_id, filter_words, num_pods):
+    def lookup(self, request_hashes, model_id, filter_words, num_pods,
+               sharded=True):
         fn = self.ops.gpu_lookup if self.is_cuda else self.ops.cpu_lookup
+        cfg = self.cfg
+        shard_id = cfg.shard_id if sharded else 0
+        num_shards = cfg.num_shards if sharded else 1
         return fn(*self._t(), request_hashes, model_id, filter_words,
-                  num_pods, self.next_epoch())
+                  num_pods, self.next_epoch(), shard_id, num_shards)
 
     def fused_score(self, hashes, counts_or_offsets, model_id, filter_words,
                     weights, num_pods, max_k=None):

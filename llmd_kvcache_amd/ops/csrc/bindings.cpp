@@ -22,7 +22,8 @@ void cpu_evict(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
 std::vector<at::Tensor> cpu_lookup(at::Tensor, at::Tensor, at::Tensor,
                                    at::Tensor, at::Tensor, at::Tensor,
                                    at::Tensor, int64_t, at::Tensor, int64_t,
-                                   at::Tensor, int64_t, int64_t);
+                                   at::Tensor, int64_t, int64_t, int64_t,
+                                   int64_t);
 at::Tensor cpu_fused_score(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                            at::Tensor, at::Tensor, at::Tensor, int64_t,
                            at::Tensor, at::Tensor, int64_t, at::Tensor,
@@ -49,7 +50,8 @@ std::vector<at::Tensor> gpu_get_request_keys(at::Tensor, at::Tensor,
 std::vector<at::Tensor> gpu_lookup(at::Tensor, at::Tensor, at::Tensor,
                                    at::Tensor, at::Tensor, at::Tensor,
                                    at::Tensor, int64_t, at::Tensor, int64_t,
-                                   at::Tensor, int64_t, int64_t);
+                                   at::Tensor, int64_t, int64_t, int64_t,
+                                   int64_t);
 at::Tensor gpu_fused_score(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                            at::Tensor, at::Tensor, at::Tensor, int64_t,
                            at::Tensor, at::Tensor, int64_t, at::Tensor,

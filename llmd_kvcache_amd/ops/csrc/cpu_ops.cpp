@@ -359,7 +359,8 @@ std::vector<at::Tensor> cpu_lookup(at::Tensor keys, at::Tensor meta,
                                    at::Tensor e_vals, int64_t pods_per_key,
                                    at::Tensor request_hashes, int64_t model_id,
                                    at::Tensor filter_words, int64_t num_pods,
-                                   int64_t epoch) {
+                                   int64_t epoch, int64_t shard_id,
+                                   int64_t num_shards) {
   auto v = make_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
                      (int)pods_per_key);
   auto rh = request_hashes.contiguous();
@@ -377,6 +378,9 @@ std::vector<at::Tensor> cpu_lookup(at::Tensor keys, at::Tensor meta,
   uint64_t* mp = reinterpret_cast<uint64_t*>(masks.data_ptr<int64_t>());
 
   for (int64_t k = 0; k < K; ++k) {
+    if (num_shards > 1 &&
+        (int64_t)(remap_hash(rhp[k]) % (uint64_t)num_shards) != shard_id)
+      continue;  // unowned key: another shard's mask contribution
     int64_t slot = table_find(v, rhp[k], (uint32_t)model_id);
     if (slot < 0) continue;
     v.stamp[slot] = (int32_t)epoch;  // LRU touch on read

@@ -286,10 +286,14 @@ __global__ void k_lookup_masks(DevTable v, const uint64_t* __restrict__ rh,
                                int64_t K, uint32_t model,
                                const uint64_t* __restrict__ filter,
                                int has_filter, int num_pods, int W,
-                               int32_t epoch, uint8_t* __restrict__ found,
+                               int32_t epoch, int shard_id, int num_shards,
+                               uint8_t* __restrict__ found,
                                unsigned long long* __restrict__ masks) {
   int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (k >= K) return;
+  if (num_shards > 1 &&
+      (int)(remap_hash(rh[k]) % (uint64_t)num_shards) != shard_id)
+    return;  // unowned key: contributes zero masks on this shard
   unsigned long long* mk = masks + (size_t)k * MAX_TIERS * W;
   int f = dev_probe_collect(
       v, rh[k], model, filter, has_filter, num_pods, W, epoch,
@@ -604,7 +608,8 @@ std::vector<at::Tensor> gpu_lookup(at::Tensor keys, at::Tensor meta,
                                    at::Tensor e_vals, int64_t pods_per_key,
                                    at::Tensor request_hashes, int64_t model_id,
                                    at::Tensor filter_words, int64_t num_pods,
-                                   int64_t epoch) {
+                                   int64_t epoch, int64_t shard_id,
+                                   int64_t num_shards) {
   auto v = dev_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
                     pods_per_key);
   int64_t K = request_hashes.numel();
@@ -619,7 +624,8 @@ std::vector<at::Tensor> gpu_lookup(at::Tensor keys, at::Tensor meta,
       k_lookup_masks, dim3(blocks), dim3(threads), 0, STREAM, v,
       U64P(request_hashes), K, (uint32_t)model_id,
       has_filter ? U64P(filter_words) : nullptr, has_filter ? 1 : 0,
-      (int)num_pods, (int)W, (int32_t)epoch, found.data_ptr<uint8_t>(),
+      (int)num_pods, (int)W, (int32_t)epoch, (int)shard_id, (int)num_shards,
+      found.data_ptr<uint8_t>(),
       reinterpret_cast<unsigned long long*>(masks.data_ptr<int64_t>()));
   return {found, masks};
 }
