@@ -126,14 +126,68 @@ class EventsPool:
     # -- workers -------------------------------------------------------
     def _worker(self, index: int) -> None:
         q = self.queues[index]
+        fast = hasattr(self.index, "apply_event_batches") and getattr(
+            getattr(self.index, "table", None), "is_cuda", False
+        )
         while True:
             msg = q.get()
-            try:
-                if msg is None:
-                    return
-                self.process_event(msg)
-            finally:
+            if msg is None:
                 q.task_done()
+                return
+            if not fast:
+                try:
+                    self.process_event(msg)
+                finally:
+                    q.task_done()
+                continue
+            # GPU fast path: drain a burst of queued messages and apply
+            # them in ONE on-device kernel launch (per-pod ordering is
+            # preserved: this queue owns its pods, and the kernel
+            # processes each pod group's events serially in FIFO order).
+            burst: List[Message] = [msg]
+            stop = False
+            while len(burst) < 64:
+                try:
+                    nxt = q.get_nowait()
+                except queue.Empty:
+                    break
+                if nxt is None:
+                    stop = True
+                    break
+                burst.append(nxt)
+            try:
+                self._process_burst_gpu(burst)
+            finally:
+                for _ in burst:
+                    q.task_done()
+                if stop:
+                    q.task_done()
+            if stop:
+                return
+
+    def _process_burst_gpu(self, burst: List[Message]) -> None:
+        batches = []
+        for msg in burst:
+            try:
+                batch = decode_event_batch(msg.payload)
+            except DecodeError as e:
+                logger.debug("dropping poison-pill message: %s", e)
+                continue
+            # split by model: the on-device kernel takes one model per call
+            batches.append((msg.pod_identifier, msg.model_name, batch.events))
+        if not batches:
+            return
+        by_model = {}
+        for pod, model, events in batches:
+            by_model.setdefault(model, []).append((pod, model, events))
+        for model, group in by_model.items():
+            try:
+                self.index.apply_event_batches(group, self.token_processor)
+            except Exception:
+                logger.exception("on-device event application failed; "
+                                 "falling back to per-event path")
+                for pod, model_name, events in group:
+                    self.digest_events(pod, model_name, events)
 
     def process_event(self, msg: Message) -> None:
         try:
