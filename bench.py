@@ -211,6 +211,8 @@ def main():
     key_offsets = torch.arange(0, (args.batch + 1) * KEYS_PER_PROMPT,
                                KEYS_PER_PROMPT, dtype=torch.int32,
                                device=device)
+    nchunks_t = torch.full((args.batch,), KEYS_PER_PROMPT, dtype=torch.int32,
+                           device=device)
     ops = index.table.ops
 
     def score_call(call_idx):
@@ -222,7 +224,13 @@ def main():
         off = torch.arange(0, (args.batch + 1) * PROMPT_TOKENS, PROMPT_TOKENS,
                            dtype=torch.int64, device=device)
         if index.table.is_cuda:
-            hashes, _ = ops.gpu_hash_chain(toks, off, parents, BLOCK_SIZE)
+            # transposed int32 staging: coalesced lane loads in the chain
+            # kernel (see profiles/r01_bench1_kernel_stats.md)
+            t32 = toks.view(args.batch, PROMPT_TOKENS).to(torch.int32)
+            hashes_t = ops.gpu_hash_chain_tr(
+                t32.t().contiguous(), parents, nchunks_t, BLOCK_SIZE,
+                KEYS_PER_PROMPT)
+            hashes = hashes_t.t().contiguous().view(-1)
             if sharded is not None:
                 scores = sharded.sharded_scores(hashes, key_offsets, MODEL,
                                                 set(), weights)

@@ -59,6 +59,8 @@ at::Tensor gpu_fused_score(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
 at::Tensor gpu_score_from_masks(at::Tensor, at::Tensor, at::Tensor, int64_t);
 std::vector<at::Tensor> gpu_hash_chain(at::Tensor, at::Tensor, at::Tensor,
                                        int64_t);
+at::Tensor gpu_hash_chain_tr(at::Tensor, at::Tensor, at::Tensor, int64_t,
+                             int64_t);
 void gpu_apply_events(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, int64_t, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, at::Tensor,
@@ -87,6 +89,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gpu_fused_score", &kvidx::gpu_fused_score);
   m.def("gpu_score_from_masks", &kvidx::gpu_score_from_masks);
   m.def("gpu_hash_chain", &kvidx::gpu_hash_chain);
+  m.def("gpu_hash_chain_tr", &kvidx::gpu_hash_chain_tr);
   m.def("gpu_apply_events", &kvidx::gpu_apply_events);
 #else
   m.attr("HAS_HIP") = false;

@@ -431,6 +431,44 @@ __global__ void k_hash_chain(const int64_t* __restrict__ tokens,
   }
 }
 
+// Transposed-layout hash chains (the read-path hot kernel): tokens are
+// staged [token_pos][prompt] in int32, so at each chunk step all 64 lanes
+// of a wave load one coalesced 256 B line per token position, and the
+// next chunk's 16 independent loads prefetch under the current chunk's
+// serial FNV ALU chain.  The row-major int64 variant (k_hash_chain)
+// measured 2.5 ms per 512x512-chunk call on MI355X - pure HBM latency on
+// strided per-lane loads; this layout removes that.
+// out is [max_chunks][B] (transposed too; callers transpose back with a
+// cheap torch op or consume the stride directly).
+template <int BS>  // compile-time block size keeps tok[] in registers
+__global__ void k_hash_chain_tr(const int32_t* __restrict__ tokens_t,  // [T,B]
+                                const uint64_t* __restrict__ parents,  // [B]
+                                const int32_t* __restrict__ n_chunks,  // [B]
+                                int64_t B, int max_chunks,
+                                uint64_t* __restrict__ out) {  // [maxC,B]
+  int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (b >= B) return;
+  uint64_t h = parents[b];
+  const int my_chunks = n_chunks[b];
+  uint32_t tok[BS];
+#pragma unroll
+  for (int j = 0; j < BS; ++j)  // prefetch chunk 0
+    tok[j] = (uint32_t)tokens_t[(int64_t)j * B + b];
+  for (int c = 0; c < max_chunks; ++c) {
+    if (c >= my_chunks) break;
+    uint32_t nxt[BS];
+    if (c + 1 < max_chunks) {  // prefetch chunk c+1 under the ALU chain
+#pragma unroll
+      for (int j = 0; j < BS; ++j)
+        nxt[j] = (uint32_t)tokens_t[(int64_t)((c + 1) * BS + j) * B + b];
+    }
+    h = chunk_hash(h, tok, BS);
+    out[(int64_t)c * B + b] = h;
+#pragma unroll
+    for (int j = 0; j < BS; ++j) tok[j] = nxt[j];
+  }
+}
+
 // Apply a batch of KV events fully on-device: one WAVE per pod-group
 // (events of one pod processed serially -> per-pod ordering preserved,
 // kvevents/pool.go:132-144); within a BlockStored event lane 0 streams the
@@ -698,6 +736,38 @@ std::vector<at::Tensor> gpu_hash_chain(at::Tensor tokens, at::Tensor tok_off,
                      (int)block_size,
                      reinterpret_cast<uint64_t*>(out.data_ptr<int64_t>()));
   return {out, chunk_off};
+}
+
+// tokens_t: int32 [T, B] (transposed); parents int64 [B]; n_chunks
+// int32 [B].  Returns out int64 [max_chunks, B] (transposed hashes).
+at::Tensor gpu_hash_chain_tr(at::Tensor tokens_t, at::Tensor parents,
+                             at::Tensor n_chunks, int64_t block_size,
+                             int64_t max_chunks) {
+  TORCH_CHECK(tokens_t.is_cuda() && tokens_t.dtype() == at::kInt);
+  TORCH_CHECK(tokens_t.dim() == 2, "tokens_t must be [T, B]");
+  int64_t B = tokens_t.size(1);
+  TORCH_CHECK(parents.numel() == B && n_chunks.numel() == B);
+  auto out = at::zeros({max_chunks, B}, parents.options());
+  if (B == 0 || max_chunks == 0) return out;
+  int threads = 256;
+  int blocks = (int)((B + threads - 1) / threads);
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(blocks), dim3(threads), 0, STREAM,
+                       tokens_t.data_ptr<int32_t>(), U64P(parents),
+                       n_chunks.data_ptr<int32_t>(), B, (int)max_chunks,
+                       reinterpret_cast<uint64_t*>(out.data_ptr<int64_t>()));
+  };
+  switch (block_size) {
+    case 16: launch(k_hash_chain_tr<16>); break;
+    case 32: launch(k_hash_chain_tr<32>); break;
+    case 64: launch(k_hash_chain_tr<64>); break;
+    case 4:  launch(k_hash_chain_tr<4>);  break;
+    case 8:  launch(k_hash_chain_tr<8>);  break;
+    default:
+      TORCH_CHECK(false, "hash_chain_tr supports block sizes 4/8/16/32/64; "
+                         "use gpu_hash_chain for other sizes");
+  }
+  return out;
 }
 
 void gpu_apply_events(at::Tensor keys, at::Tensor meta, at::Tensor stamp,

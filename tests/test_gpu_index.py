@@ -163,6 +163,42 @@ class TestHashChainKernel:
                 assert _to_u64(out[chunk_off[b] + c]) == h, (b, c)
 
 
+class TestHashChainTrKernel:
+    @pytest.mark.parametrize("block_size", [4, 16])
+    def test_matches_row_major_kernel(self, block_size):
+        from llmd_kvcache_amd.ops import cpu_ext
+
+        mod = cpu_ext.require()
+        rng = random.Random(13)
+        B, max_chunks = 130, 9
+        n_chunks = [rng.randrange(1, max_chunks + 1) for _ in range(B)]
+        tok = torch.zeros((max_chunks * block_size, B), dtype=torch.int32)
+        row_toks, row_off = [], [0]
+        for b in range(B):
+            n = n_chunks[b] * block_size
+            vals = [rng.randrange(0, 1 << 31) for _ in range(n)]
+            tok[:n, b] = torch.tensor(vals, dtype=torch.int32)
+            row_toks.extend(vals)
+            row_off.append(len(row_toks))
+        parents = torch.tensor(
+            [_to_i64(hashing.init_hash(""))] * B, dtype=torch.int64,
+            device="cuda")
+        out_t = mod.gpu_hash_chain_tr(
+            tok.cuda(), parents,
+            torch.tensor(n_chunks, dtype=torch.int32, device="cuda"),
+            block_size, max_chunks)
+        ref, ref_off = mod.gpu_hash_chain(
+            torch.tensor(row_toks, dtype=torch.int64, device="cuda"),
+            torch.tensor(row_off, dtype=torch.int64, device="cuda"),
+            parents, block_size)
+        out_t = out_t.cpu()
+        ref = ref.cpu()
+        ref_off = ref_off.cpu()
+        for b in range(B):
+            for c in range(n_chunks[b]):
+                assert out_t[c, b] == ref[ref_off[b] + c], (b, c)
+
+
 class TestApplyEventsKernel:
     def test_stored_then_removed_in_order(self):
         gpu = GpuIndex(GpuIndexConfig(capacity=1 << 12, pods_per_key=10))
