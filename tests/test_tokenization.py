@@ -1,0 +1,307 @@
+"""Tokenization subsystem tests: local provider + discovery, composite
+fallback chain, pool sync/async flow with the 0.8 overlap threshold, and
+the UDS sidecar service over a real unix socket
+(mirrors pkg/tokenization/pool_test.go and tokenizer.go behaviors)."""
+
+import asyncio
+import json
+import os
+import threading
+import time
+
+import pytest
+
+from llmd_kvcache_amd.tokenization.pool import (
+    TokenizationConfig,
+    TokenizationPool,
+)
+from llmd_kvcache_amd.tokenization.prefixstore import (
+    LRUStoreConfig,
+    LRUTokenStore,
+)
+from llmd_kvcache_amd.tokenization.tokenizer import (
+    CompositeTokenizer,
+    LocalTokenizerConfig,
+    TokenizationError,
+    Tokenizer,
+    discover_local_tokenizers,
+    new_cached_local_tokenizer,
+)
+
+
+@pytest.fixture(scope="module")
+def tokenizer_fixture_dir(tmp_path_factory):
+    """Builds a real (tiny) HF tokenizers WordLevel tokenizer.json."""
+    import tokenizers
+    from tokenizers import models, pre_tokenizers
+
+    words = ["hello", "world", "foo", "bar", "baz", "the", "quick", "brown"]
+    vocab = {w: i for i, w in enumerate(words)}
+    vocab["[UNK]"] = len(vocab)
+    tok = tokenizers.Tokenizer(models.WordLevel(vocab, unk_token="[UNK]"))
+    tok.pre_tokenizer = pre_tokenizers.Whitespace()
+
+    root = tmp_path_factory.mktemp("tokenizers")
+    plain = root / "test-model"
+    plain.mkdir()
+    tok.save(str(plain / "tokenizer.json"))
+    # HF-cache layout for discovery
+    hf = root / "models--test-org--cached-model" / "snapshots" / "abc123"
+    hf.mkdir(parents=True)
+    tok.save(str(hf / "tokenizer.json"))
+    return str(root)
+
+
+class TestLocalProvider:
+    def test_discovery_plain_and_hf_layout(self, tokenizer_fixture_dir):
+        mapping = discover_local_tokenizers(tokenizer_fixture_dir)
+        assert "test-model" in mapping
+        assert "test-org/cached-model" in mapping
+
+    def test_encode_with_offsets(self, tokenizer_fixture_dir):
+        tok = new_cached_local_tokenizer(
+            LocalTokenizerConfig(auto_discover_dir=tokenizer_fixture_dir)
+        )
+        ids, offsets = tok.encode("hello world foo", "test-model")
+        assert len(ids) == 3
+        assert offsets == [(0, 5), (6, 11), (12, 15)]
+
+    def test_unknown_model_raises(self, tokenizer_fixture_dir):
+        tok = new_cached_local_tokenizer(
+            LocalTokenizerConfig(auto_discover_dir=tokenizer_fixture_dir)
+        )
+        with pytest.raises(TokenizationError):
+            tok.encode("x", "nope/never")
+
+    def test_cache_returns_same_instance(self, tokenizer_fixture_dir):
+        tok = new_cached_local_tokenizer(
+            LocalTokenizerConfig(auto_discover_dir=tokenizer_fixture_dir)
+        )
+        a = tok._get("test-model")
+        b = tok._get("test-model")
+        assert a is b
+
+
+class FailingTokenizer(Tokenizer):
+    @property
+    def type(self):
+        return "failing"
+
+    def encode(self, prompt, model_name, add_special_tokens=True):
+        raise TokenizationError("backend down")
+
+    def render_chat_template(self, req):
+        raise TokenizationError("backend down")
+
+
+class StaticTokenizer(Tokenizer):
+    @property
+    def type(self):
+        return "static"
+
+    def encode(self, prompt, model_name, add_special_tokens=True):
+        n = len(prompt) // 4
+        return [7] * n, [(i * 4, (i + 1) * 4) for i in range(n)]
+
+    def render_chat_template(self, req):
+        return "rendered!"
+
+
+class TestComposite:
+    def test_fallback_chain(self):
+        comp = CompositeTokenizer([FailingTokenizer(), StaticTokenizer()])
+        ids, offsets = comp.encode("x" * 16, "m")
+        assert ids == [7, 7, 7, 7]
+
+    def test_all_fail_raises_with_accumulated_errors(self):
+        comp = CompositeTokenizer([FailingTokenizer(), FailingTokenizer()])
+        with pytest.raises(TokenizationError, match="failing"):
+            comp.encode("x", "m")
+
+    def test_type_name(self):
+        comp = CompositeTokenizer([FailingTokenizer(), StaticTokenizer()])
+        assert comp.type == "composite(failing,static)"
+
+
+class CountingTokenizer(StaticTokenizer):
+    def __init__(self):
+        self.encode_calls = 0
+
+    def encode(self, prompt, model_name, add_special_tokens=True):
+        self.encode_calls += 1
+        return super().encode(prompt, model_name, add_special_tokens)
+
+
+class TestPool:
+    def test_sync_tokenize_and_prefix_cache_skip(self):
+        tok = CountingTokenizer()
+        store = LRUTokenStore(LRUStoreConfig(cache_size=1000, block_size=16))
+        pool = TokenizationPool(
+            TokenizationConfig(workers_count=2), indexer=store, tokenizer=tok
+        )
+        pool.run()
+        try:
+            prompt = "abcd" * 16  # 64 chars -> 16 tokens
+            t1 = pool.tokenize(None, prompt, "m")
+            assert len(t1) == 16
+            assert tok.encode_calls == 1
+            # same prompt again: full coverage -> cached, no re-encode
+            t2 = pool.tokenize(None, prompt, "m")
+            assert t2 == t1
+            assert tok.encode_calls == 1
+            # sufficiently-extended prompt: coverage drops below 0.8 ->
+            # full re-encode
+            pool.tokenize(None, prompt + "zzzz" * 8, "m")
+            assert tok.encode_calls == 2
+        finally:
+            pool.shutdown()
+
+    def test_async_enqueue(self):
+        tok = CountingTokenizer()
+        store = LRUTokenStore(LRUStoreConfig(cache_size=1000, block_size=16))
+        pool = TokenizationPool(
+            TokenizationConfig(workers_count=2), indexer=store, tokenizer=tok
+        )
+        pool.run()
+        try:
+            pool.enqueue_tokenization(None, "abcd" * 16, "m")
+            deadline = time.monotonic() + 5
+            while tok.encode_calls == 0 and time.monotonic() < deadline:
+                time.sleep(0.01)
+            assert tok.encode_calls == 1
+        finally:
+            pool.shutdown()
+
+    def test_render_req_renders_first(self):
+        class RenderCheck(StaticTokenizer):
+            def render_chat_template(self, req):
+                return "abcd" * 8
+
+        pool = TokenizationPool(
+            TokenizationConfig(workers_count=1), tokenizer=RenderCheck()
+        )
+        tokens = pool.tokenize(object(), "ignored", "m")
+        assert len(tokens) == 8
+
+    def test_error_propagates_to_caller(self):
+        pool = TokenizationPool(
+            TokenizationConfig(workers_count=1), tokenizer=FailingTokenizer()
+        )
+        pool.run()
+        try:
+            with pytest.raises(TokenizationError):
+                pool.tokenize(None, "x" * 100, "m")
+        finally:
+            pool.shutdown()
+
+
+@pytest.fixture
+def uds_server(tokenizer_fixture_dir, tmp_path):
+    """Runs the real aiohttp UDS sidecar in a background thread."""
+    import sys
+
+    sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+    from services.uds_tokenizer import server as uds
+
+    sock_path = str(tmp_path / "tok.socket")
+    uds.core.reload({"local_dir": tokenizer_fixture_dir})
+    app = uds.make_app()
+
+    loop = asyncio.new_event_loop()
+    started = threading.Event()
+    runner_box = {}
+
+    def run():
+        asyncio.set_event_loop(loop)
+
+        async def start():
+            runner = web_runner = None
+            from aiohttp import web
+
+            runner = web.AppRunner(app)
+            await runner.setup()
+            site = web.UnixSite(runner, sock_path)
+            await site.start()
+            runner_box["runner"] = runner
+            started.set()
+
+        loop.run_until_complete(start())
+        loop.run_forever()
+
+    t = threading.Thread(target=run, daemon=True)
+    t.start()
+    assert started.wait(10)
+    yield sock_path
+    loop.call_soon_threadsafe(loop.stop)
+    t.join(timeout=5)
+
+
+class TestUdsService:
+    def test_tokenize_round_trip(self, uds_server):
+        from llmd_kvcache_amd.tokenization.uds import (
+            UdsTokenizer,
+            UdsTokenizerConfig,
+        )
+
+        client = UdsTokenizer(UdsTokenizerConfig(socket_path=uds_server))
+        ids, offsets = client.encode("hello world", "test-model")
+        assert len(ids) == 2
+        assert offsets == [(0, 5), (6, 11)]
+
+    def test_health_and_config(self, uds_server):
+        import http.client
+        import socket as socketlib
+
+        class Conn(http.client.HTTPConnection):
+            def __init__(self):
+                super().__init__("localhost", timeout=5)
+
+            def connect(self):
+                s = socketlib.socket(socketlib.AF_UNIX, socketlib.SOCK_STREAM)
+                s.connect(uds_server)
+                self.sock = s
+
+        conn = Conn()
+        conn.request("GET", "/health")
+        assert json.loads(conn.getresponse().read())["status"] == "ok"
+        conn = Conn()
+        conn.request("GET", "/config")
+        cfg = json.loads(conn.getresponse().read())
+        assert "local_dir" in cfg
+        # hot reload
+        conn = Conn()
+        body = json.dumps({"add_special_tokens": False}).encode()
+        conn.request("POST", "/config", body=body,
+                     headers={"Content-Type": "application/json"})
+        resp = json.loads(conn.getresponse().read())
+        assert resp["config"]["add_special_tokens"] is False
+
+    def test_bad_request_400(self, uds_server):
+        from llmd_kvcache_amd.tokenization.uds import (
+            UdsTokenizer,
+            UdsTokenizerConfig,
+        )
+
+        client = UdsTokenizer(
+            UdsTokenizerConfig(socket_path=uds_server, max_retries=0)
+        )
+        with pytest.raises(TokenizationError):
+            client._post("/tokenize", {"nope": 1})
+
+    def test_retry_on_dead_socket(self, tmp_path):
+        from llmd_kvcache_amd.tokenization.uds import (
+            UdsTokenizer,
+            UdsTokenizerConfig,
+        )
+
+        client = UdsTokenizer(
+            UdsTokenizerConfig(
+                socket_path=str(tmp_path / "missing.socket"),
+                max_retries=1,
+                backoff_base_s=0.01,
+            )
+        )
+        t0 = time.monotonic()
+        with pytest.raises(TokenizationError):
+            client.encode("x", "m")
+        assert time.monotonic() - t0 >= 0.01  # backoff happened
