@@ -95,12 +95,42 @@ class Indexer:
         )
         if not block_keys:
             return {}
+        return self._score_keys(block_keys, pod_identifiers)
 
-        # 3. index lookup (empty filter set = all pods)
+    def _score_keys(
+        self, block_keys: List[Key], pod_identifiers: Sequence[str]
+    ) -> Dict[str, float]:
+        # Fast path: table-backed indexes (CPU C++ or gfx950 HIP) fuse
+        # lookup + longest-prefix scoring into one native call/kernel.
+        fused = getattr(self._kv_block_index, "fused_scores", None)
+        if fused is not None:
+            import torch
+
+            from .kvblock.gpu_index import _to_i64
+
+            idx = self._kv_block_index
+            device = idx.device
+            hashes = torch.tensor(
+                [_to_i64(k.chunk_hash) for k in block_keys],
+                dtype=torch.int64, device=device,
+            )
+            if idx.table.is_cuda:
+                counts = torch.tensor([0, len(block_keys)],
+                                      dtype=torch.int32, device=device)
+            else:
+                counts = torch.tensor([len(block_keys)], dtype=torch.int32)
+            weights = idx.tier_weights(
+                {b.name: b.weight for b in self.config.backend_configs}
+            )
+            scores = fused(hashes, counts, block_keys[0].model_name,
+                           set(pod_identifiers), weights,
+                           max_k=len(block_keys))
+            return idx.scores_to_map(scores)[0]
+
+        # 3. generic index lookup (empty filter set = all pods)
         key_to_pods = self._kv_block_index.lookup(
             block_keys, set(pod_identifiers)
         )
-
         # 4. score
         return self.kv_block_scorer.score(block_keys, key_to_pods)
 
@@ -117,7 +147,4 @@ class Indexer:
         )
         if not block_keys:
             return {}
-        key_to_pods = self._kv_block_index.lookup(
-            block_keys, set(pod_identifiers)
-        )
-        return self.kv_block_scorer.score(block_keys, key_to_pods)
+        return self._score_keys(block_keys, pod_identifiers)

@@ -12,6 +12,8 @@ std::vector<uint64_t> tokens_to_chunk_hashes(std::vector<uint64_t> tokens,
                                              int64_t block_size);
 std::vector<at::Tensor> hash_chain_batch(at::Tensor, at::Tensor, at::Tensor,
                                          int64_t);
+std::vector<uint64_t> tokens_to_chunk_hashes_fast(std::vector<uint64_t>,
+                                                  uint64_t, int64_t);
 void cpu_insert(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                 at::Tensor, at::Tensor, int64_t, at::Tensor, at::Tensor,
                 int64_t, at::Tensor, int64_t, int64_t, int64_t);
@@ -74,6 +76,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native KV-block index ops (CPU reference + gfx950 HIP)";
   m.def("tokens_to_chunk_hashes", &kvidx::tokens_to_chunk_hashes);
   m.def("hash_chain_batch", &kvidx::hash_chain_batch);
+  m.def("tokens_to_chunk_hashes_fast", &kvidx::tokens_to_chunk_hashes_fast);
   m.def("cpu_insert", &kvidx::cpu_insert);
   m.def("cpu_evict", &kvidx::cpu_evict);
   m.def("cpu_lookup", &kvidx::cpu_lookup);

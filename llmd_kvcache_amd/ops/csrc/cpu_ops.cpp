@@ -201,6 +201,23 @@ std::vector<uint64_t> tokens_to_chunk_hashes(std::vector<uint64_t> tokens,
   return out;
 }
 
+// Branchless-variant twin of tokens_to_chunk_hashes - exists so tests
+// can verify chunk_hash_fast (the GPU hot path) == chunk_hash on every
+// CBOR length-boundary case without a GPU.
+std::vector<uint64_t> tokens_to_chunk_hashes_fast(
+    std::vector<uint64_t> tokens, uint64_t parent, int64_t block_size) {
+  std::vector<uint32_t> toks(tokens.begin(), tokens.end());
+  int64_t n_chunks = (int64_t)toks.size() / block_size;
+  std::vector<uint64_t> out;
+  out.reserve(n_chunks);
+  uint64_t h = parent;
+  for (int64_t c = 0; c < n_chunks; ++c) {
+    h = chunk_hash_fast(h, toks.data() + c * block_size, (int)block_size);
+    out.push_back(h);
+  }
+  return out;
+}
+
 // Batched chain hashing: tokens (int64 [total]), offsets (int64 [B+1]),
 // parents (int64 [B], u64 bits) -> per-prompt chunk hashes, flat, with
 // chunk-count offsets returned alongside.

@@ -440,32 +440,50 @@ __global__ void k_hash_chain(const int64_t* __restrict__ tokens,
 // strided per-lane loads; this layout removes that.
 // out is [max_chunks][B] (transposed too; callers transpose back with a
 // cheap torch op or consume the stride directly).
-template <int BS>  // compile-time block size keeps tok[] in registers
+// ILP chains: each lane advances ILP independent prompt chains in one
+// straight-line (branchless chunk_hash_fast) loop body, so the scheduler
+// interleaves their serial FNV dependency chains - the chain is
+// dependent-ALU-latency-bound at 1 wave/SIMD, and ILP-x recovers ~x of
+// that latency.  Lane l of the grid owns prompts {l, l+L, .., l+(ILP-1)L}
+// where L = total lanes, keeping every token load coalesced.
+template <int BS, int ILP>
 __global__ void k_hash_chain_tr(const int32_t* __restrict__ tokens_t,  // [T,B]
                                 const uint64_t* __restrict__ parents,  // [B]
                                 const int32_t* __restrict__ n_chunks,  // [B]
-                                int64_t B, int max_chunks,
+                                int64_t B, int64_t L, int max_chunks,
                                 uint64_t* __restrict__ out) {  // [maxC,B]
-  int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (b >= B) return;
-  uint64_t h = parents[b];
-  const int my_chunks = n_chunks[b];
-  uint32_t tok[BS];
+  const int64_t lane = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (lane >= L) return;
+  int64_t b[ILP];
+  uint64_t h[ILP];
+  int mc[ILP];
+  bool live[ILP];
 #pragma unroll
-  for (int j = 0; j < BS; ++j)  // prefetch chunk 0
-    tok[j] = (uint32_t)tokens_t[(int64_t)j * B + b];
+  for (int i = 0; i < ILP; ++i) {
+    b[i] = lane + (int64_t)i * L;
+    live[i] = b[i] < B;
+    h[i] = live[i] ? parents[b[i]] : 0;
+    mc[i] = live[i] ? n_chunks[b[i]] : 0;
+  }
   for (int c = 0; c < max_chunks; ++c) {
-    if (c >= my_chunks) break;
-    uint32_t nxt[BS];
-    if (c + 1 < max_chunks) {  // prefetch chunk c+1 under the ALU chain
+    uint32_t tok[ILP][BS];
+    // loads and hash bodies are UNCONDITIONAL (clamped addresses, dead
+    // chains compute garbage that a select discards): keeps the whole
+    // chunk step one basic block so the ILP chains actually interleave.
+#pragma unroll
+    for (int i = 0; i < ILP; ++i) {
+      const int64_t bi = live[i] ? b[i] : 0;
 #pragma unroll
       for (int j = 0; j < BS; ++j)
-        nxt[j] = (uint32_t)tokens_t[(int64_t)((c + 1) * BS + j) * B + b];
+        tok[i][j] = (uint32_t)tokens_t[(int64_t)(c * BS + j) * B + bi];
     }
-    h = chunk_hash(h, tok, BS);
-    out[(int64_t)c * B + b] = h;
 #pragma unroll
-    for (int j = 0; j < BS; ++j) tok[j] = nxt[j];
+    for (int i = 0; i < ILP; ++i) {
+      const bool active = live[i] && c < mc[i];
+      const uint64_t h2 = chunk_hash_fast(h[i], tok[i], BS);
+      h[i] = active ? h2 : h[i];
+      if (active) out[(int64_t)c * B + b[i]] = h[i];
+    }
   }
 }
 
@@ -749,20 +767,22 @@ at::Tensor gpu_hash_chain_tr(at::Tensor tokens_t, at::Tensor parents,
   TORCH_CHECK(parents.numel() == B && n_chunks.numel() == B);
   auto out = at::zeros({max_chunks, B}, parents.options());
   if (B == 0 || max_chunks == 0) return out;
+  constexpr int ILP = 4;
+  int64_t L = (B + ILP - 1) / ILP;  // lanes
   int threads = 256;
-  int blocks = (int)((B + threads - 1) / threads);
+  int blocks = (int)((L + threads - 1) / threads);
   auto launch = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(blocks), dim3(threads), 0, STREAM,
                        tokens_t.data_ptr<int32_t>(), U64P(parents),
-                       n_chunks.data_ptr<int32_t>(), B, (int)max_chunks,
+                       n_chunks.data_ptr<int32_t>(), B, L, (int)max_chunks,
                        reinterpret_cast<uint64_t*>(out.data_ptr<int64_t>()));
   };
   switch (block_size) {
-    case 16: launch(k_hash_chain_tr<16>); break;
-    case 32: launch(k_hash_chain_tr<32>); break;
-    case 64: launch(k_hash_chain_tr<64>); break;
-    case 4:  launch(k_hash_chain_tr<4>);  break;
-    case 8:  launch(k_hash_chain_tr<8>);  break;
+    case 16: launch(k_hash_chain_tr<16, ILP>); break;
+    case 32: launch(k_hash_chain_tr<32, ILP>); break;
+    case 64: launch(k_hash_chain_tr<64, 2>); break;
+    case 4:  launch(k_hash_chain_tr<4, ILP>);  break;
+    case 8:  launch(k_hash_chain_tr<8, ILP>);  break;
     default:
       TORCH_CHECK(false, "hash_chain_tr supports block sizes 4/8/16/32/64; "
                          "use gpu_hash_chain for other sizes");

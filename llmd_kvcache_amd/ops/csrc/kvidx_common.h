@@ -99,6 +99,63 @@ KVIDX_HD uint64_t fnv_cbor_uint(uint64_t h, uint64_t v, uint8_t major) {
   return h;
 }
 
+// Branchless (predicated) variants: identical output to fnv_cbor_uint but
+// straight-line code - on the GPU this lets the scheduler interleave
+// several independent chains' instruction streams (ILP hash kernels) and
+// removes lane divergence on data-dependent encoding lengths.
+KVIDX_HD uint64_t fnv_cbor_u32_branchless(uint64_t h, uint32_t v,
+                                          uint8_t major) {
+  const uint8_t mt = major << 5;
+  const int sel = (v >= 24) + (v > 0xFF) + (v > 0xFFFF);  // 0..3
+  const int n = sel == 3 ? 4 : sel;  // value bytes
+  const uint8_t header = sel ? (uint8_t)(mt | (0x17 + sel))
+                             : (uint8_t)(mt | v);
+  h = fnv1a_64_byte(h, header);
+#if defined(__HIP_DEVICE_COMPILE__)
+#pragma unroll
+#endif
+  for (int k = 0; k < 4; ++k) {
+    const uint8_t b = (uint8_t)(v >> (8 * (3 - k)));
+    const uint64_t h2 = fnv1a_64_byte(h, b);
+    h = (k >= 4 - n) ? h2 : h;
+  }
+  return h;
+}
+
+KVIDX_HD uint64_t fnv_cbor_u64_branchless(uint64_t h, uint64_t v,
+                                          uint8_t major) {
+  const uint8_t mt = major << 5;
+  const int sel = (v >= 24) + (v > 0xFF) + (v > 0xFFFF) +
+                  (v > 0xFFFFFFFFull);  // 0..4
+  const int n = sel == 0 ? 0 : (1 << (sel - 1));  // 0,1,2,4,8 value bytes
+  const uint8_t header = sel ? (uint8_t)(mt | (0x17 + sel))
+                             : (uint8_t)(mt | v);
+  h = fnv1a_64_byte(h, header);
+#if defined(__HIP_DEVICE_COMPILE__)
+#pragma unroll
+#endif
+  for (int k = 0; k < 8; ++k) {
+    const uint8_t b = (uint8_t)(v >> (8 * (7 - k)));
+    const uint64_t h2 = fnv1a_64_byte(h, b);
+    h = (k >= 8 - n) ? h2 : h;
+  }
+  return h;
+}
+
+// Branchless chain link (bit-identical to chunk_hash; GPU hot path).
+template <typename TokT>
+KVIDX_HD uint64_t chunk_hash_fast(uint64_t parent, const TokT* tokens,
+                                  int n) {
+  uint64_t h = FNV64_OFFSET;
+  h = fnv1a_64_byte(h, 0x83);
+  h = fnv_cbor_u64_branchless(h, parent, 0);
+  h = fnv_cbor_u32_branchless(h, (uint32_t)n, 4);
+  for (int i = 0; i < n; ++i)
+    h = fnv_cbor_u32_branchless(h, (uint32_t)tokens[i], 0);
+  h = fnv1a_64_byte(h, 0xF6);
+  return h;
+}
+
 // One chain link: FNV-64a(CBOR([parent, tokens[0..n), null])), streamed.
 template <typename TokT>
 KVIDX_HD uint64_t chunk_hash(uint64_t parent, const TokT* tokens, int n) {

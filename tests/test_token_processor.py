@@ -96,3 +96,45 @@ class TestNativeParity:
             py.append(h)
         native = mod.tokens_to_chunk_hashes(tokens, hashing.init_hash(""), 16)
         assert list(native) == py
+
+
+class TestBranchlessParity:
+    """chunk_hash_fast (the GPU hot path) must be bit-identical to the
+    branchy reference on every CBOR shortest-form boundary."""
+
+    def test_boundary_values(self):
+        from llmd_kvcache_amd.ops import cpu_ext
+
+        mod = cpu_ext.maybe_load()
+        if mod is None:
+            import pytest
+
+            pytest.skip("native extension not built")
+        boundary_tokens = [0, 1, 23, 24, 255, 256, 65535, 65536,
+                           2**31 - 1, 2**32 - 1]
+        # parents spanning every u64 encoding length
+        parents = [0, 5, 23, 24, 200, 256, 65535, 70000, 2**32 - 1, 2**32,
+                   2**63, 2**64 - 1, hashing.init_hash("")]
+        import itertools
+
+        for parent in parents:
+            toks = boundary_tokens[:8]
+            a = mod.tokens_to_chunk_hashes(toks, parent, 4)
+            b = mod.tokens_to_chunk_hashes_fast(toks, parent, 4)
+            assert list(a) == list(b), parent
+
+    def test_random_parity(self):
+        from llmd_kvcache_amd.ops import cpu_ext
+
+        mod = cpu_ext.maybe_load()
+        if mod is None:
+            import pytest
+
+            pytest.skip("native extension not built")
+        import random
+
+        rng = random.Random(3)
+        toks = [rng.randrange(0, 2**32) for _ in range(320)]
+        a = mod.tokens_to_chunk_hashes(toks, hashing.init_hash(""), 16)
+        b = mod.tokens_to_chunk_hashes_fast(toks, hashing.init_hash(""), 16)
+        assert list(a) == list(b)
