@@ -760,15 +760,20 @@ std::vector<at::Tensor> gpu_hash_chain(at::Tensor tokens, at::Tensor tok_off,
 // int32 [B].  Returns out int64 [max_chunks, B] (transposed hashes).
 at::Tensor gpu_hash_chain_tr(at::Tensor tokens_t, at::Tensor parents,
                              at::Tensor n_chunks, int64_t block_size,
-                             int64_t max_chunks) {
+                             int64_t max_chunks, int64_t ilp) {
   TORCH_CHECK(tokens_t.is_cuda() && tokens_t.dtype() == at::kInt);
   TORCH_CHECK(tokens_t.dim() == 2, "tokens_t must be [T, B]");
   int64_t B = tokens_t.size(1);
   TORCH_CHECK(parents.numel() == B && n_chunks.numel() == B);
   auto out = at::zeros({max_chunks, B}, parents.options());
   if (B == 0 || max_chunks == 0) return out;
-  constexpr int ILP = 4;
-  int64_t L = (B + ILP - 1) / ILP;  // lanes
+  // ILP trades wave count for per-lane chain interleave: only pay the
+  // wave reduction when there are waves to spare (tuned by
+  // scripts/sweep_chain.py on MI355X).
+  int64_t want_ilp = ilp;
+  if (want_ilp <= 0)
+    want_ilp = B >= 16384 ? 8 : B >= 4096 ? 4 : B >= 1024 ? 2 : 1;
+  int64_t L = (B + want_ilp - 1) / want_ilp;  // lanes
   int threads = 256;
   int blocks = (int)((L + threads - 1) / threads);
   auto launch = [&](auto kern) {
@@ -777,15 +782,29 @@ at::Tensor gpu_hash_chain_tr(at::Tensor tokens_t, at::Tensor parents,
                        n_chunks.data_ptr<int32_t>(), B, L, (int)max_chunks,
                        reinterpret_cast<uint64_t*>(out.data_ptr<int64_t>()));
   };
-  switch (block_size) {
-    case 16: launch(k_hash_chain_tr<16, ILP>); break;
-    case 32: launch(k_hash_chain_tr<32, ILP>); break;
-    case 64: launch(k_hash_chain_tr<64, 2>); break;
-    case 4:  launch(k_hash_chain_tr<4, ILP>);  break;
-    case 8:  launch(k_hash_chain_tr<8, ILP>);  break;
-    default:
-      TORCH_CHECK(false, "hash_chain_tr supports block sizes 4/8/16/32/64; "
-                         "use gpu_hash_chain for other sizes");
+  bool done = false;
+  if (block_size == 16) {
+    done = true;
+    switch (want_ilp) {
+      case 1: launch(k_hash_chain_tr<16, 1>); break;
+      case 2: launch(k_hash_chain_tr<16, 2>); break;
+      case 8: launch(k_hash_chain_tr<16, 8>); break;
+      default: launch(k_hash_chain_tr<16, 4>); break;
+    }
+  }
+  if (!done) {
+    int fixed_ilp = block_size == 64 ? 2 : 4;
+    L = (B + fixed_ilp - 1) / fixed_ilp;
+    blocks = (int)((L + threads - 1) / threads);
+    switch (block_size) {
+      case 32: launch(k_hash_chain_tr<32, 4>); break;
+      case 64: launch(k_hash_chain_tr<64, 2>); break;
+      case 4:  launch(k_hash_chain_tr<4, 4>);  break;
+      case 8:  launch(k_hash_chain_tr<8, 4>);  break;
+      default:
+        TORCH_CHECK(false, "hash_chain_tr supports block sizes 4/8/16/32/64;"
+                           " use gpu_hash_chain for other sizes");
+    }
   }
   return out;
 }
