@@ -146,9 +146,9 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--batch", type=int, default=512,
+    ap.add_argument("--batch", type=int, default=2048,
                     help="prompts per scoring call (sub-batch)")
-    ap.add_argument("--calls-per-step", type=int, default=4)
+    ap.add_argument("--calls-per-step", type=int, default=2)
     ap.add_argument("--device", default=None, help="cpu to force CPU tables")
     ap.add_argument("--blocks", type=int, default=NUM_BLOCKS)
     args = ap.parse_args()

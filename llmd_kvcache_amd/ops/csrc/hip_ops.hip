@@ -771,8 +771,10 @@ at::Tensor gpu_hash_chain_tr(at::Tensor tokens_t, at::Tensor parents,
   // wave reduction when there are waves to spare (tuned by
   // scripts/sweep_chain.py on MI355X).
   int64_t want_ilp = ilp;
-  if (want_ilp <= 0)
-    want_ilp = B >= 16384 ? 8 : B >= 4096 ? 4 : B >= 1024 ? 2 : 1;
+  // Sweep on MI355X (profiles/r01_chain_sweep.md): ILP>1 always loses -
+  // register pressure + extra loads outweigh the interleave; the
+  // branchless body alone keeps the SIMD pipeline fed.  Default 1.
+  if (want_ilp <= 0) want_ilp = 1;
   int64_t L = (B + want_ilp - 1) / want_ilp;  // lanes
   int threads = 256;
   int blocks = (int)((L + threads - 1) / threads);
