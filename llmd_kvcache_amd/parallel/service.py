@@ -1,0 +1,130 @@
+"""Multi-rank service driver: one ingest/API rank, N table shards.
+
+Completes BASELINE.json config 3 as a *service*: rank 0 owns the external
+surfaces (ZMQ KVEvents subscriber, gRPC/HTTP scoring) and replicates work
+to the other ranks of the node with torch.distributed object broadcasts;
+every rank holds one shard of the table (parallel/sharded.py) and
+participates in the mask-merge all_reduce.
+
+Why broadcast instead of per-rank ZMQ subscribers: vLLM publishers
+connect to ONE endpoint (the SUB binds, zmq_subscriber.go:90), so a
+single rank must own the socket; event decode is cheap relative to the
+collective, so raw payload bytes are shipped and decoded per rank.
+
+Protocol (collective ops driven by rank 0):
+  ("events", [(pod, model, payload_bytes), ...])  apply on every rank
+  ("score", hashes_i64_list, offsets, model, pods) score collectively
+  ("stop",)                                        exit serve loops
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Sequence, Set, Tuple
+
+import torch
+import torch.distributed as dist
+
+from ..kvblock.gpu_index import _to_i64
+from ..kvblock.token_processor import ChunkedTokenDatabase
+from ..kvevents.events import decode_event_batch
+from .sharded import ShardedIndex
+
+
+class ShardedIndexService:
+    def __init__(
+        self,
+        sharded: ShardedIndex,
+        token_processor: Optional[ChunkedTokenDatabase] = None,
+    ):
+        self.sharded = sharded
+        self.token_processor = token_processor or ChunkedTokenDatabase()
+        self.rank = sharded.rank
+        self.group = sharded.group
+        self._running = True
+
+    # -- collective plumbing -------------------------------------------
+    def _broadcast(self, op):
+        buf = [op]
+        dist.broadcast_object_list(buf, src=0, group=self.group)
+        return buf[0]
+
+    def _dispatch(self, op) -> Optional[object]:
+        kind = op[0]
+        if kind == "events":
+            batches = []
+            for pod, model, payload in op[1]:
+                try:
+                    batch = decode_event_batch(payload)
+                except Exception:
+                    continue
+                batches.append((pod, model, batch.events))
+            self._apply(batches)
+            return None
+        if kind == "score":
+            _, hashes, offsets, model, pods = op
+            device = self.sharded.device
+            h = torch.tensor(hashes, dtype=torch.int64, device=device)
+            offs = torch.tensor(offsets, dtype=torch.int32)
+            scores = self.sharded.sharded_scores(h, offs, model, set(pods))
+            return scores
+        if kind == "stop":
+            self._running = False
+            return None
+        raise ValueError(f"unknown op {kind!r}")
+
+    def _apply(self, batches: List[Tuple[str, str, list]]) -> None:
+        local = self.sharded.local
+        if local.table.is_cuda:
+            from ..kvblock.gpu_index import GpuIndex
+
+            GpuIndex.apply_event_batches.__get__(local)(
+                batches, self.token_processor
+            )
+        else:
+            from ..kvevents.pool import EventsConfig, EventsPool
+
+            pool = EventsPool(EventsConfig(concurrency=1), local,
+                              self.token_processor)
+            for pod, model, events in batches:
+                pool.digest_events(pod, model, events)
+
+    # -- rank-0 API ----------------------------------------------------
+    def apply_messages(
+        self, messages: Sequence[Tuple[str, str, bytes]]
+    ) -> None:
+        """(pod_identifier, model_name, raw msgpack payload) triples -
+        exactly what the ZMQ subscriber hands the events pool."""
+        assert self.rank == 0
+        op = ("events", list(messages))
+        self._broadcast(op)
+        self._dispatch(op)
+
+    def score(
+        self,
+        request_keys: Sequence,
+        pod_identifier_set: Set[str],
+    ) -> Dict[str, float]:
+        assert self.rank == 0
+        if not request_keys:
+            return {}
+        hashes = [_to_i64(k.chunk_hash) for k in request_keys]
+        offsets = [0, len(request_keys)]
+        op = ("score", hashes, offsets, request_keys[0].model_name,
+              sorted(pod_identifier_set))
+        self._broadcast(op)
+        scores = self._dispatch(op)
+        return self.sharded.local.scores_to_map(scores)[0]
+
+    def stop(self) -> None:
+        assert self.rank == 0
+        op = ("stop",)
+        self._broadcast(op)
+        self._dispatch(op)
+
+    # -- non-rank-0 loop -----------------------------------------------
+    def serve(self) -> None:
+        """Follower ranks: execute broadcast ops until stop."""
+        assert self.rank != 0
+        while self._running:
+            op = self._broadcast(None)
+            self._dispatch(op)

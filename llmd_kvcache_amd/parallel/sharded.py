@@ -97,10 +97,9 @@ class ShardedIndex:
         dist.all_reduce(masks, op=dist.ReduceOp.SUM, group=self.group)
 
         offs = offsets.to(dtype=torch.int32, device=self.device)
-        flat = masks.reshape(-1, MAX_TIERS * masks.shape[2])
         if self.local.table.is_cuda:
             return self.local.table.ops.gpu_score_from_masks(
-                flat.contiguous().view(masks.shape), offs, weights, num_pods
+                masks.contiguous(), offs, weights, num_pods
             )
         return self.local.table.ops.cpu_score_from_masks(
             masks.contiguous(), offs.cpu(), weights.cpu(), num_pods

@@ -180,3 +180,51 @@ def test_sharded_batch_scores():
         assert maps[0] == {"pod-a": 3.0}
         assert maps[1] == {"pod-a": 6.0}
         assert maps[2] == {}
+
+
+def _body_service_bridge(rank, world_size):
+    import time as _time
+
+    from llmd_kvcache_amd.kvblock.gpu_index import TableIndexConfig
+    from llmd_kvcache_amd.kvblock.token_processor import (
+        ChunkedTokenDatabase,
+        TokenProcessorConfig,
+    )
+    from llmd_kvcache_amd.kvevents.events import BlockStored, EventBatch
+    from llmd_kvcache_amd.parallel.service import ShardedIndexService
+    from llmd_kvcache_amd.parallel.sharded import ShardedIndex
+
+    tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=4))
+    sharded = ShardedIndex(TableIndexConfig(capacity=1 << 12, pods_per_key=10))
+    svc = ShardedIndexService(sharded, tp)
+
+    if rank == 0:
+        tokens = list(range(32))
+        batch = EventBatch(
+            ts=_time.time(),
+            events=[BlockStored(list(range(100, 108)), None, tokens, 4)],
+        )
+        svc.apply_messages([("vllm-pod-1", MODEL, batch.encode())])
+        keys = tp.tokens_to_kv_block_keys(None, tokens, MODEL)
+        scores = svc.score(keys, set())
+        filtered = svc.score(keys, {"nobody"})
+        svc.stop()
+        return (scores, filtered)
+    svc.serve()
+    # follower rank must own part of the shard
+    from llmd_kvcache_amd.kvblock.keys import Key
+
+    keys = tp.tokens_to_kv_block_keys(None, list(range(32)), MODEL)
+    owned = sum(
+        1 for k in keys
+        if sharded.local.lookup([k, Key(MODEL, 1)], set()).get(k)
+    )
+    return owned
+
+
+def test_sharded_service_bridge():
+    results = run_distributed("_body_service_bridge")
+    scores, filtered = results[0]
+    assert scores == {"vllm-pod-1": 8.0}
+    assert filtered == {}
+    assert results[1] > 0  # rank 1 holds part of the index
