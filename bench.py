@@ -84,10 +84,11 @@ def populate_index(index, device, rank):
     for e in range(n_events):
         tokens = rng.integers(0, 1 << 31, size=BLOCKS_PER_EVENT * BLOCK_SIZE,
                               dtype=np.int64)
-        hashes = list(range(engine_hash, engine_hash + BLOCKS_PER_EVENT))
+        hashes = np.arange(engine_hash, engine_hash + BLOCKS_PER_EVENT,
+                           dtype=np.uint64)
         engine_hash += BLOCKS_PER_EVENT
         pod = pods[e % NUM_PODS]
-        ev = BlockStored(hashes, None, tokens.tolist(), BLOCK_SIZE)
+        ev = BlockStored(hashes, None, tokens, BLOCK_SIZE)
         batch.append((pod, MODEL, [ev]))
         if e < 2048:  # keep a sample of chains for the read workload
             chains.append(tokens)

@@ -410,9 +410,28 @@ class GpuIndex(TableIndex):
         block_size = token_processor.block_size
         init_hash = token_processor.config.init_hash()
 
-        tokens: List[int] = []
+        import numpy as np
+
+        def _hashes_np(raw_list):
+            """Coerce a block-hash list to u64 bits as an int64 ndarray;
+            fast path for plain-int lists/ndarrays, per-element fallback
+            for mixed/bytes forms (pool.go:343-367 coercion)."""
+            try:
+                return np.asarray(raw_list, dtype=np.uint64).view(np.int64)
+            except (TypeError, ValueError, OverflowError):
+                out = []
+                for raw in raw_list:
+                    try:
+                        out.append(get_hash_as_uint64(raw))
+                    except Exception:
+                        continue
+                return np.asarray(out, dtype=np.uint64).view(np.int64)
+
+        token_arrays: List = []
+        n_tokens = 0
         tok_off = [0]
-        ehashes: List[int] = []
+        hash_arrays: List = []
+        n_hashes = 0
         eh_off = [0]
         parents: List[int] = []
         has_parent: List[int] = []
@@ -440,18 +459,16 @@ class GpuIndex(TableIndex):
                 if isinstance(ev, BlockStored):
                     tier = self.registry.tier_id(
                         ev.medium.lower() if ev.medium else "gpu")
-                    hs = []
-                    for raw in ev.block_hashes:
-                        try:
-                            hs.append(get_hash_as_uint64(raw))
-                        except Exception:
-                            continue
-                    if not hs:
+                    hs = _hashes_np(ev.block_hashes)
+                    if hs.size == 0:
                         continue
-                    ehashes.extend(_to_i64(h) for h in hs)
-                    eh_off.append(len(ehashes))
-                    tokens.extend(int(t) for t in ev.token_ids)
-                    tok_off.append(len(tokens))
+                    hash_arrays.append(hs)
+                    n_hashes += hs.size
+                    eh_off.append(n_hashes)
+                    toks = np.asarray(ev.token_ids, dtype=np.int64)
+                    token_arrays.append(toks)
+                    n_tokens += toks.size
+                    tok_off.append(n_tokens)
                     if ev.parent_block_hash is not None:
                         try:
                             parents.append(
@@ -469,17 +486,13 @@ class GpuIndex(TableIndex):
                 elif isinstance(ev, BlockRemoved):
                     tier = self.registry.tier_id(
                         ev.medium.lower() if ev.medium else "gpu")
-                    hs = []
-                    for raw in ev.block_hashes:
-                        try:
-                            hs.append(get_hash_as_uint64(raw))
-                        except Exception:
-                            continue
-                    if not hs:
+                    hs = _hashes_np(ev.block_hashes)
+                    if hs.size == 0:
                         continue
-                    ehashes.extend(_to_i64(h) for h in hs)
-                    eh_off.append(len(ehashes))
-                    tok_off.append(len(tokens))
+                    hash_arrays.append(hs)
+                    n_hashes += hs.size
+                    eh_off.append(n_hashes)
+                    tok_off.append(n_tokens)
                     parents.append(0)
                     has_parent.append(0)
                     ev_type.append(1)
@@ -498,11 +511,15 @@ class GpuIndex(TableIndex):
 
         d = self.device
         i32 = torch.int32
+        tokens_np = (np.concatenate(token_arrays) if token_arrays
+                     else np.zeros(0, dtype=np.int64))
+        hashes_np = (np.concatenate(hash_arrays) if hash_arrays
+                     else np.zeros(0, dtype=np.int64))
         self.table.ops.gpu_apply_events(
             *self.table._t(),
-            torch.tensor(tokens, dtype=torch.int64, device=d),
+            torch.from_numpy(tokens_np).to(d, non_blocking=True),
             torch.tensor(tok_off, dtype=i32, device=d),
-            torch.tensor(ehashes, dtype=torch.int64, device=d),
+            torch.from_numpy(hashes_np).to(d, non_blocking=True),
             torch.tensor(eh_off, dtype=i32, device=d),
             torch.tensor(parents, dtype=torch.int64, device=d),
             torch.tensor(has_parent, dtype=torch.uint8, device=d),
