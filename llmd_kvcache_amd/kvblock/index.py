@@ -49,6 +49,7 @@ class Index:
 class IndexConfig:
     in_memory: Optional["InMemoryIndexConfig"] = None
     gpu: Optional["GpuIndexConfig"] = None
+    tiered: Optional["TieredIndexConfig"] = None
     native: Optional["TableIndexConfig"] = None
     cost_aware: Optional["CostAwareMemoryIndexConfig"] = None
     valkey: Optional["RedisIndexConfig"] = None
@@ -77,6 +78,10 @@ def new_index(cfg: Optional[IndexConfig] = None) -> Index:
         from .gpu_index import GpuIndex
 
         idx = GpuIndex(cfg.gpu)
+    elif cfg.tiered is not None:
+        from .tiered import TieredIndex
+
+        idx = TieredIndex(cfg.tiered)
     elif cfg.native is not None:
         from .gpu_index import NativeIndex
 

@@ -302,3 +302,30 @@ class TestConcurrentGpuInserts:
         sample = [Key(MODEL, 1_000_001 + i) for i in range(0, n, 977)]
         result = gpu.lookup(sample, set())
         assert set(result.keys()) == set(sample)
+
+
+class TestTieredGpu:
+    def test_gpu_hot_cpu_cold(self):
+        from llmd_kvcache_amd.kvblock.gpu_index import (
+            NativeIndex,
+            TableIndexConfig,
+        )
+        from llmd_kvcache_amd.kvblock.tiered import TieredIndex
+
+        hot = GpuIndex(GpuIndexConfig(capacity=256, pods_per_key=4))
+        cold = NativeIndex(
+            TableIndexConfig(capacity=1 << 14, pods_per_key=10),
+            registry=hot.registry,
+        )
+        t = TieredIndex(hot=hot, cold=cold)
+        all_keys = []
+        for h in range(0, 3000, 100):
+            ks = [Key(MODEL, 10_000 + h + i) for i in range(100)]
+            t.add(ks, ks, [PodEntry("pod-a", "gpu")])
+            all_keys.extend(ks)
+        torch.cuda.synchronize()
+        early = all_keys[:50]
+        merged = t.lookup(early, set())
+        assert len(merged) == len(early)
+        hot_only = t.hot.lookup(early + [Key(MODEL, 1)], set())
+        assert len(hot_only) < len(early)  # HBM tier evicted under pressure

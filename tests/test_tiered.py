@@ -1,0 +1,85 @@
+"""Two-tier index tests (CPU-injected tiers; the GPU-hot variant shares
+all logic and is exercised by the gpu-marked suite)."""
+
+import pytest
+
+torch = pytest.importorskip("torch")
+
+from llmd_kvcache_amd.kvblock.gpu_index import NativeIndex, TableIndexConfig
+from llmd_kvcache_amd.kvblock.keys import Key, PodEntry
+from llmd_kvcache_amd.kvblock.tiered import TieredIndex
+from llmd_kvcache_amd.ops import cpu_ext
+
+pytestmark = pytest.mark.skipif(
+    cpu_ext.maybe_load() is None, reason="native extension not built"
+)
+
+MODEL = "m"
+
+
+def make_tiered(hot_capacity=256, cold_capacity=1 << 14):
+    hot = NativeIndex(TableIndexConfig(capacity=hot_capacity, pods_per_key=4))
+    cold = NativeIndex(
+        TableIndexConfig(capacity=cold_capacity, pods_per_key=10),
+        registry=hot.registry,
+    )
+    return TieredIndex(hot=hot, cold=cold)
+
+
+class TestTieredIndex:
+    def test_basic_add_goes_to_both_tiers(self):
+        t = make_tiered()
+        keys = [Key(MODEL, 1), Key(MODEL, 2)]
+        t.add(keys, keys, [PodEntry("pod-a", "gpu")])
+        assert t.hot.lookup(keys, set())[keys[0]] == [PodEntry("pod-a", "gpu")]
+        assert t.cold.lookup(keys, set())[keys[0]] == [PodEntry("pod-a", "gpu")]
+        assert t.lookup(keys, set())[keys[1]] == [PodEntry("pod-a", "gpu")]
+
+    def test_cold_tier_serves_hot_evictions(self):
+        """Overfill the tiny hot tier: stolen slots must still resolve
+        from the capacity tier."""
+        t = make_tiered(hot_capacity=256)
+        all_keys = []
+        for h in range(3000):
+            k = [Key(MODEL, 10_000 + h)]
+            t.add(k, k, [PodEntry("pod-a", "gpu")])
+            all_keys.append(k[0])
+        # early keys likely evicted from hot; tiered lookup still finds them
+        early = all_keys[:50]
+        merged = t.lookup(early, set())
+        assert len(merged) == len(early)
+        hot_only = t.hot.lookup(early + [Key(MODEL, 1)], set())
+        assert len(hot_only) < len(early)  # hot tier really did evict
+
+    def test_eviction_removes_from_both(self):
+        t = make_tiered()
+        keys = [Key(MODEL, 77)]
+        t.add(keys, keys, [PodEntry("pod-a", "gpu")])
+        t.evict(keys[0], [PodEntry("pod-a", "gpu")])
+        assert keys[0] not in t.lookup(keys + [Key(MODEL, 78)], set())
+
+    def test_get_request_key_falls_back_to_cold(self):
+        t = make_tiered(hot_capacity=256)
+        for h in range(3000):
+            ek = [Key(MODEL, 50_000 + h)]
+            rk = [Key(MODEL, 90_000 + h)]
+            t.add(ek, rk, [PodEntry("pod-a", "gpu")])
+        # all engine mappings resolvable through the tier stack
+        misses = sum(
+            1
+            for h in range(0, 3000, 37)
+            if t.get_request_key(Key(MODEL, 50_000 + h)) != Key(MODEL, 90_000 + h)
+        )
+        assert misses == 0
+
+    def test_fused_scores_routes_to_hot(self):
+        t = make_tiered()
+        keys = [Key(MODEL, 5), Key(MODEL, 6)]
+        t.add(keys, keys, [PodEntry("pod-a", "gpu")])
+        from llmd_kvcache_amd.kvblock.gpu_index import _to_i64
+
+        hashes = torch.tensor([_to_i64(k.chunk_hash) for k in keys],
+                              dtype=torch.int64)
+        counts = torch.tensor([2], dtype=torch.int32)
+        maps = t.scores_to_map(t.fused_scores(hashes, counts, MODEL, set()))
+        assert maps[0] == {"pod-a": 2.0}
