@@ -198,3 +198,75 @@ class TestSubscriberToIndex:
         finally:
             pub.close()
             pool.shutdown()
+
+
+class TestReconnect:
+    def test_publisher_reconnect_after_drop(self):
+        """A dropped publisher connection must not poison the bound SUB;
+        new publishers keep flowing (reference reconnect semantics,
+        zmq_subscriber.go:55-77)."""
+        import struct
+
+        received = []
+        lock = threading.Lock()
+
+        def on_message(parts):
+            with lock:
+                received.append(parts[0])
+
+        sub = SubSocket(on_message)
+        sub.subscribe(b"kv@")
+        sub.bind("tcp://127.0.0.1:0")
+        try:
+            pub1 = PubSocket()
+            pub1.connect(f"tcp://127.0.0.1:{sub.port}")
+            assert pub1.wait_for_subscriber(5.0)
+            pub1.send_multipart([b"kv@p1@m", struct.pack(">Q", 1), b"x"])
+            pub1.close()  # abrupt drop
+
+            pub2 = PubSocket()
+            pub2.connect(f"tcp://127.0.0.1:{sub.port}")
+            assert pub2.wait_for_subscriber(5.0)
+            pub2.send_multipart([b"kv@p2@m", struct.pack(">Q", 1), b"y"])
+            deadline = time.monotonic() + 5.0
+            while time.monotonic() < deadline:
+                with lock:
+                    if b"kv@p2@m" in received:
+                        break
+                time.sleep(0.01)
+            with lock:
+                assert b"kv@p2@m" in received
+            pub2.close()
+        finally:
+            sub.close()
+
+    def test_garbage_connection_rejected_cleanly(self):
+        """A non-ZMTP client connecting to the bound SUB is dropped
+        without affecting real peers."""
+        import socket as socketlib
+        import struct
+
+        received = []
+        done = threading.Event()
+
+        def on_message(parts):
+            received.append(parts)
+            done.set()
+
+        sub = SubSocket(on_message)
+        sub.subscribe(b"kv@")
+        sub.bind("tcp://127.0.0.1:0")
+        try:
+            garbage = socketlib.create_connection(("127.0.0.1", sub.port))
+            garbage.sendall(b"GET / HTTP/1.1\r\n\r\n")
+            time.sleep(0.1)
+            garbage.close()
+
+            pub = PubSocket()
+            pub.connect(f"tcp://127.0.0.1:{sub.port}")
+            assert pub.wait_for_subscriber(5.0)
+            pub.send_multipart([b"kv@p@m", struct.pack(">Q", 1), b"z"])
+            assert done.wait(5.0)
+            pub.close()
+        finally:
+            sub.close()
