@@ -204,6 +204,35 @@ class KvTable:
               else self.ops.cpu_get_request_keys)
         return fn(*self._t(), engine_hashes, model_id)
 
+    # -- checkpoint/restore --------------------------------------------
+    # The reference index is ephemeral by design (rebuilds from the event
+    # stream; durability delegated to Redis - docs/architecture.md:129).
+    # The table layout makes snapshots nearly free on MI355X, so we offer
+    # them as a native extra: one D2H copy of the tensor bundle.
+    def state_dict(self):
+        return {
+            "keys": self.keys.cpu(),
+            "meta": self.meta.cpu(),
+            "stamp": self.stamp.cpu(),
+            "pods": self.pods.cpu(),
+            "e_keys": self.e_keys.cpu(),
+            "e_meta": self.e_meta.cpu(),
+            "e_vals": self.e_vals.cpu(),
+            "capacity": self.keys.numel(),
+            "pods_per_key": self.cfg.pods_per_key,
+            "epoch": self._epoch,
+        }
+
+    def load_state_dict(self, state):
+        if state["capacity"] != self.keys.numel() or (
+            state["pods_per_key"] != self.cfg.pods_per_key
+        ):
+            raise ValueError("snapshot shape mismatch")
+        for name in ("keys", "meta", "stamp", "pods", "e_keys", "e_meta",
+                     "e_vals"):
+            getattr(self, name).copy_(state[name].to(self.device))
+        self._epoch = int(state.get("epoch", 0))
+
 
 class TableIndex(Index):
     """Index contract over a KvTable (CPU or GPU device)."""
@@ -344,6 +373,29 @@ class TableIndex(Index):
         for i, name in enumerate(self.registry.id_to_tier):
             w[i] = weight_map.get(name, 1.0)
         return torch.tensor(w, dtype=torch.float32, device=self.device)
+
+    def save(self, path: str) -> None:
+        """Checkpoint the table + string registries to a file."""
+        state = self.table.state_dict()
+        state["registry"] = {
+            "pods": list(self.registry.id_to_pod),
+            "models": list(self.registry.id_to_model),
+            "tiers": list(self.registry.id_to_tier),
+        }
+        torch.save(state, path)
+
+    def load(self, path: str) -> None:
+        state = torch.load(path, map_location="cpu", weights_only=False)
+        reg = state.pop("registry")
+        self.table.load_state_dict(state)
+        r = self.registry
+        with r._lock:
+            r.id_to_pod = list(reg["pods"])
+            r.pod_to_id = {p: i for i, p in enumerate(r.id_to_pod)}
+            r.id_to_model = list(reg["models"])
+            r.model_to_id = {m: i for i, m in enumerate(r.id_to_model)}
+            r.id_to_tier = list(reg["tiers"])
+            r.tier_to_id = {t: i for i, t in enumerate(r.id_to_tier)}
 
     def scores_to_map(self, scores: torch.Tensor) -> List[Dict[str, float]]:
         """float [B, num_pods] -> per-prompt {pod: score} (nonzero only)."""

@@ -170,3 +170,25 @@ def test_registry_interning():
     assert reg.pod_id("pod-b") == a + 1
     m = reg.model_id("model-x")
     assert reg.model_id("model-x") == m
+
+
+def test_checkpoint_restore(tmp_path):
+    """Snapshot/restore of the table + registries (MI355X-native extra;
+    the reference index is ephemeral and delegates durability to Redis)."""
+    nat = NativeIndex(TableIndexConfig(capacity=1 << 10, pods_per_key=4))
+    keys = [Key(MODEL, h) for h in range(1, 20)]
+    nat.add(keys, keys, [PodEntry("pod-a", "gpu"), PodEntry("pod-b", "cpu")])
+    path = str(tmp_path / "index.pt")
+    nat.save(path)
+
+    fresh = NativeIndex(TableIndexConfig(capacity=1 << 10, pods_per_key=4))
+    fresh.load(path)
+    result = fresh.lookup(keys, set())
+    assert set(result.keys()) == set(keys)
+    assert set(result[keys[0]]) == {PodEntry("pod-a", "gpu"),
+                                    PodEntry("pod-b", "cpu")}
+    assert fresh.get_request_key(keys[0]) == keys[0]
+
+    wrong = NativeIndex(TableIndexConfig(capacity=1 << 11, pods_per_key=4))
+    with pytest.raises(ValueError):
+        wrong.load(path)
