@@ -147,7 +147,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--batch", type=int, default=2048,
+    ap.add_argument("--batch", type=int, default=4096,
                     help="prompts per scoring call (sub-batch)")
     ap.add_argument("--calls-per-step", type=int, default=2)
     ap.add_argument("--device", default=None, help="cpu to force CPU tables")
@@ -172,7 +172,8 @@ def main():
         backend = "nccl" if use_gpu else "gloo"
         dist.init_process_group(backend)
 
-    capacity = 1 << 22  # 4M slots for ~1M keys (plus shard headroom)
+    # power-of-two capacity at <=0.5 load factor for the requested blocks
+    capacity = 1 << max(22, (NUM_BLOCKS * 2 - 1).bit_length())
     if world > 1:
         from llmd_kvcache_amd.parallel.sharded import ShardedIndex
 

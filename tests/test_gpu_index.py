@@ -329,3 +329,29 @@ class TestTieredGpu:
         assert len(merged) == len(early)
         hot_only = t.hot.lookup(early + [Key(MODEL, 1)], set())
         assert len(hot_only) < len(early)  # HBM tier evicted under pressure
+
+
+class TestManyPods:
+    def test_fused_score_256_pods(self):
+        """Multi-word pod masks (W=4): 256-pod fleet through the fused
+        kernel (BASELINE config-5 scale)."""
+        gpu = GpuIndex(GpuIndexConfig(capacity=1 << 14, pods_per_key=10))
+        keys = [Key(MODEL, 100 + i) for i in range(4)]
+        # register 250 pods; store entries for a spread of pod ids
+        for i in range(250):
+            gpu.registry.pod_id(f"pod-{i}")
+        for pid in (0, 63, 64, 127, 128, 249):
+            gpu.add(keys, keys, [PodEntry(f"pod-{pid}", "gpu")])
+        gpu.add(keys[:2], keys[:2], [PodEntry("pod-200", "cpu")])
+        hashes = torch.tensor([_to_i64(k.chunk_hash) for k in keys],
+                              dtype=torch.int64, device="cuda")
+        offs = torch.tensor([0, 4], dtype=torch.int32, device="cuda")
+        maps = gpu.scores_to_map(
+            gpu.fused_scores(hashes, offs, MODEL, set(), max_k=4))
+        expected = {f"pod-{p}": 4.0 for p in (0, 63, 64, 127, 128, 249)}
+        expected["pod-200"] = 1.6
+        assert maps[0] == expected
+        # and agreement with the python scorer on the generic path
+        scorer = new_kv_block_scorer()
+        ref = scorer.score(keys, gpu.lookup(keys, set()))
+        assert maps[0] == {p: s for p, s in ref.items() if s != 0}
