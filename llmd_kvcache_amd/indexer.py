@@ -20,6 +20,7 @@ from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Sequence
 
 from .kvblock.index import Index, IndexConfig, new_index
+from .utils.logging import get_logger, trace
 from .kvblock.keys import Key
 from .kvblock.token_processor import ChunkedTokenDatabase, TokenProcessorConfig
 from .scorer import (
@@ -46,6 +47,9 @@ class Config:
     backend_configs: List[KVCacheBackendConfig] = field(
         default_factory=default_kv_cache_backend_configs
     )
+
+
+_log = get_logger("indexer")
 
 
 class Indexer:
@@ -88,13 +92,17 @@ class Indexer:
     ) -> Dict[str, float]:
         # 1. tokenize prompt (blocking on the pool)
         tokens = self.tokenizers_pool.tokenize(render_req, prompt, model_name)
+        trace(_log, "tokenized prompt: %d tokens", len(tokens))
 
         # 2. block keys
         block_keys = self.tokens_processor.tokens_to_kv_block_keys(
             None, tokens, model_name
         )
         if not block_keys:
+            trace(_log, "no block keys found, returning empty scores")
             return {}
+        trace(_log, "block keys: %d (first=%s)", len(block_keys),
+              block_keys[0])
         return self._score_keys(block_keys, pod_identifiers)
 
     def _score_keys(
@@ -131,8 +139,12 @@ class Indexer:
         key_to_pods = self._kv_block_index.lookup(
             block_keys, set(pod_identifiers)
         )
+        trace(_log, "lookup hit %d/%d keys", len(key_to_pods),
+              len(block_keys))
         # 4. score
-        return self.kv_block_scorer.score(block_keys, key_to_pods)
+        scores = self.kv_block_scorer.score(block_keys, key_to_pods)
+        trace(_log, "pod scores: %s", scores)
+        return scores
 
     def score_tokens(
         self,

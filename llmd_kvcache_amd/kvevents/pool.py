@@ -12,9 +12,9 @@ Parity with reference pkg/kvcache/kvevents/pool.go:
    Index.evict per hash; AllBlocksCleared -> no-op; Medium (lowercased)
    selects the device tier, defaulting to "gpu" (:35,255-259).
 
-MI355X note: with a GpuIndex the adds/evicts are batched into pinned host
-staging buffers and applied by HIP update kernels on a dedicated stream
-(gpu_index.py handles the batching transparently via add_batch/evict_batch).
+MI355X note: with a GpuIndex the workers burst-drain their queues and
+apply up to 64 messages per k_apply_events kernel launch
+(_process_burst_gpu), keeping per-pod ordering within each shard queue.
 """
 
 from __future__ import annotations
@@ -198,6 +198,8 @@ class EventsPool:
         self.digest_events(msg.pod_identifier, msg.model_name, batch.events)
 
     def digest_events(self, pod_identifier: str, model_name: str, events) -> None:
+        logger.debug("digesting %d events from %s/%s", len(events),
+                     pod_identifier, model_name)
         for ev in events:
             if isinstance(ev, BlockStored):
                 self._digest_block_stored(pod_identifier, model_name, ev)

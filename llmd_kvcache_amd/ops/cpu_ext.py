@@ -17,6 +17,8 @@ def maybe_load():
         return _mod
     _tried = True
     try:
+        import torch  # noqa: F401  (loads libc10/libtorch for the ext)
+
         from . import _kvidx_C  # type: ignore
 
         _mod = _kvidx_C
