@@ -350,8 +350,13 @@ class TestManyPods:
             gpu.fused_scores(hashes, offs, MODEL, set(), max_k=4))
         expected = {f"pod-{p}": 4.0 for p in (0, 63, 64, 127, 128, 249)}
         expected["pod-200"] = 1.6
-        assert maps[0] == expected
+        assert maps[0].keys() == expected.keys()
+        for p, v in expected.items():
+            assert maps[0][p] == pytest.approx(v)
         # and agreement with the python scorer on the generic path
         scorer = new_kv_block_scorer()
-        ref = scorer.score(keys, gpu.lookup(keys, set()))
-        assert maps[0] == {p: s for p, s in ref.items() if s != 0}
+        ref = {p: s for p, s in
+               scorer.score(keys, gpu.lookup(keys, set())).items() if s != 0}
+        assert maps[0].keys() == ref.keys()
+        for p, v in ref.items():
+            assert maps[0][p] == pytest.approx(v)
