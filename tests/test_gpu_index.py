@@ -360,3 +360,48 @@ class TestManyPods:
         assert maps[0].keys() == ref.keys()
         for p, v in ref.items():
             assert maps[0][p] == pytest.approx(v)
+
+
+class TestConcurrentReadWrite:
+    def test_reads_race_writes_across_streams(self):
+        """Fused scores on the default stream racing k_apply_events on a
+        side stream: must not fault, and post-sync lookups must reflect
+        all writes (lock-free table invariants under real concurrency)."""
+        import numpy as np
+
+        from llmd_kvcache_amd.kvblock.token_processor import (
+            ChunkedTokenDatabase,
+            TokenProcessorConfig,
+        )
+        from llmd_kvcache_amd.kvevents.events import BlockStored
+
+        tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=16))
+        gpu = GpuIndex(GpuIndexConfig(capacity=1 << 16, pods_per_key=10))
+        rng = np.random.default_rng(3)
+        side = torch.cuda.Stream()
+
+        all_tokens = []
+        h = 1
+        for step in range(10):
+            toks = rng.integers(0, 1 << 31, size=16 * 16, dtype=np.int64)
+            hs = np.arange(h, h + 16, dtype=np.uint64)
+            h += 16
+            all_tokens.append(toks)
+            with torch.cuda.stream(side):
+                gpu.apply_event_batches(
+                    [(f"pod-{step % 4}", MODEL,
+                      [BlockStored(hs, None, toks, 16)])], tp)
+            # concurrent reads on the default stream
+            probe = torch.randint(-(2**62), 2**62, (2048,),
+                                  dtype=torch.int64, device="cuda")
+            offs = torch.arange(0, 2049, 512, dtype=torch.int32,
+                                device="cuda")[:5]
+            gpu.fused_scores(probe, offs, MODEL, set(), max_k=512)
+        torch.cuda.synchronize()
+
+        # every write is visible after sync
+        for step, toks in enumerate(all_tokens):
+            keys = tp.tokens_to_kv_block_keys(None, toks.tolist(), MODEL)
+            result = gpu.lookup(keys, set())
+            assert set(result.keys()) == set(keys), step
+            assert result[keys[0]] == [PodEntry(f"pod-{step % 4}", "gpu")]
