@@ -358,7 +358,20 @@ class TableIndex(Index):
         filt = self._filter_tensor(pod_identifier_set, num_pods)
         if weights is None:
             weights = self.tier_weights()
-        with self._write_lock if not self.table.is_cuda else _nullcontext():
+        if self.table.is_cuda:
+            W = (num_pods + 63) // 64
+            k = max_k if max_k is not None else 512
+            if k * MAX_TIERS * W * 8 > 64 * 1024:
+                # LDS would overflow (huge fleet x long prompts): fall back
+                # to the two-kernel path - global-mask lookup + mask walk.
+                found, masks = self.table.lookup(hashes, model_id, filt,
+                                                 num_pods)
+                del found
+                return self.table.ops.gpu_score_from_masks(
+                    masks.contiguous(), counts_or_offsets, weights, num_pods)
+            return self.table.fused_score(hashes, counts_or_offsets, model_id,
+                                          filt, weights, num_pods, max_k)
+        with self._write_lock:
             return self.table.fused_score(hashes, counts_or_offsets, model_id,
                                           filt, weights, num_pods, max_k)
 

@@ -405,3 +405,22 @@ class TestConcurrentReadWrite:
             result = gpu.lookup(keys, set())
             assert set(result.keys()) == set(keys), step
             assert result[keys[0]] == [PodEntry(f"pod-{step % 4}", "gpu")]
+
+
+class TestLdsOverflowFallback:
+    def test_huge_fleet_long_prompt_falls_back(self):
+        """2048 registered pods x 512-key prompt exceeds the fused
+        kernel's LDS budget; fused_scores must route through the
+        two-kernel global-mask path with identical results."""
+        gpu = GpuIndex(GpuIndexConfig(capacity=1 << 14, pods_per_key=10))
+        for i in range(2048):
+            gpu.registry.pod_id(f"pod-{i}")
+        K = 512
+        keys = [Key(MODEL, 40_000 + i) for i in range(K)]
+        gpu.add(keys, keys, [PodEntry("pod-2000", "gpu")])
+        hashes = torch.tensor([_to_i64(k.chunk_hash) for k in keys],
+                              dtype=torch.int64, device="cuda")
+        offs = torch.tensor([0, K], dtype=torch.int32, device="cuda")
+        maps = gpu.scores_to_map(
+            gpu.fused_scores(hashes, offs, MODEL, set(), max_k=K))
+        assert maps[0] == {"pod-2000": float(K)}
