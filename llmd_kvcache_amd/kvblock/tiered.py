@@ -99,17 +99,14 @@ class TieredIndex(Index):
     def apply_event_batches(self, batches, token_processor=None) -> None:
         self.hot.apply_event_batches(batches, token_processor)
         # mirror into the capacity tier through the CPU digest path
-        from ..kvevents.pool import EventsConfig, EventsPool
+        from ..kvevents.pool import digest_events
 
-        pool = EventsPool.__new__(EventsPool)
-        pool.index = self.cold
-        pool.token_processor = token_processor
         if token_processor is None:
             from .token_processor import ChunkedTokenDatabase
 
-            pool.token_processor = ChunkedTokenDatabase()
+            token_processor = ChunkedTokenDatabase()
         for pod, model, events in batches:
-            pool.digest_events(pod, model, events)
+            digest_events(self.cold, token_processor, pod, model, events)
 
     def fused_scores(self, *args, **kwargs):
         return self.hot.fused_scores(*args, **kwargs)

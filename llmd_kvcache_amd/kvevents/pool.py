@@ -198,66 +198,76 @@ class EventsPool:
         self.digest_events(msg.pod_identifier, msg.model_name, batch.events)
 
     def digest_events(self, pod_identifier: str, model_name: str, events) -> None:
-        logger.debug("digesting %d events from %s/%s", len(events),
-                     pod_identifier, model_name)
-        for ev in events:
-            if isinstance(ev, BlockStored):
-                self._digest_block_stored(pod_identifier, model_name, ev)
-            elif isinstance(ev, BlockRemoved):
-                self._digest_block_removed(pod_identifier, model_name, ev)
-            elif isinstance(ev, AllBlocksCleared):
-                continue
-            else:
-                logger.debug("unknown event %r from %s", ev, pod_identifier)
+        digest_events(self.index, self.token_processor, pod_identifier,
+                      model_name, events)
 
-    def _tier(self, medium: Optional[str]) -> str:
-        return medium.lower() if medium else DEFAULT_DEVICE_TIER
 
-    def _digest_block_stored(
-        self, pod_identifier: str, model_name: str, ev: BlockStored
-    ) -> None:
-        entries = [PodEntry(pod_identifier, self._tier(ev.medium))]
+def _tier(medium: Optional[str]) -> str:
+    return medium.lower() if medium else DEFAULT_DEVICE_TIER
 
-        engine_keys: List[Key] = []
-        for raw in ev.block_hashes:
-            try:
-                engine_keys.append(Key(model_name, get_hash_as_uint64(raw)))
-            except DecodeError as e:
-                logger.debug("bad block hash %r: %s", raw, e)
-                continue
 
-        parent_request_key: Optional[Key] = None
-        if ev.parent_block_hash is not None:
-            try:
-                parent_engine_key = Key(
-                    model_name, get_hash_as_uint64(ev.parent_block_hash)
-                )
-            except DecodeError as e:
-                logger.debug("bad parent hash %r: %s", ev.parent_block_hash, e)
-                return
-            parent_request_key = self.index.get_request_key(parent_engine_key)
+def digest_events(index: Index, token_processor: ChunkedTokenDatabase,
+                  pod_identifier: str, model_name: str, events) -> None:
+    """Apply decoded events to an index (pool.go:246-338 semantics);
+    usable standalone by the tiered index and the sharded service."""
+    logger.debug("digesting %d events from %s/%s", len(events),
+                 pod_identifier, model_name)
+    for ev in events:
+        if isinstance(ev, BlockStored):
+            _digest_block_stored(index, token_processor, pod_identifier,
+                                 model_name, ev)
+        elif isinstance(ev, BlockRemoved):
+            _digest_block_removed(index, pod_identifier, model_name, ev)
+        elif isinstance(ev, AllBlocksCleared):
+            continue
+        else:
+            logger.debug("unknown event %r from %s", ev, pod_identifier)
 
-        request_keys = self.token_processor.tokens_to_kv_block_keys(
-            parent_request_key, ev.token_ids, model_name
-        )
 
-        if engine_keys:
-            try:
-                self.index.add(engine_keys, request_keys, entries)
-            except Exception as e:
-                logger.debug("failed to add event to index: %s", e)
+def _digest_block_stored(index, token_processor, pod_identifier: str,
+                         model_name: str, ev: BlockStored) -> None:
+    entries = [PodEntry(pod_identifier, _tier(ev.medium))]
 
-    def _digest_block_removed(
-        self, pod_identifier: str, model_name: str, ev: BlockRemoved
-    ) -> None:
-        entries = [PodEntry(pod_identifier, self._tier(ev.medium))]
-        for raw in ev.block_hashes:
-            try:
-                engine_key = Key(model_name, get_hash_as_uint64(raw))
-            except DecodeError as e:
-                logger.debug("bad block hash %r: %s", raw, e)
-                continue
-            try:
-                self.index.evict(engine_key, entries)
-            except Exception as e:
-                logger.debug("failed to evict from index: %s", e)
+    engine_keys: List[Key] = []
+    for raw in ev.block_hashes:
+        try:
+            engine_keys.append(Key(model_name, get_hash_as_uint64(raw)))
+        except DecodeError as e:
+            logger.debug("bad block hash %r: %s", raw, e)
+            continue
+
+    parent_request_key: Optional[Key] = None
+    if ev.parent_block_hash is not None:
+        try:
+            parent_engine_key = Key(
+                model_name, get_hash_as_uint64(ev.parent_block_hash)
+            )
+        except DecodeError as e:
+            logger.debug("bad parent hash %r: %s", ev.parent_block_hash, e)
+            return
+        parent_request_key = index.get_request_key(parent_engine_key)
+
+    request_keys = token_processor.tokens_to_kv_block_keys(
+        parent_request_key, ev.token_ids, model_name
+    )
+
+    if engine_keys:
+        try:
+            index.add(engine_keys, request_keys, entries)
+        except Exception as e:
+            logger.debug("failed to add event to index: %s", e)
+
+
+def _digest_block_removed(index, pod_identifier: str, model_name: str,
+                          ev: BlockRemoved) -> None:
+    entries = [PodEntry(pod_identifier, _tier(ev.medium))]
+    for raw in ev.block_hashes:
+        try:
+            engine_key = Key(model_name, get_hash_as_uint64(raw))
+        except DecodeError as e:
+            logger.debug("bad block hash %r: %s", raw, e)
+            continue
+        try:
+            index.evict(engine_key, entries)
+        except Exception as e:
+            logger.debug("failed to evict from index: %s", e)

@@ -81,12 +81,11 @@ class ShardedIndexService:
                 batches, self.token_processor
             )
         else:
-            from ..kvevents.pool import EventsConfig, EventsPool
+            from ..kvevents.pool import digest_events
 
-            pool = EventsPool(EventsConfig(concurrency=1), local,
-                              self.token_processor)
             for pod, model, events in batches:
-                pool.digest_events(pod, model, events)
+                digest_events(local, self.token_processor, pod, model,
+                              events)
 
     # -- rank-0 API ----------------------------------------------------
     def apply_messages(

@@ -94,6 +94,8 @@ def main():
         staged.append(batch)
 
     for mode in ("reads-only", "mixed"):
+        for _ in range(2):  # per-mode warmup
+            read_call()
         torch.cuda.synchronize()
         t0 = time.monotonic()
         blocks = 0
