@@ -487,6 +487,40 @@ __global__ void k_hash_chain_tr(const int32_t* __restrict__ tokens_t,  // [T,B]
   }
 }
 
+// Double-buffered single-chain variant: chunk c+1's coalesced loads are
+// issued BEFORE chunk c's hash chain, so HBM latency hides under the
+// serial ALU work (the plain variant exposes ~full load latency at each
+// chunk boundary).  Token loop in chunk_hash_fast is force-unrolled so
+// the buffers stay in named registers (no gpr_idx, no vmcnt(0) drain).
+template <int BS>
+__global__ void k_hash_chain_tr_pf(const int32_t* __restrict__ tokens_t,
+                                   const uint64_t* __restrict__ parents,
+                                   const int32_t* __restrict__ n_chunks,
+                                   int64_t B, int64_t L, int max_chunks,
+                                   uint64_t* __restrict__ out) {
+  const int64_t lane = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (lane >= L) return;
+  uint64_t h = parents[lane];
+  const int mc = n_chunks[lane];
+  uint32_t cur[BS], nxt[BS];
+#pragma unroll
+  for (int j = 0; j < BS; ++j)
+    cur[j] = (uint32_t)tokens_t[(int64_t)j * B + lane];
+  for (int c = 0; c < max_chunks; ++c) {
+    if (c + 1 < max_chunks) {
+#pragma unroll
+      for (int j = 0; j < BS; ++j)
+        nxt[j] = (uint32_t)tokens_t[(int64_t)((c + 1) * BS + j) * B + lane];
+    }
+    const bool active = c < mc;
+    const uint64_t h2 = chunk_hash_fast(h, cur, BS);
+    h = active ? h2 : h;
+    if (active) out[(int64_t)c * B + lane] = h;
+#pragma unroll
+    for (int j = 0; j < BS; ++j) cur[j] = nxt[j];
+  }
+}
+
 // Apply a batch of KV events fully on-device: one WAVE per pod-group
 // (events of one pod processed serially -> per-pod ordering preserved,
 // kvevents/pool.go:132-144); within a BlockStored event lane 0 streams the
@@ -788,9 +822,10 @@ at::Tensor gpu_hash_chain_tr(at::Tensor tokens_t, at::Tensor parents,
   if (block_size == 16) {
     done = true;
     switch (want_ilp) {
-      case 1: launch(k_hash_chain_tr<16, 1>); break;
+      case 1: launch(k_hash_chain_tr_pf<16>); break;  // default fast path
       case 2: launch(k_hash_chain_tr<16, 2>); break;
       case 8: launch(k_hash_chain_tr<16, 8>); break;
+      case 9: launch(k_hash_chain_tr<16, 1>); break;  // A/B: no prefetch
       default: launch(k_hash_chain_tr<16, 4>); break;
     }
   }

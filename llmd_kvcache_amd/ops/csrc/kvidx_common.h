@@ -143,6 +143,10 @@ KVIDX_HD uint64_t fnv_cbor_u64_branchless(uint64_t h, uint64_t v,
 }
 
 // Branchless chain link (bit-identical to chunk_hash; GPU hot path).
+// The token loop is force-unrolled on device: a rolled loop makes hipcc
+// read tok[] via gpr_idx (register-indexed) with a vmcnt(0) wait at the
+// loop head, which both serializes the reads and drains any prefetched
+// next-chunk loads (measured in the k_hash_chain_tr ISA).
 template <typename TokT>
 KVIDX_HD uint64_t chunk_hash_fast(uint64_t parent, const TokT* tokens,
                                   int n) {
@@ -150,6 +154,9 @@ KVIDX_HD uint64_t chunk_hash_fast(uint64_t parent, const TokT* tokens,
   h = fnv1a_64_byte(h, 0x83);
   h = fnv_cbor_u64_branchless(h, parent, 0);
   h = fnv_cbor_u32_branchless(h, (uint32_t)n, 4);
+#if defined(__HIP_DEVICE_COMPILE__)
+#pragma unroll
+#endif
   for (int i = 0; i < n; ++i)
     h = fnv_cbor_u32_branchless(h, (uint32_t)tokens[i], 0);
   h = fnv1a_64_byte(h, 0xF6);
