@@ -369,18 +369,24 @@ __global__ void __launch_bounds__(256) k_fused_score(
 }
 
 // Walk precomputed masks (e.g. after an RCCL all-reduce merge of shard
-// masks) -> scores. Grid = B, one wave per block.
+// masks) -> scores.  Per-pod scoring is independent across 64-pod words
+// (the all-empty break is only an optimization), so huge fleets split
+// across blockIdx.y word-groups of up to 16 words each.
+// Grid = (B, ceil(W/16)), one wave per block.
 __global__ void __launch_bounds__(64) k_score_from_masks(
     const unsigned long long* __restrict__ masks,  // [Ktot, T, W]
     const int32_t* __restrict__ offsets,           // [B+1]
     const float* __restrict__ weights, int num_pods, int W,
     float* __restrict__ scores) {
   const int b = blockIdx.x;
+  const int w_lo = blockIdx.y * 16;
+  const int w_hi = min(W, w_lo + 16);
+  const int WG = w_hi - w_lo;
   const int K = offsets[b + 1] - offsets[b];
   const int lane = threadIdx.x & 63;
   float score[16];
   int active[16];
-  for (int w = 0; w < W; ++w) {
+  for (int w = 0; w < WG; ++w) {
     score[w] = 0.f;
     active[w] = 0;
   }
@@ -388,11 +394,11 @@ __global__ void __launch_bounds__(64) k_score_from_masks(
     const unsigned long long* mk =
         masks + (size_t)(offsets[b] + k) * MAX_TIERS * W;
     bool any = false;
-    for (int w = 0; w < W; ++w) {
+    for (int w = 0; w < WG; ++w) {
       int cur = 0;
       float wmax = 0.f;
       for (int t = 0; t < MAX_TIERS; ++t) {
-        if ((mk[t * W + w] >> lane) & 1) {
+        if ((mk[t * W + w_lo + w] >> lane) & 1) {
           cur = 1;
           wmax = fmaxf(wmax, weights[t]);
         }
@@ -404,8 +410,8 @@ __global__ void __launch_bounds__(64) k_score_from_masks(
     }
     if (!any) break;
   }
-  for (int w = 0; w < W; ++w) {
-    int pid = w * 64 + lane;
+  for (int w = 0; w < WG; ++w) {
+    int pid = (w_lo + w) * 64 + lane;
     if (pid < num_pods) scores[(size_t)b * num_pods + pid] = score[w];
   }
 }
@@ -752,11 +758,12 @@ at::Tensor gpu_score_from_masks(at::Tensor masks, at::Tensor offsets,
                                 at::Tensor weights, int64_t num_pods) {
   int64_t B = offsets.numel() - 1;
   int64_t W = (num_pods + 63) / 64;
-  TORCH_CHECK(W <= 16, "score_from_masks supports up to 1024 pods");
+  TORCH_CHECK(W <= 64, "score_from_masks supports up to 4096 pods");
   auto scores = at::zeros({B, num_pods}, masks.options().dtype(at::kFloat));
   if (B == 0) return scores;
+  int wgroups = (int)((W + 15) / 16);
   hipLaunchKernelGGL(
-      k_score_from_masks, dim3((int)B), dim3(64), 0, STREAM,
+      k_score_from_masks, dim3((int)B, wgroups), dim3(64), 0, STREAM,
       reinterpret_cast<const unsigned long long*>(masks.data_ptr<int64_t>()),
       offsets.data_ptr<int32_t>(), weights.data_ptr<float>(), (int)num_pods,
       (int)W, scores.data_ptr<float>());
