@@ -432,7 +432,7 @@ __global__ void k_hash_chain(const int64_t* __restrict__ tokens,
   int64_t n_chunks = chunk_off[b + 1] - chunk_off[b];
   uint64_t* o = out + chunk_off[b];
   for (int64_t c = 0; c < n_chunks; ++c) {
-    h = chunk_hash(h, t0 + c * block_size, block_size);
+    h = chunk_hash_fast(h, t0 + c * block_size, block_size);
     o[c] = h;
   }
 }
@@ -592,7 +592,7 @@ __global__ void k_apply_events(
       const int64_t* t0 = tokens + tok_off[e];
       uint64_t h = parent;
       for (int c = 0; c < n_chunks && c < nh; ++c) {
-        h = chunk_hash(h, t0 + (int64_t)c * block_size, block_size);
+        h = chunk_hash_fast(h, t0 + (int64_t)c * block_size, block_size);
         req[c] = h;
       }
       // engine hashes beyond the token-derived chain keep prior behavior:
