@@ -828,12 +828,16 @@ at::Tensor gpu_hash_chain_tr(at::Tensor tokens_t, at::Tensor parents,
   bool done = false;
   if (block_size == 16) {
     done = true;
+    // Measured (profiles/r01_chain_sweep.md): the plain single-chain
+    // kernel with the force-unrolled token loop wins at every batch
+    // (0.95 ms/call vs 1.12-1.24 ms for the explicit double-buffer
+    // variant - the prefetch's register moves cost more than the load
+    // latency they hide once the unroll removed the gpr_idx waits).
     switch (want_ilp) {
-      case 1: launch(k_hash_chain_tr_pf<16>); break;  // default fast path
       case 2: launch(k_hash_chain_tr<16, 2>); break;
-      case 8: launch(k_hash_chain_tr<16, 8>); break;
-      case 9: launch(k_hash_chain_tr<16, 1>); break;  // A/B: no prefetch
-      default: launch(k_hash_chain_tr<16, 4>); break;
+      case 8: launch(k_hash_chain_tr_pf<16>); break;  // A/B: prefetch
+      case 4: launch(k_hash_chain_tr<16, 4>); break;
+      default: launch(k_hash_chain_tr<16, 1>); break;
     }
   }
   if (!done) {

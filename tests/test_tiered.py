@@ -83,3 +83,40 @@ class TestTieredIndex:
         counts = torch.tensor([2], dtype=torch.int32)
         maps = t.scores_to_map(t.fused_scores(hashes, counts, MODEL, set()))
         assert maps[0] == {"pod-a": 2.0}
+
+
+class TestValkeyBackedTier:
+    """BASELINE config 5: distributed Valkey index fronted by the
+    fast-tier table (GPU in production; CPU table here) - composition
+    through the Index contract."""
+
+    def test_table_hot_valkey_cold(self):
+        from llmd_kvcache_amd.kvblock.fake_redis import FakeRedisServer
+        from llmd_kvcache_amd.kvblock.redis_index import (
+            RedisIndexConfig,
+            ValkeyIndex,
+        )
+
+        server = FakeRedisServer()
+        server.start()
+        try:
+            hot = NativeIndex(TableIndexConfig(capacity=256, pods_per_key=4))
+            cold = ValkeyIndex(
+                RedisIndexConfig(address=f"valkey://127.0.0.1:{server.port}")
+            )
+            t = TieredIndex(hot=hot, cold=cold)
+            all_keys = []
+            for h in range(2000):
+                k = [Key(MODEL, 70_000 + h)]
+                t.add(k, k, [PodEntry("pod-v", "gpu")])
+                all_keys.append(k[0])
+            # hot tier evicted early keys; valkey still serves them
+            early = all_keys[:40]
+            merged = t.lookup(early, set())
+            assert len(merged) == len(early)
+            hot_only = t.hot.lookup(early + [Key(MODEL, 1)], set())
+            assert len(hot_only) < len(early)
+            # dual keys resolve through the stack
+            assert t.get_request_key(all_keys[0]) == all_keys[0]
+        finally:
+            server.stop()
