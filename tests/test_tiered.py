@@ -100,16 +100,15 @@ class TestValkeyBackedTier:
         server = FakeRedisServer()
         server.start()
         try:
-            hot = NativeIndex(TableIndexConfig(capacity=256, pods_per_key=4))
+            hot = NativeIndex(TableIndexConfig(capacity=128, pods_per_key=4))
             cold = ValkeyIndex(
                 RedisIndexConfig(address=f"valkey://127.0.0.1:{server.port}")
             )
             t = TieredIndex(hot=hot, cold=cold)
-            all_keys = []
-            for h in range(2000):
-                k = [Key(MODEL, 70_000 + h)]
-                t.add(k, k, [PodEntry("pod-v", "gpu")])
-                all_keys.append(k[0])
+            all_keys = [Key(MODEL, 70_000 + h) for h in range(600)]
+            for lo in range(0, 600, 50):
+                ks = all_keys[lo:lo + 50]
+                t.add(ks, ks, [PodEntry("pod-v", "gpu")])
             # hot tier evicted early keys; valkey still serves them
             early = all_keys[:40]
             merged = t.lookup(early, set())
