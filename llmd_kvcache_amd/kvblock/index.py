@@ -14,7 +14,7 @@ open-addressing hash table with wave-cooperative HIP kernels (gpu_index.py).
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Sequence, Set
 
 from .keys import Key, PodEntry

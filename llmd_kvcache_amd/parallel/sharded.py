@@ -24,17 +24,12 @@ Design:
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Sequence, Set
+from typing import Dict, Optional, Sequence, Set
 
 import torch
 import torch.distributed as dist
 
-from ..kvblock.gpu_index import (
-    MAX_TIERS,
-    TableIndex,
-    TableIndexConfig,
-    _to_i64,
-)
+from ..kvblock.gpu_index import TableIndex, TableIndexConfig, _to_i64
 
 
 class ShardedIndex:

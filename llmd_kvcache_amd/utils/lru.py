@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import threading
 from collections import OrderedDict
-from typing import Any, Hashable, Iterator, Optional, Tuple
+from typing import Any, Hashable, Iterator, Tuple
 
 
 class LRUCache:
