@@ -325,6 +325,29 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t[0])
 
+    # single-prompt request latency (separate, not part of the headline):
+    # one 8k-token prompt through hash-chain + probe + score + D2H.
+    single_ms = None
+    if index.table.is_cuda and sharded is None:
+        one_tok = call_tokens[0][:, :1].contiguous()
+        one_par = parents[:1]
+        one_nch = nchunks_t[:1]
+        one_off = torch.tensor([0, KEYS_PER_PROMPT], dtype=torch.int32,
+                               device=device)
+        lat1 = []
+        for _ in range(30):
+            t0 = time.monotonic()
+            hh = ops.gpu_hash_chain_tr(one_tok, one_par, one_nch,
+                                       BLOCK_SIZE, KEYS_PER_PROMPT, 0)
+            sc = ops.gpu_fused_score(
+                *index.table._t(), hh.view(-1), one_off, model_id,
+                no_filter, weights, num_pods, index.table.next_epoch(),
+                KEYS_PER_PROMPT)
+            sc.argmax(dim=1).cpu()
+            torch.cuda.synchronize()
+            lat1.append(time.monotonic() - t0)
+        single_ms = statistics.median(lat1) * 1000.0
+
     # strong scaling: all ranks score the same global prompt stream
     # cooperatively (index sharded N ways, probes split by ownership,
     # masks merged over RCCL); total work is fixed as N grows.
@@ -358,6 +381,8 @@ def main():
                 "block_size": BLOCK_SIZE,
                 "keys_per_prompt": KEYS_PER_PROMPT,
                 "p50_batch_latency_ms": round(p50_ms, 3),
+                "p50_single_prompt_ms": (round(single_ms, 3)
+                                         if single_ms is not None else None),
                 "batch_per_call": args.batch,
                 "ingest_blocks_per_sec": round(ingest_rate, 1),
                 "timed_path": "hash-chain + probe + longest-prefix score "
