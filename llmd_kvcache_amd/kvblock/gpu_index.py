@@ -580,8 +580,7 @@ class GpuIndex(TableIndex):
                      else np.zeros(0, dtype=np.int64))
         hashes_np = (np.concatenate(hash_arrays) if hash_arrays
                      else np.zeros(0, dtype=np.int64))
-        self.table.ops.gpu_apply_events(
-            *self.table._t(),
+        common = (
             torch.from_numpy(tokens_np).to(d, non_blocking=True),
             torch.tensor(tok_off, dtype=i32, device=d),
             torch.from_numpy(hashes_np).to(d, non_blocking=True),
@@ -591,6 +590,33 @@ class GpuIndex(TableIndex):
             torch.tensor(ev_type, dtype=torch.uint8, device=d),
             torch.tensor(pod_entry, dtype=i32, device=d),
             torch.tensor(grp_off, dtype=i32, device=d),
-            model_id, _to_i64(init_hash), block_size,
-            self.table.next_epoch(), self.cfg.shard_id, self.cfg.num_shards,
         )
+        # Phase-split fast path (chains lane-per-group, inserts
+        # thread-per-block) unless this batch both stores AND removes some
+        # engine hash - then per-pod op ORDER matters and the serial
+        # wave-per-group kernel preserves it.
+        use_split = True
+        if any(ev_type):
+            stored, removed = set(), set()
+            eh_list = hashes_np.tolist()
+            for e, t in enumerate(ev_type):
+                span = eh_list[eh_off[e]:eh_off[e + 1]]
+                (removed if t else stored).update(span)
+            use_split = not (stored & removed)
+        if use_split:
+            counts = np.diff(np.asarray(eh_off, dtype=np.int64))
+            ev_of = np.repeat(np.arange(len(counts), dtype=np.int32), counts)
+            self.table.ops.gpu_apply_events_split(
+                *self.table._t(), *common,
+                torch.from_numpy(ev_of).to(d, non_blocking=True),
+                model_id, _to_i64(init_hash), block_size,
+                self.table.next_epoch(), self.cfg.shard_id,
+                self.cfg.num_shards,
+            )
+        else:
+            self.table.ops.gpu_apply_events(
+                *self.table._t(), *common,
+                model_id, _to_i64(init_hash), block_size,
+                self.table.next_epoch(), self.cfg.shard_id,
+                self.cfg.num_shards,
+            )

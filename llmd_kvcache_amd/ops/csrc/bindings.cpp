@@ -68,6 +68,12 @@ void gpu_apply_events(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, at::Tensor, int64_t,
                       int64_t, int64_t, int64_t, int64_t, int64_t);
+void gpu_apply_events_split(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                            at::Tensor, at::Tensor, at::Tensor, int64_t,
+                            at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                            at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                            at::Tensor, at::Tensor, int64_t, int64_t, int64_t,
+                            int64_t, int64_t, int64_t);
 #endif
 
 }  // namespace kvidx
@@ -96,6 +102,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("tokens_t"), py::arg("parents"), py::arg("n_chunks"),
         py::arg("block_size"), py::arg("max_chunks"), py::arg("ilp") = 0);
   m.def("gpu_apply_events", &kvidx::gpu_apply_events);
+  m.def("gpu_apply_events_split", &kvidx::gpu_apply_events_split);
 #else
   m.attr("HAS_HIP") = false;
 #endif

@@ -614,6 +614,139 @@ __global__ void k_apply_events(
 }
 
 // ---------------------------------------------------------------------
+// Phase-split event application (the serial wave-per-group kernel above
+// left 63 lanes idle during each chain; here chains run one LANE per
+// pod-group and the per-block inserts fan out thread-per-block).
+// The host guards ordering: a batch where some engine hash is both
+// stored and removed falls back to the serial kernel.
+// ---------------------------------------------------------------------
+
+// Batch-local engine-hash -> global block index map (open addressing,
+// zeroed tensors per call; value = idx+1 so 0 means empty).
+__global__ void k_batch_bmap(const uint64_t* __restrict__ ehashes, int64_t n,
+                             uint64_t* __restrict__ bkeys,
+                             int32_t* __restrict__ bvals, int64_t bmask) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  uint64_t h = remap_hash(ehashes[i]);
+  uint64_t s = probe_start(h, (uint64_t)bmask);
+  for (int t = 0; t < PROBE_MAX; ++t) {
+    uint64_t j = (s + t) & (uint64_t)bmask;
+    uint64_t prev = atomicCAS((unsigned long long*)&bkeys[j], 0ull,
+                              (unsigned long long)h);
+    if (prev == 0) {
+      bvals[j] = (int32_t)(i + 1);  // consumers run in a later kernel
+      return;
+    }
+    if (prev == h) return;  // duplicate hash: first writer's index stands
+  }
+}
+
+DEV int64_t dev_bmap_find(const uint64_t* __restrict__ bkeys,
+                          const int32_t* __restrict__ bvals, int64_t bmask,
+                          uint64_t h) {
+  h = remap_hash(h);
+  uint64_t s = probe_start(h, (uint64_t)bmask);
+  for (int t = 0; t < PROBE_MAX; ++t) {
+    uint64_t j = (s + t) & (uint64_t)bmask;
+    uint64_t k = bkeys[j];
+    if (k == 0) return -1;
+    if (k == h) {
+      int32_t v = bvals[j];
+      return v > 0 ? (int64_t)v - 1 : -1;
+    }
+  }
+  return -1;
+}
+
+// Phase A: one lane per pod-group walks its events in order computing
+// request-key chains into req_scratch.  Parent resolution: earlier block
+// of the SAME group in this batch (bmap) first, then the persistent
+// engine map (cross-batch), else the chain root.
+__global__ void k_event_chains(
+    DevTable v, const int64_t* __restrict__ tokens,
+    const int32_t* __restrict__ tok_off, const uint64_t* __restrict__ ehashes,
+    const int32_t* __restrict__ eh_off, const uint64_t* __restrict__ parents,
+    const uint8_t* __restrict__ has_parent, const uint8_t* __restrict__ ev_type,
+    const int32_t* __restrict__ grp_off, int64_t G, uint32_t model,
+    uint64_t init_hash, int block_size,
+    const uint64_t* __restrict__ bkeys, const int32_t* __restrict__ bvals,
+    int64_t bmask, uint64_t* __restrict__ req_scratch) {
+  int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (g >= G) return;
+  const int e_begin = grp_off[g];
+  const int e_end = grp_off[g + 1];
+  const int grp_first_block = eh_off[e_begin];
+  for (int e = e_begin; e < e_end; ++e) {
+    if (ev_type[e] == 1) continue;  // removals have no chain
+    const int nh = eh_off[e + 1] - eh_off[e];
+    const int n_chunks = (tok_off[e + 1] - tok_off[e]) / block_size;
+    uint64_t parent = init_hash;
+    if (has_parent[e]) {
+      int64_t bi = dev_bmap_find(bkeys, bvals, bmask, parents[e]);
+      if (bi >= grp_first_block && bi < eh_off[e]) {
+        parent = req_scratch[bi];  // earlier event of this group
+      } else {
+        int64_t ei = dev_emap_find(v, parents[e], model);
+        if (ei >= 0) parent = v.e_vals[ei];
+      }
+    }
+    const int64_t* t0 = tokens + tok_off[e];
+    uint64_t h = parent;
+    const int n_ins = min(n_chunks, nh);
+    for (int c = 0; c < n_ins; ++c) {
+      h = chunk_hash_fast(h, t0 + (int64_t)c * block_size, block_size);
+      req_scratch[eh_off[e] + c] = h;
+    }
+  }
+}
+
+// Phase B: one thread per block applies its insert/removal.
+__global__ void k_event_inserts(
+    DevTable v, const uint64_t* __restrict__ ehashes,
+    const int32_t* __restrict__ eh_off, const int32_t* __restrict__ ev_of,
+    const uint8_t* __restrict__ ev_type, const int32_t* __restrict__ tok_off,
+    const uint32_t* __restrict__ pod_entry, int64_t n_blocks, uint32_t model,
+    int block_size, int32_t epoch, int shard_id, int num_shards,
+    const uint64_t* __restrict__ req_scratch) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n_blocks) return;
+  const int e = ev_of[i];
+  if (ev_type[e] == 1) {  // BlockRemoved
+    int64_t ei = dev_emap_find(v, ehashes[i], model);
+    if (ei < 0) return;
+    uint64_t req = v.e_vals[ei];
+    int64_t slot = dev_table_find(v, req, model);
+    if (slot < 0) {
+      atomicOr(&v.e_meta[ei], META_TOMB);
+      return;
+    }
+    uint32_t* p = v.pods + slot * v.pods_per_key;
+    for (int k = 0; k < v.pods_per_key; ++k)
+      atomicCAS(&p[k], pod_entry[e], 0u);
+    bool empty = true;
+    for (int k = 0; k < v.pods_per_key; ++k)
+      if (p[k] != 0) { empty = false; break; }
+    if (empty) {
+      atomicOr(&v.meta[slot], META_TOMB);
+      atomicOr(&v.e_meta[ei], META_TOMB);
+    }
+    return;
+  }
+  // BlockStored: only blocks covered by the token chain insert
+  const int local = (int)(i - eh_off[e]);
+  const int n_chunks = (tok_off[e + 1] - tok_off[e]) / block_size;
+  if (local >= n_chunks) return;
+  const uint64_t req = req_scratch[i];
+  dev_emap_put(v, ehashes[i], model, remap_hash(req));
+  if (num_shards > 1 &&
+      (int)(remap_hash(req) % (uint64_t)num_shards) != shard_id)
+    return;
+  int64_t slot = dev_table_put(v, req, model, epoch);
+  dev_pod_set_add(v, slot, pod_entry[e], epoch);
+}
+
+// ---------------------------------------------------------------------
 // Launchers
 // ---------------------------------------------------------------------
 
@@ -885,6 +1018,60 @@ void gpu_apply_events(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
       (uint64_t)init_hash_bits, (int)block_size, (int32_t)epoch,
       (int)shard_id, (int)num_shards,
       reinterpret_cast<uint64_t*>(req_scratch.data_ptr<int64_t>()));
+}
+
+// Phase-split apply: bmap build -> chains (lane per group) -> inserts
+// (thread per block).  ev_of: int32 [total blocks] = owning event index.
+void gpu_apply_events_split(
+    at::Tensor keys, at::Tensor meta, at::Tensor stamp, at::Tensor pods,
+    at::Tensor e_keys, at::Tensor e_meta, at::Tensor e_vals,
+    int64_t pods_per_key, at::Tensor tokens, at::Tensor tok_off,
+    at::Tensor ehashes, at::Tensor eh_off, at::Tensor parents,
+    at::Tensor has_parent, at::Tensor ev_type, at::Tensor pod_entry,
+    at::Tensor grp_off, at::Tensor ev_of, int64_t model_id,
+    int64_t init_hash_bits, int64_t block_size, int64_t epoch,
+    int64_t shard_id, int64_t num_shards) {
+  auto v = dev_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
+                    pods_per_key);
+  int64_t G = grp_off.numel() - 1;
+  int64_t n = ehashes.numel();
+  if (G == 0 || n == 0) return;
+  int64_t bcap = 4;
+  while (bcap < n * 2) bcap <<= 1;
+  auto bkeys = at::zeros({bcap}, ehashes.options());
+  auto bvals = at::zeros({bcap}, ehashes.options().dtype(at::kInt));
+  auto req_scratch = at::zeros({n}, ehashes.options());
+  int threads = 256;
+
+  hipLaunchKernelGGL(k_batch_bmap,
+                     dim3((int)((n + threads - 1) / threads)), dim3(threads),
+                     0, STREAM, U64P(ehashes), n,
+                     reinterpret_cast<uint64_t*>(bkeys.data_ptr<int64_t>()),
+                     bvals.data_ptr<int32_t>(), bcap - 1);
+  hipLaunchKernelGGL(k_event_chains,
+                     dim3((int)((G + threads - 1) / threads)), dim3(threads),
+                     0, STREAM, v, tokens.data_ptr<int64_t>(),
+                     tok_off.data_ptr<int32_t>(), U64P(ehashes),
+                     eh_off.data_ptr<int32_t>(), U64P(parents),
+                     has_parent.data_ptr<uint8_t>(),
+                     ev_type.data_ptr<uint8_t>(), grp_off.data_ptr<int32_t>(),
+                     G, (uint32_t)model_id, (uint64_t)init_hash_bits,
+                     (int)block_size,
+                     reinterpret_cast<uint64_t*>(bkeys.data_ptr<int64_t>()),
+                     bvals.data_ptr<int32_t>(), bcap - 1,
+                     reinterpret_cast<uint64_t*>(
+                         req_scratch.data_ptr<int64_t>()));
+  hipLaunchKernelGGL(k_event_inserts,
+                     dim3((int)((n + threads - 1) / threads)), dim3(threads),
+                     0, STREAM, v, U64P(ehashes), eh_off.data_ptr<int32_t>(),
+                     ev_of.data_ptr<int32_t>(), ev_type.data_ptr<uint8_t>(),
+                     tok_off.data_ptr<int32_t>(),
+                     reinterpret_cast<const uint32_t*>(
+                         pod_entry.data_ptr<int32_t>()),
+                     n, (uint32_t)model_id, (int)block_size, (int32_t)epoch,
+                     (int)shard_id, (int)num_shards,
+                     reinterpret_cast<uint64_t*>(
+                         req_scratch.data_ptr<int64_t>()));
 }
 
 }  // namespace kvidx
