@@ -42,7 +42,23 @@ def config_to_json(cfg: Config, **kw) -> str:
     return json.dumps(config_to_dict(cfg), **kw)
 
 
+def _dataclass_of(field_type: Any) -> Any:
+    """Resolve Optional[SomeDataclass] / SomeDataclass annotations."""
+    import typing
+
+    for t in typing.get_args(field_type) or (field_type,):
+        if dataclasses.is_dataclass(t):
+            return t
+    return None
+
+
 def _apply(obj: Any, data: Dict[str, Any]) -> Any:
+    import typing
+
+    try:
+        hints = typing.get_type_hints(type(obj))
+    except Exception:
+        hints = {}
     for f in dataclasses.fields(obj):
         if f.name not in data or f.name.startswith("_"):
             continue
@@ -50,6 +66,14 @@ def _apply(obj: Any, data: Dict[str, Any]) -> Any:
         cur = getattr(obj, f.name)
         if dataclasses.is_dataclass(cur) and isinstance(val, dict):
             _apply(cur, val)
+        elif cur is None and isinstance(val, dict):
+            # e.g. tokenizers_pool.uds defaults to None: construct the
+            # annotated dataclass and overlay into it
+            dc = _dataclass_of(hints.get(f.name, f.type))
+            if dc is not None:
+                setattr(obj, f.name, _apply(dc(), val))
+            else:
+                setattr(obj, f.name, val)
         else:
             setattr(obj, f.name, val)
     return obj

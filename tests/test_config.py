@@ -115,3 +115,15 @@ class TestEventsPoolBurst:
         pool.drain()
         pool.shutdown()
         assert index.applied == []
+
+
+class TestNestedNoneOverlay:
+    def test_none_default_nested_dataclass_constructed(self):
+        cfg = config_from_dict({
+            "tokenizers_pool": {"uds": {"socket_path": "/tmp/x.sock"}}
+        })
+        from llmd_kvcache_amd.tokenization.uds import UdsTokenizerConfig
+
+        assert isinstance(cfg.tokenizers_pool.uds, UdsTokenizerConfig)
+        assert cfg.tokenizers_pool.uds.socket_path == "/tmp/x.sock"
+        assert cfg.tokenizers_pool.uds.timeout_s == 5.0  # default kept
