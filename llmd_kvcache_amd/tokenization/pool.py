@@ -22,6 +22,7 @@ from dataclasses import dataclass, field
 from typing import List, Optional
 
 from .prefixstore import LRUTokenStore
+from .uds import UdsTokenizerConfig
 from .tokenizer import (
     CompositeTokenizer,
     HFTokenizerConfig,
@@ -42,7 +43,7 @@ class TokenizationConfig:
     local: Optional[LocalTokenizerConfig] = field(
         default_factory=LocalTokenizerConfig.from_env
     )
-    uds: Optional["UdsTokenizerConfig"] = None
+    uds: Optional[UdsTokenizerConfig] = None
     hf: Optional[HFTokenizerConfig] = field(default_factory=HFTokenizerConfig)
 
 
