@@ -345,15 +345,20 @@ void cpu_evict(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
   auto pe = pod_entries.contiguous();
   const uint64_t* ehp = reinterpret_cast<uint64_t*>(eh.data_ptr<int64_t>());
   const uint32_t* pep = reinterpret_cast<uint32_t*>(pe.data_ptr<int32_t>());
+  // Engine->request mappings are deliberately RETAINED on eviction
+  // (unlike in_memory.go:252-255): with the sharded deployment the emap
+  // is replicated while the main table is ownership-filtered, so only
+  // the owning shard can observe "pod set now empty" - dropping the
+  // mapping locally would diverge the replicas and break parent-chain
+  // stitching after removals.  A retained mapping also keeps child
+  // chains content-consistent (the reference restarts them from the
+  // root); stale entries are reclaimed by the emap's own slot-steal.
   for (int64_t i = 0; i < eh.numel(); ++i) {
     int64_t ei = emap_find(v, ehp[i], (uint32_t)model_id);
     if (ei < 0) continue;
     uint64_t req = v.e_vals[ei];
     int64_t slot = table_find(v, req, (uint32_t)model_id);
-    if (slot < 0) {
-      v.e_meta[ei] = META_OCC | META_TOMB;  // dangling mapping: drop
-      continue;
-    }
+    if (slot < 0) continue;
     uint32_t* p = v.pods + slot * v.pods_per_key;
     for (int64_t j = 0; j < pe.numel(); ++j)
       for (int k = 0; k < v.pods_per_key; ++k)
@@ -361,10 +366,7 @@ void cpu_evict(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
     bool empty = true;
     for (int k = 0; k < v.pods_per_key; ++k)
       if (p[k] != 0) { empty = false; break; }
-    if (empty) {
-      v.meta[slot] |= META_TOMB;
-      v.e_meta[ei] = META_OCC | META_TOMB;
-    }
+    if (empty) v.meta[slot] |= META_TOMB;
   }
 }
 

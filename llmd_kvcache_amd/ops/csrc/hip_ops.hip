@@ -247,14 +247,13 @@ __global__ void k_evict(DevTable v, const uint64_t* __restrict__ eh,
                         const uint32_t* __restrict__ entries, int n_entries) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
+  // emap entries retained on eviction (replica determinism under
+  // sharding + chain continuity; see cpu_evict note).
   int64_t ei = dev_emap_find(v, eh[i], model);
   if (ei < 0) return;
   uint64_t req = v.e_vals[ei];
   int64_t slot = dev_table_find(v, req, model);
-  if (slot < 0) {
-    atomicOr(&v.e_meta[ei], META_TOMB);
-    return;
-  }
+  if (slot < 0) return;
   uint32_t* p = v.pods + slot * v.pods_per_key;
   for (int j = 0; j < n_entries; ++j)
     for (int k = 0; k < v.pods_per_key; ++k)
@@ -262,10 +261,7 @@ __global__ void k_evict(DevTable v, const uint64_t* __restrict__ eh,
   bool empty = true;
   for (int k = 0; k < v.pods_per_key; ++k)
     if (p[k] != 0) { empty = false; break; }
-  if (empty) {
-    atomicOr(&v.meta[slot], META_TOMB);
-    atomicOr(&v.e_meta[ei], META_TOMB);
-  }
+  if (empty) atomicOr(&v.meta[slot], META_TOMB);
 }
 
 __global__ void k_get_request_keys(DevTable v,
@@ -561,20 +557,14 @@ __global__ void k_apply_events(
         if (ei < 0) continue;
         uint64_t req = v.e_vals[ei];
         int64_t slot = dev_table_find(v, req, model);
-        if (slot < 0) {
-          atomicOr(&v.e_meta[ei], META_TOMB);
-          continue;
-        }
+        if (slot < 0) continue;  // emap retained (see cpu_evict note)
         uint32_t* p = v.pods + slot * v.pods_per_key;
         for (int k = 0; k < v.pods_per_key; ++k)
           atomicCAS(&p[k], pod_entry[e], 0u);
         bool empty = true;
         for (int k = 0; k < v.pods_per_key; ++k)
           if (p[k] != 0) { empty = false; break; }
-        if (empty) {
-          atomicOr(&v.meta[slot], META_TOMB);
-          atomicOr(&v.e_meta[ei], META_TOMB);
-        }
+        if (empty) atomicOr(&v.meta[slot], META_TOMB);
       }
       __builtin_amdgcn_wave_barrier();
       continue;
@@ -717,20 +707,14 @@ __global__ void k_event_inserts(
     if (ei < 0) return;
     uint64_t req = v.e_vals[ei];
     int64_t slot = dev_table_find(v, req, model);
-    if (slot < 0) {
-      atomicOr(&v.e_meta[ei], META_TOMB);
-      return;
-    }
+    if (slot < 0) return;  // emap retained (see cpu_evict note)
     uint32_t* p = v.pods + slot * v.pods_per_key;
     for (int k = 0; k < v.pods_per_key; ++k)
       atomicCAS(&p[k], pod_entry[e], 0u);
     bool empty = true;
     for (int k = 0; k < v.pods_per_key; ++k)
       if (p[k] != 0) { empty = false; break; }
-    if (empty) {
-      atomicOr(&v.meta[slot], META_TOMB);
-      atomicOr(&v.e_meta[ei], META_TOMB);
-    }
+    if (empty) atomicOr(&v.meta[slot], META_TOMB);
     return;
   }
   // BlockStored: only blocks covered by the token chain insert

@@ -79,12 +79,15 @@ def test_differential_vs_in_memory(seed):
             assert set(r_ref.keys()) == set(r_nat.keys()), op
             for k in r_ref:
                 assert set(r_ref[k]) == set(r_nat[k]), (op, k)
-    # final dual-key mapping parity on a sample
+    # Dual-key mapping: the table backends deliberately RETAIN engine
+    # mappings on eviction (replica determinism under sharding + chain
+    # continuity; see ops/csrc/cpu_ops.cpp cpu_evict note), so the native
+    # map is a superset of the reference's; where both exist they agree.
     for h in range(1000, 1040):
         k = Key(MODEL, h)
-        assert (ref.get_request_key(k) is None) == (
-            nat.get_request_key(k) is None
-        )
+        ref_rk = ref.get_request_key(k)
+        if ref_rk is not None:
+            assert nat.get_request_key(k) == ref_rk
 
 
 @pytest.mark.parametrize("seed", [11, 12])
