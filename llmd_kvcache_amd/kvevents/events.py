@@ -136,8 +136,18 @@ def decode_event_batch(payload: bytes) -> EventBatch:
     if not isinstance(arr, (list, tuple)) or len(arr) < 2:
         raise DecodeError("event batch is not a >=2 element array")
 
-    ts = float(arr[0])
-    dp_rank = int(arr[2]) if len(arr) > 2 and arr[2] is not None else None
+    try:
+        ts = float(arr[0])
+    except (TypeError, ValueError) as e:
+        raise DecodeError(f"bad batch timestamp: {e}") from e
+    if not isinstance(arr[1], (list, tuple)):
+        raise DecodeError("batch events field is not an array")
+    dp_rank = None
+    if len(arr) > 2 and arr[2] is not None:
+        try:
+            dp_rank = int(arr[2])
+        except (TypeError, ValueError):
+            dp_rank = None
 
     events = []
     for raw in arr[1]:
