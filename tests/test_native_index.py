@@ -28,27 +28,40 @@ MODEL = "m"
 
 
 def random_workload(rng, n_ops=300, key_space=200, n_pods=8):
-    """Generates (op, args) tuples; chain-shaped adds like real events."""
+    """Generates (op, args) tuples; chain-shaped adds like real events,
+    including distinct engine/request keys and filtered lookups."""
     ops = []
     for _ in range(n_ops):
         r = rng.random()
-        if r < 0.6:
+        if r < 0.55:
             start = rng.randrange(key_space)
             n = rng.randrange(1, 8)
-            keys = [Key(MODEL, 1000 + (start + i) % key_space) for i in range(n)]
+            rkeys = [Key(MODEL, 1000 + (start + i) % key_space)
+                     for i in range(n)]
+            # half the adds use distinct engine keys (dual-key design)
+            if rng.random() < 0.5:
+                ekeys = [Key(MODEL, 500_000 + k.chunk_hash) for k in rkeys]
+            else:
+                ekeys = rkeys
             pod = f"pod-{rng.randrange(n_pods)}"
             tier = rng.choice(["gpu", "cpu"])
-            ops.append(("add", keys, keys, [PodEntry(pod, tier)]))
-        elif r < 0.8:
+            ops.append(("add", ekeys, rkeys, [PodEntry(pod, tier)]))
+        elif r < 0.75:
             h = 1000 + rng.randrange(key_space)
+            if rng.random() < 0.5:
+                h += 500_000  # engine-keyed eviction
             pod = f"pod-{rng.randrange(n_pods)}"
             tier = rng.choice(["gpu", "cpu"])
             ops.append(("evict", Key(MODEL, h), [PodEntry(pod, tier)]))
         else:
             start = rng.randrange(key_space)
             n = rng.randrange(1, 16)
-            keys = [Key(MODEL, 1000 + (start + i) % key_space) for i in range(n)]
-            ops.append(("lookup", keys))
+            keys = [Key(MODEL, 1000 + (start + i) % key_space)
+                    for i in range(n)]
+            filt = (set() if rng.random() < 0.5 else
+                    {f"pod-{rng.randrange(n_pods)}"
+                     for _ in range(rng.randrange(1, 4))})
+            ops.append(("lookup", keys, filt))
     return ops
 
 
@@ -60,7 +73,8 @@ def run_op(index, op):
         elif kind == "evict":
             index.evict(op[1], op[2])
         elif kind == "lookup":
-            return index.lookup(op[1], set())
+            filt = op[2] if len(op) > 2 else set()
+            return index.lookup(op[1], filt)
     except ValueError:
         return "error"
     return None
