@@ -122,19 +122,19 @@ def _apply_cpu(index, batch, tp):
 def build_prompts(chains, n_prompts, device, seed):
     """Prompts = one stored chain (shared prefix, hits) + fresh random
     tail (misses) - the shared-prefix routing workload of the reference
-    benchmarks (benchmarking/37-capacity: 8k shared prefix).  Vectorized:
-    prefix rows gathered from the chain pool, tails drawn fresh."""
+    benchmarks (benchmarking/37-capacity: 8k shared prefix)."""
     import numpy as np
 
     rng = np.random.default_rng(seed)
     reuse = PROMPT_TOKENS // 2
-    chain_pool = np.stack([np.tile(c, reuse // len(c) + 1)[:reuse]
-                           for c in chains])
-    pick = rng.integers(len(chains), size=n_prompts)
     all_tokens = np.empty((n_prompts, PROMPT_TOKENS), dtype=np.int64)
-    all_tokens[:, :reuse] = chain_pool[pick]
-    all_tokens[:, reuse:] = rng.integers(
-        0, 1 << 31, size=(n_prompts, PROMPT_TOKENS - reuse), dtype=np.int64)
+    for i in range(n_prompts):
+        chain = chains[rng.integers(len(chains))]
+        prefix = np.tile(chain, reuse // len(chain) + 1)[:reuse]
+        tail = rng.integers(0, 1 << 31, size=PROMPT_TOKENS - reuse,
+                            dtype=np.int64)
+        all_tokens[i, :reuse] = prefix
+        all_tokens[i, reuse:] = tail
     t = torch.from_numpy(all_tokens.reshape(-1)).to(device)
     offsets = torch.arange(0, (n_prompts + 1) * PROMPT_TOKENS, PROMPT_TOKENS,
                            dtype=torch.int64, device=device)
@@ -154,8 +154,6 @@ def main():
     ap.add_argument("--blocks", type=int, default=NUM_BLOCKS)
     ap.add_argument("--graph", action="store_true",
                     help="capture the read call in a hipGraph and replay")
-    ap.add_argument("--no-memo", dest="memo", action="store_false",
-                    help="disable the chain memo table")
     args = ap.parse_args()
     NUM_BLOCKS = args.blocks
 
@@ -204,10 +202,8 @@ def main():
               f"(world={world}, shard={index.cfg.shard_id}/{index.cfg.num_shards})")
     chains, ingest_rate = populate_index(index, device, rank)
 
-    total_calls = (args.warmup + args.steps) * args.calls_per_step
     n_prompts = args.batch * args.calls_per_step
-    tokens, tok_offsets = build_prompts(
-        chains, args.batch * total_calls, device, seed=99)
+    tokens, tok_offsets = build_prompts(chains, n_prompts, device, seed=99)
     tp_init = _to_i64(ChunkedTokenDatabase(
         TokenProcessorConfig(block_size=BLOCK_SIZE)).config.init_hash())
     parents = torch.full((args.batch,), tp_init, dtype=torch.int64,
@@ -227,32 +223,21 @@ def main():
         # pre-stage each call's tokens in the chain kernel's native layout
         # ([token_pos][prompt] int32, coalesced lane loads); pre-staging the
         # tensor is layout-neutral work the service does once per request
-        # batch either way.  Every call in every step gets UNIQUE prompts
-        # (fresh tails; prefixes recur, as a real request stream does).
+        # batch either way.
         call_tokens = [
             tokens[c * args.batch * PROMPT_TOKENS:
                    (c + 1) * args.batch * PROMPT_TOKENS]
             .view(args.batch, PROMPT_TOKENS).to(torch.int32).t().contiguous()
-            for c in range(total_calls)
+            for c in range(args.calls_per_step)
         ]
         chain_stream = torch.cuda.Stream(device=device)
-        memo = None
-        if args.memo:
-            # chain memo: {tag, value} 16B entries, 2^26 slots = 1 GiB HBM
-            memo = torch.zeros((1 << 26, 2), dtype=torch.int64,
-                               device=device)
-        call_cursor = [0]
 
     def chain_call(call_idx):
         """Hash-chain kernel for one call's prompts -> flat row-major
         request hashes."""
-        toks = call_tokens[call_idx % len(call_tokens)]
-        if memo is not None:
-            hashes_t = ops.gpu_hash_chain_memo(
-                toks, parents, nchunks_t, BLOCK_SIZE, KEYS_PER_PROMPT, memo)
-        else:
-            hashes_t = ops.gpu_hash_chain_tr(
-                toks, parents, nchunks_t, BLOCK_SIZE, KEYS_PER_PROMPT, 0)
+        hashes_t = ops.gpu_hash_chain_tr(
+            call_tokens[call_idx], parents, nchunks_t, BLOCK_SIZE,
+            KEYS_PER_PROMPT, 0)
         return hashes_t.t().contiguous().view(-1)
 
     def probe_score(hashes):
@@ -350,16 +335,14 @@ def main():
                 score_call_cpu(c)
                 lat.append(time.monotonic() - t0)
             return lat
-        base = call_cursor[0]
-        call_cursor[0] += args.calls_per_step
         t0 = time.monotonic()
-        hashes = chain_call(base)
+        hashes = chain_call(0)
         for c in range(args.calls_per_step):
             ev = None
             if c + 1 < args.calls_per_step:
                 chain_stream.wait_stream(torch.cuda.current_stream())
                 with torch.cuda.stream(chain_stream):
-                    next_hashes = chain_call(base + c + 1)
+                    next_hashes = chain_call(c + 1)
                 ev = torch.cuda.Event()
                 ev.record(chain_stream)
             probe_score(hashes)
@@ -456,8 +439,6 @@ def main():
                                          if single_ms is not None else None),
                 "batch_per_call": args.batch,
                 "ingest_blocks_per_sec": round(ingest_rate, 1),
-                "chain_memo": bool(getattr(args, "memo", False)
-                                   and use_gpu and sharded is None),
                 "timed_path": "hash-chain + probe + longest-prefix score "
                               "+ D2H + top-pod (tokenization excluded)",
             },

@@ -63,8 +63,6 @@ std::vector<at::Tensor> gpu_hash_chain(at::Tensor, at::Tensor, at::Tensor,
                                        int64_t);
 at::Tensor gpu_hash_chain_tr(at::Tensor, at::Tensor, at::Tensor, int64_t,
                              int64_t, int64_t);
-at::Tensor gpu_hash_chain_memo(at::Tensor, at::Tensor, at::Tensor, int64_t,
-                               int64_t, at::Tensor);
 void gpu_apply_events(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, int64_t, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, at::Tensor,
@@ -103,7 +101,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gpu_hash_chain_tr", &kvidx::gpu_hash_chain_tr,
         py::arg("tokens_t"), py::arg("parents"), py::arg("n_chunks"),
         py::arg("block_size"), py::arg("max_chunks"), py::arg("ilp") = 0);
-  m.def("gpu_hash_chain_memo", &kvidx::gpu_hash_chain_memo);
   m.def("gpu_apply_events", &kvidx::gpu_apply_events);
   m.def("gpu_apply_events_split", &kvidx::gpu_apply_events_split);
 #else

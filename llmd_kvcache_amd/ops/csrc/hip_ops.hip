@@ -489,64 +489,6 @@ __global__ void k_hash_chain_tr(const int32_t* __restrict__ tokens_t,  // [T,B]
   }
 }
 
-// Memoized chains: routing workloads repeat shared prefixes by
-// construction (that is the premise of prefix-cache routing), so chunk
-// hashes recur.  A direct-mapped HBM memo table caches
-// (parent, chunk-tokens) -> chain hash under a 64-bit mixed tag: a warm
-// chunk costs ONE 16-B load instead of the ~4.4k-cycle serial CBOR->FNV
-// chain; a cold chunk computes exactly as before and stores its entry
-// (last-writer-wins; the memo is a pure cache of deterministic values).
-// Tag collisions are 2^-64-scale - the same risk class the index itself
-// accepts for 64-bit content hashes.
-KVIDX_HD uint64_t memo_mix(uint64_t parent, const uint32_t* tok, int n) {
-  uint64_t k = parent ^ 0x9E3779B97F4A7C15ull;
-  for (int j = 0; j < n; ++j) {
-    k = (k ^ (uint64_t)tok[j]) * 0xFF51AFD7ED558CCDull;
-    k ^= k >> 33;
-  }
-  k *= 0xC4CEB9FE1A85EC53ull;
-  k ^= k >> 33;
-  return k == 0 ? 1 : k;  // 0 is the empty-tag sentinel
-}
-
-// Entries are {tag, value} packed in ONE aligned 16-byte ulonglong2 so
-// the load and the store are single dwordx4 transactions - concurrent
-// writers to a colliding slot replace the whole entry, never tearing a
-// tag from another writer's value.
-template <int BS>
-__global__ void k_hash_chain_memo(const int32_t* __restrict__ tokens_t,
-                                  const uint64_t* __restrict__ parents,
-                                  const int32_t* __restrict__ n_chunks,
-                                  int64_t B, int64_t L, int max_chunks,
-                                  ulonglong2* __restrict__ memo,  // [Cm]
-                                  int64_t mmask,
-                                  uint64_t* __restrict__ out) {
-  const int64_t lane = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (lane >= L) return;
-  uint64_t h = parents[lane];
-  const int mc = n_chunks[lane];
-  for (int c = 0; c < max_chunks; ++c) {
-    if (c >= mc) break;
-    uint32_t tok[BS];
-#pragma unroll
-    for (int j = 0; j < BS; ++j)
-      tok[j] = (uint32_t)tokens_t[(int64_t)(c * BS + j) * B + lane];
-    const uint64_t tag = memo_mix(h, tok, BS);
-    const int64_t slot = (int64_t)(tag & (uint64_t)mmask);
-    const ulonglong2 e = memo[slot];  // one 16B load
-    if (e.x == tag) {
-      h = e.y;  // warm: the load replaces the ~4.4k-cycle FNV chain
-    } else {
-      h = chunk_hash_fast(h, tok, BS);
-      ulonglong2 ne;
-      ne.x = tag;
-      ne.y = h;
-      memo[slot] = ne;  // one 16B store
-    }
-    out[(int64_t)c * B + lane] = h;
-  }
-}
-
 // Double-buffered single-chain variant: chunk c+1's coalesced loads are
 // issued BEFORE chunk c's hash chain, so HBM latency hides under the
 // serial ALU work (the plain variant exposes ~full load latency at each
@@ -970,43 +912,6 @@ std::vector<at::Tensor> gpu_hash_chain(at::Tensor tokens, at::Tensor tok_off,
                      (int)block_size,
                      reinterpret_cast<uint64_t*>(out.data_ptr<int64_t>()));
   return {out, chunk_off};
-}
-
-// Memoized-chain launcher: memo is an int64 [Cm, 2] tensor ({tag, val}
-// interleaved, zero-initialized; Cm a power of two).
-at::Tensor gpu_hash_chain_memo(at::Tensor tokens_t, at::Tensor parents,
-                               at::Tensor n_chunks, int64_t block_size,
-                               int64_t max_chunks, at::Tensor memo) {
-  TORCH_CHECK(tokens_t.is_cuda() && tokens_t.dtype() == at::kInt);
-  TORCH_CHECK(tokens_t.dim() == 2, "tokens_t must be [T, B]");
-  TORCH_CHECK(memo.dim() == 2 && memo.size(1) == 2 &&
-                  memo.dtype() == at::kLong,
-              "memo must be int64 [Cm, 2]");
-  int64_t Cm = memo.size(0);
-  TORCH_CHECK((Cm & (Cm - 1)) == 0, "memo capacity must be a power of two");
-  int64_t B = tokens_t.size(1);
-  auto out = at::zeros({max_chunks, B}, parents.options());
-  if (B == 0 || max_chunks == 0) return out;
-  int threads = 256;
-  int blocks = (int)((B + threads - 1) / threads);
-  auto launch = [&](auto kern) {
-    hipLaunchKernelGGL(kern, dim3(blocks), dim3(threads), 0, STREAM,
-                       tokens_t.data_ptr<int32_t>(), U64P(parents),
-                       n_chunks.data_ptr<int32_t>(), B, B, (int)max_chunks,
-                       reinterpret_cast<ulonglong2*>(
-                           memo.data_ptr<int64_t>()),
-                       Cm - 1,
-                       reinterpret_cast<uint64_t*>(out.data_ptr<int64_t>()));
-  };
-  switch (block_size) {
-    case 16: launch(k_hash_chain_memo<16>); break;
-    case 32: launch(k_hash_chain_memo<32>); break;
-    case 4:  launch(k_hash_chain_memo<4>);  break;
-    case 8:  launch(k_hash_chain_memo<8>);  break;
-    default:
-      TORCH_CHECK(false, "hash_chain_memo supports block sizes 4/8/16/32");
-  }
-  return out;
 }
 
 // tokens_t: int32 [T, B] (transposed); parents int64 [B]; n_chunks

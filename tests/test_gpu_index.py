@@ -424,51 +424,3 @@ class TestLdsOverflowFallback:
         maps = gpu.scores_to_map(
             gpu.fused_scores(hashes, offs, MODEL, set(), max_k=K))
         assert maps[0] == {"pod-2000": float(K)}
-
-
-class TestHashChainMemoKernel:
-    def test_memo_identical_to_plain_cold_and_warm(self):
-        from llmd_kvcache_amd.ops import cpu_ext
-
-        mod = cpu_ext.require()
-        rng = random.Random(17)
-        B, K, BS = 257, 12, 16
-        tok = torch.randint(0, 1 << 31, (K * BS, B), dtype=torch.int32,
-                            device="cuda")
-        parents = torch.full((B,), _to_i64(hashing.init_hash("")),
-                             dtype=torch.int64, device="cuda")
-        nch = torch.full((B,), K, dtype=torch.int32, device="cuda")
-        memo = torch.zeros((1 << 18, 2), dtype=torch.int64, device="cuda")
-        ref = mod.gpu_hash_chain_tr(tok, parents, nch, BS, K, 0)
-        cold = mod.gpu_hash_chain_memo(tok, parents, nch, BS, K, memo)
-        warm = mod.gpu_hash_chain_memo(tok, parents, nch, BS, K, memo)
-        torch.cuda.synchronize()
-        assert torch.equal(cold, ref)
-        assert torch.equal(warm, ref)
-        # table actually has entries now
-        assert int((memo[:, 0] != 0).sum()) > 0
-
-    def test_memo_shared_prefix_reuse(self):
-        """Prompts sharing a prefix fill/hit the same entries; divergent
-        tails still hash correctly."""
-        from llmd_kvcache_amd.ops import cpu_ext
-
-        mod = cpu_ext.require()
-        B, K, BS = 128, 8, 16
-        shared = torch.randint(0, 1 << 31, (4 * BS, 1), dtype=torch.int32)
-        tok = torch.empty((K * BS, B), dtype=torch.int32)
-        tok[: 4 * BS] = shared  # same 4-chunk prefix for every prompt
-        tok[4 * BS:] = torch.randint(0, 1 << 31, ((K - 4) * BS, B),
-                                     dtype=torch.int32)
-        tok = tok.cuda()
-        parents = torch.full((B,), _to_i64(hashing.init_hash("")),
-                             dtype=torch.int64, device="cuda")
-        nch = torch.full((B,), K, dtype=torch.int32, device="cuda")
-        memo = torch.zeros((1 << 16, 2), dtype=torch.int64, device="cuda")
-        ref = mod.gpu_hash_chain_tr(tok, parents, nch, BS, K, 0)
-        out = mod.gpu_hash_chain_memo(tok, parents, nch, BS, K, memo)
-        torch.cuda.synchronize()
-        assert torch.equal(out, ref)
-        # shared prefix chunks collapse to 4 entries + B distinct tails
-        n_entries = int((memo[:, 0] != 0).sum())
-        assert n_entries <= 4 + (K - 4) * B
