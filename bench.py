@@ -154,6 +154,9 @@ def main():
     ap.add_argument("--blocks", type=int, default=NUM_BLOCKS)
     ap.add_argument("--graph", action="store_true",
                     help="capture the read call in a hipGraph and replay")
+    ap.add_argument("--force-sharded", action="store_true",
+                    help="use the ShardedIndex/RCCL path even at world=1 "
+                         "(single-GPU validation of the collective path)")
     args = ap.parse_args()
     NUM_BLOCKS = args.blocks
 
@@ -174,9 +177,21 @@ def main():
         backend = "nccl" if use_gpu else "gloo"
         dist.init_process_group(backend)
 
+    if args.force_sharded and world == 1 and dist is None:
+        import torch.distributed as tdist
+
+        dist = tdist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29517")
+        os.environ.setdefault("RANK", "0")
+        os.environ.setdefault("WORLD_SIZE", "1")
+        dist.init_process_group("nccl" if use_gpu else "gloo",
+                                rank=0, world_size=1)
+        world = 1
+
     # power-of-two capacity at <=0.5 load factor for the requested blocks
     capacity = 1 << max(22, (NUM_BLOCKS * 2 - 1).bit_length())
-    if world > 1:
+    if world > 1 or (args.force_sharded and dist is not None):
         from llmd_kvcache_amd.parallel.sharded import ShardedIndex
 
         cfg = TableIndexConfig(capacity=capacity, pods_per_key=10,
