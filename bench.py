@@ -21,7 +21,11 @@ ingest rate as config.ingest_blocks_per_sec.
 Usage (driver contract):
   python bench.py --gpus N --steps K --warmup W
 N>1 is launched by the driver via torch.distributed.run (one rank per
-GPU over RCCL); per-GPU work is fixed (weak scaling).
+GPU over RCCL). Default multi-GPU mode REPLICATES the index (it fits in
+one MI355X's HBM by orders of magnitude) and gives each rank its own
+request stream - per-GPU work fixed, weak scaling.  --sharded switches
+to hash%%N ownership with RCCL mask merges (capacity mode, BASELINE
+config 3, strong scaling).
 """
 
 import argparse
@@ -157,6 +161,10 @@ def main():
     ap.add_argument("--force-sharded", action="store_true",
                     help="use the ShardedIndex/RCCL path even at world=1 "
                          "(single-GPU validation of the collective path)")
+    ap.add_argument("--sharded", action="store_true",
+                    help="N>1: shard the index by hash%%N with RCCL mask "
+                         "merge (capacity mode, BASELINE config 3) instead "
+                         "of the default full-replication mode")
     args = ap.parse_args()
     NUM_BLOCKS = args.blocks
 
@@ -191,7 +199,16 @@ def main():
 
     # power-of-two capacity at <=0.5 load factor for the requested blocks
     capacity = 1 << max(22, (NUM_BLOCKS * 2 - 1).bit_length())
-    if world > 1 or (args.force_sharded and dist is not None):
+    # Multi-GPU modes:
+    #  - default (replicated): the index FITS in one MI355X's 288 GB HBM by
+    #    orders of magnitude, so each rank holds the full index (the same
+    #    replicated event stream every rank already consumes) and scores
+    #    its own slice of the request load - no per-request collectives,
+    #    linear scaling ("weak": per-GPU work fixed).
+    #  - --sharded: hash%%N ownership + RCCL mask all_reduce per batch
+    #    (BASELINE config 3; for indexes beyond one GPU's memory).
+    use_sharded = (args.sharded or args.force_sharded) and dist is not None
+    if use_sharded:
         from llmd_kvcache_amd.parallel.sharded import ShardedIndex
 
         cfg = TableIndexConfig(capacity=capacity, pods_per_key=10,
@@ -218,7 +235,8 @@ def main():
     chains, ingest_rate = populate_index(index, device, rank)
 
     n_prompts = args.batch * args.calls_per_step
-    tokens, tok_offsets = build_prompts(chains, n_prompts, device, seed=99)
+    tokens, tok_offsets = build_prompts(chains, n_prompts, device,
+                                        seed=99 + (0 if sharded else rank))
     tp_init = _to_i64(ChunkedTokenDatabase(
         TokenProcessorConfig(block_size=BLOCK_SIZE)).config.init_hash())
     parents = torch.full((args.batch,), tp_init, dtype=torch.int64,
@@ -420,10 +438,17 @@ def main():
             lat1.append(time.monotonic() - t0)
         single_ms = statistics.median(lat1) * 1000.0
 
-    # strong scaling: all ranks score the same global prompt stream
-    # cooperatively (index sharded N ways, probes split by ownership,
-    # masks merged over RCCL); total work is fixed as N grows.
-    prompts_per_step = n_prompts
+    if sharded is not None:
+        # strong scaling: all ranks score the SAME global prompt stream
+        # cooperatively (index sharded N ways, probes split by ownership,
+        # masks merged over RCCL); total work is fixed as N grows.
+        prompts_per_step = n_prompts
+        scaling = "strong"
+    else:
+        # replicated: each rank scores its own distinct stream; per-GPU
+        # work fixed as N grows (weak scaling), whole-job QPS aggregates.
+        prompts_per_step = n_prompts * max(world, 1)
+        scaling = "weak"
     total_prompts = prompts_per_step * args.steps
     qps = total_prompts / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
@@ -439,7 +464,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": round(ms_per_step, 3),
             "higher_is_better": True,
-            "scaling": "strong",
+            "scaling": scaling,
             "vs_baseline": None,
             "dtype": "int64-hash/fp32-score",
             "data": "synthetic",
@@ -447,7 +472,9 @@ def main():
                 "model": MODEL,
                 "global_batch": prompts_per_step,
                 "seq_len": PROMPT_TOKENS,
-                "parallelism": f"shard{world}" if world > 1 else "single",
+                "parallelism": (f"shard{world}" if sharded is not None
+                                else f"replicated{world}" if world > 1
+                                else "single"),
                 "index_blocks": NUM_BLOCKS,
                 "num_pods": NUM_PODS,
                 "block_size": BLOCK_SIZE,
