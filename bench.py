@@ -156,6 +156,8 @@ def main():
     ap.add_argument("--calls-per-step", type=int, default=2)
     ap.add_argument("--device", default=None, help="cpu to force CPU tables")
     ap.add_argument("--blocks", type=int, default=NUM_BLOCKS)
+    ap.add_argument("--pods", type=int, default=NUM_PODS,
+                    help="fleet size (config 5 scale: 256+)")
     ap.add_argument("--graph", action="store_true",
                     help="capture the read call in a hipGraph and replay")
     ap.add_argument("--force-sharded", action="store_true",
@@ -167,6 +169,8 @@ def main():
                          "of the default full-replication mode")
     args = ap.parse_args()
     NUM_BLOCKS = args.blocks
+    global NUM_PODS
+    NUM_PODS = args.pods
 
     rank = int(os.environ.get("RANK", 0))
     world = int(os.environ.get("WORLD_SIZE", 1))

@@ -72,8 +72,8 @@ def _recv_frame(sock: socket.socket) -> Tuple[int, bytes]:
     flags = _recv_exact(sock, 1)[0]
     if flags & FLAG_LONG:
         size = struct.unpack(">Q", _recv_exact(sock, 8))[0]
-        if size > (1 << 31):
-            raise ConnectionError(f"oversized ZMTP frame: {size}")
+        if size > (1 << 28):  # 256 MiB cap: untrusted peers must not be
+            raise ConnectionError(f"oversized ZMTP frame: {size}")  # able to force huge allocations
     else:
         size = _recv_exact(sock, 1)[0]
     body = _recv_exact(sock, size) if size else b""
@@ -347,7 +347,10 @@ class PubSocket:
                 except socket.timeout:
                     continue
                 if flags & FLAG_COMMAND:
-                    name, _props = _parse_command(body)
+                    try:
+                        name, _props = _parse_command(body)
+                    except Exception:
+                        continue  # malformed command from peer: ignore
                     if name == "SUBSCRIBE":  # ZMTP 3.1 form
                         body = b"\x01" + body[1 + len(b"SUBSCRIBE") :]
                     elif name == "CANCEL":
