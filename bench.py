@@ -146,7 +146,7 @@ def build_prompts(chains, n_prompts, device, seed):
 
 
 def main():
-    global NUM_BLOCKS
+    global NUM_BLOCKS, NUM_PODS
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
@@ -169,7 +169,6 @@ def main():
                          "of the default full-replication mode")
     args = ap.parse_args()
     NUM_BLOCKS = args.blocks
-    global NUM_PODS
     NUM_PODS = args.pods
 
     rank = int(os.environ.get("RANK", 0))
