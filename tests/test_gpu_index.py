@@ -424,3 +424,26 @@ class TestLdsOverflowFallback:
         maps = gpu.scores_to_map(
             gpu.fused_scores(hashes, offs, MODEL, set(), max_k=K))
         assert maps[0] == {"pod-2000": float(K)}
+
+
+class TestHighLoadFactor:
+    def test_steal_path_under_pressure(self):
+        """GPU twin of the CPU high-load test plus concurrent duplicate
+        pressure: fill to 85%, then hammer the hottest window."""
+        cap = 1 << 14
+        gpu = GpuIndex(GpuIndexConfig(capacity=cap, pods_per_key=4))
+        n = int(cap * 0.85)
+        eh = torch.arange(1, n + 1, dtype=torch.int64, device="cuda")
+        rh = eh + 5_000_000
+        pe = gpu._entries_tensor([PodEntry("pod-a", "gpu")])
+        gpu.table.insert(eh, rh, gpu.registry.model_id(MODEL), pe)
+        torch.cuda.synchronize()
+        recent = [Key(MODEL, 5_000_000 + i) for i in range(n - 500, n)]
+        found = gpu.lookup(recent, set())
+        assert len(found) >= len(recent) * 0.95
+        # duplicate hammering on a full table must not corrupt
+        dup_eh = torch.full((4096,), n, dtype=torch.int64, device="cuda")
+        dup_rh = dup_eh + 5_000_000
+        gpu.table.insert(dup_eh, dup_rh, gpu.registry.model_id(MODEL), pe)
+        torch.cuda.synchronize()
+        assert gpu.lookup([Key(MODEL, 5_000_000 + n)], set())

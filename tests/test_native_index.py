@@ -209,3 +209,20 @@ def test_checkpoint_restore(tmp_path):
     wrong = NativeIndex(TableIndexConfig(capacity=1 << 11, pods_per_key=4))
     with pytest.raises(ValueError):
         wrong.load(path)
+
+
+def test_high_load_factor_inserts_remain_findable():
+    """At 85%+ occupancy the probe-window steal path engages; recent
+    inserts must remain findable and lookups must not degrade into
+    misses for freshly-added keys."""
+    cap = 1 << 12
+    nat = NativeIndex(TableIndexConfig(capacity=cap, pods_per_key=4))
+    n = int(cap * 0.85)
+    keys = [Key(MODEL, 1_000_000 + i) for i in range(n)]
+    for lo in range(0, n, 512):
+        ks = keys[lo:lo + 512]
+        nat.add(ks, ks, [PodEntry("pod-a", "gpu")])
+    # the most recent 10% must be present (approx-LRU steals old slots)
+    recent = keys[int(n * 0.9):]
+    found = nat.lookup(recent, set())
+    assert len(found) >= len(recent) * 0.95
