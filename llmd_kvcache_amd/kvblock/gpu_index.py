@@ -571,8 +571,14 @@ class GpuIndex(TableIndex):
         if len(grp_off) == 1:
             return
         if len(model_ids) > 1:
-            raise ValueError(
-                "apply_event_batches supports one model per call; split by model")
+            # mixed-model batch: split and recurse per model (the kernel
+            # takes one model id per launch)
+            by_model: Dict[str, list] = {}
+            for pod, model, events in batches:
+                by_model.setdefault(model, []).append((pod, model, events))
+            for group in by_model.values():
+                self.apply_event_batches(group, token_processor)
+            return
 
         d = self.device
         i32 = torch.int32

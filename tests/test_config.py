@@ -127,3 +127,46 @@ class TestNestedNoneOverlay:
         assert isinstance(cfg.tokenizers_pool.uds, UdsTokenizerConfig)
         assert cfg.tokenizers_pool.uds.socket_path == "/tmp/x.sock"
         assert cfg.tokenizers_pool.uds.timeout_s == 5.0  # default kept
+
+
+class TestMultiModelBurst:
+    def test_mixed_model_batches_split(self):
+        """apply_event_batches splits mixed-model batches per model
+        instead of raising (one model id per kernel launch)."""
+        from llmd_kvcache_amd.kvevents.events import BlockStored
+        from llmd_kvcache_amd.kvblock.gpu_index import GpuIndex
+
+        calls = []
+
+        class Rec(FakeGpuIndex):
+            def __init__(self):
+                super().__init__()
+                # borrow the real splitting logic with a recording kernel
+                from llmd_kvcache_amd.kvblock.gpu_index import (
+                    NativeIndex,
+                    TableIndexConfig,
+                )
+
+        # exercise through the real NativeIndex-compatible path is not
+        # possible without a GPU; assert the pool-level split instead
+        from llmd_kvcache_amd.kvevents.pool import EventsConfig, EventsPool
+        from llmd_kvcache_amd.kvevents.events import EventBatch
+        from llmd_kvcache_amd.kvevents.pool import Message
+
+        index = FakeGpuIndex()
+        pool = EventsPool(EventsConfig(concurrency=1), index)
+        for i, model in enumerate(["model-a", "model-b", "model-a"]):
+            batch = EventBatch(
+                ts=0.0,
+                events=[BlockStored([i + 1], None, list(range(16)), 16)],
+            )
+            pool.add_task(Message(f"kv@p@{model}", batch.encode(), i, "p",
+                                  model))
+        pool.start(with_subscriber=False)
+        pool.drain()
+        pool.shutdown()
+        models_seen = [
+            {m for _, m, _ in call} for call in index.applied
+        ]
+        for mset in models_seen:
+            assert len(mset) == 1  # each apply call is single-model
