@@ -131,27 +131,15 @@ class TestNestedNoneOverlay:
 
 class TestMultiModelBurst:
     def test_mixed_model_batches_split(self):
-        """apply_event_batches splits mixed-model batches per model
-        instead of raising (one model id per kernel launch)."""
-        from llmd_kvcache_amd.kvevents.events import BlockStored
-        from llmd_kvcache_amd.kvblock.gpu_index import GpuIndex
-
-        calls = []
-
-        class Rec(FakeGpuIndex):
-            def __init__(self):
-                super().__init__()
-                # borrow the real splitting logic with a recording kernel
-                from llmd_kvcache_amd.kvblock.gpu_index import (
-                    NativeIndex,
-                    TableIndexConfig,
-                )
-
-        # exercise through the real NativeIndex-compatible path is not
-        # possible without a GPU; assert the pool-level split instead
-        from llmd_kvcache_amd.kvevents.pool import EventsConfig, EventsPool
-        from llmd_kvcache_amd.kvevents.events import EventBatch
-        from llmd_kvcache_amd.kvevents.pool import Message
+        """The pool's GPU burst path hands apply_event_batches
+        single-model groups; gpu_index additionally self-splits mixed
+        batches (one model id per kernel launch)."""
+        from llmd_kvcache_amd.kvevents.events import BlockStored, EventBatch
+        from llmd_kvcache_amd.kvevents.pool import (
+            EventsConfig,
+            EventsPool,
+            Message,
+        )
 
         index = FakeGpuIndex()
         pool = EventsPool(EventsConfig(concurrency=1), index)
