@@ -281,7 +281,8 @@ void cpu_insert(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
   TORCH_CHECK(rh.numel() == n, "engine/request key length mismatch");
   for (int64_t i = 0; i < n; ++i) {
     emap_put(v, ehp[i], (uint32_t)model_id, remap_hash(rhp[i]), (int32_t)epoch);
-    if (num_shards > 1 && (int64_t)(rhp[i] % (uint64_t)num_shards) != shard_id)
+    if (num_shards > 1 &&
+        (int64_t)(remap_hash(rhp[i]) % (uint64_t)num_shards) != shard_id)
       continue;
     int64_t slot = table_put(v, rhp[i], (uint32_t)model_id, (int32_t)epoch);
     for (int64_t j = 0; j < m; ++j)

@@ -235,7 +235,8 @@ __global__ void k_insert(DevTable v, const uint64_t* __restrict__ eh,
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   dev_emap_put(v, eh[i], model, remap_hash(rh[i]));
-  if (num_shards > 1 && (int)(rh[i] % (uint64_t)num_shards) != shard_id)
+  if (num_shards > 1 &&
+      (int)(remap_hash(rh[i]) % (uint64_t)num_shards) != shard_id)
     return;  // engine map replicated; main table sharded by ownership
   int64_t slot = dev_table_put(v, rh[i], model, epoch);
   for (int j = 0; j < n_entries; ++j)
