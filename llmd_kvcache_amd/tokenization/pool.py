@@ -18,6 +18,7 @@ from __future__ import annotations
 
 import queue
 import threading
+import time
 from dataclasses import dataclass, field
 from typing import List, Optional
 
@@ -53,6 +54,7 @@ class _Task:
     prompt: str
     model_name: str
     result: Optional["queue.Queue"]  # None => fire-and-forget
+    attempts: int = 0
 
 
 class TokenizationPool:
@@ -143,6 +145,8 @@ class TokenizationPool:
                 return
             self._process(task)
 
+    MAX_ASYNC_RETRIES = 2
+
     def _process(self, task: _Task) -> None:
         try:
             tokens = self._process_task(task)
@@ -151,6 +155,17 @@ class TokenizationPool:
         except Exception as e:
             if task.result is not None:
                 task.result.put(("err", e))
+            elif task.attempts < self.MAX_ASYNC_RETRIES:
+                # fire-and-forget tasks are re-queued with backoff, the
+                # reference's rate-limited retry (pool.go:187-191)
+                task.attempts += 1
+
+                def requeue():
+                    time.sleep(0.05 * (2 ** task.attempts))
+                    if self._running:
+                        self._queue.put(task)
+
+                threading.Thread(target=requeue, daemon=True).start()
 
     def _process_task(self, task: _Task) -> List[int]:
         prompt = task.prompt

@@ -305,3 +305,47 @@ class TestUdsService:
         with pytest.raises(TokenizationError):
             client.encode("x", "m")
         assert time.monotonic() - t0 >= 0.01  # backoff happened
+
+
+class FlakyTokenizer(StaticTokenizer):
+    """Fails the first N encodes, then succeeds."""
+
+    def __init__(self, fail_times):
+        self.fail_times = fail_times
+        self.calls = 0
+
+    def encode(self, prompt, model_name, add_special_tokens=True):
+        self.calls += 1
+        if self.calls <= self.fail_times:
+            raise TokenizationError("transient failure")
+        return super().encode(prompt, model_name, add_special_tokens)
+
+
+class TestAsyncRetry:
+    def test_fire_and_forget_retries_with_backoff(self):
+        tok = FlakyTokenizer(fail_times=1)
+        pool = TokenizationPool(
+            TokenizationConfig(workers_count=1), tokenizer=tok
+        )
+        pool.run()
+        try:
+            pool.enqueue_tokenization(None, "abcd" * 16, "m")
+            deadline = time.monotonic() + 5
+            while tok.calls < 2 and time.monotonic() < deadline:
+                time.sleep(0.02)
+            assert tok.calls >= 2  # retried after the transient failure
+        finally:
+            pool.shutdown()
+
+    def test_sync_does_not_retry(self):
+        tok = FlakyTokenizer(fail_times=1)
+        pool = TokenizationPool(
+            TokenizationConfig(workers_count=1), tokenizer=tok
+        )
+        pool.run()
+        try:
+            with pytest.raises(TokenizationError):
+                pool.tokenize(None, "abcd" * 16, "m")
+            assert tok.calls == 1
+        finally:
+            pool.shutdown()
