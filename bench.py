@@ -123,21 +123,22 @@ def _apply_cpu(index, batch, tp):
                       [PodEntry(pod, "gpu")])
 
 
-def build_prompts(chains, n_prompts, device, seed):
+def build_prompts(chains, n_prompts, device, seed, prefix_frac=0.5):
     """Prompts = one stored chain (shared prefix, hits) + fresh random
     tail (misses) - the shared-prefix routing workload of the reference
     benchmarks (benchmarking/37-capacity: 8k shared prefix)."""
     import numpy as np
 
     rng = np.random.default_rng(seed)
-    reuse = PROMPT_TOKENS // 2
+    reuse = int(PROMPT_TOKENS * prefix_frac) // BLOCK_SIZE * BLOCK_SIZE
     all_tokens = np.empty((n_prompts, PROMPT_TOKENS), dtype=np.int64)
     for i in range(n_prompts):
         chain = chains[rng.integers(len(chains))]
-        prefix = np.tile(chain, reuse // len(chain) + 1)[:reuse]
+        if reuse:
+            prefix = np.tile(chain, reuse // len(chain) + 1)[:reuse]
+            all_tokens[i, :reuse] = prefix
         tail = rng.integers(0, 1 << 31, size=PROMPT_TOKENS - reuse,
                             dtype=np.int64)
-        all_tokens[i, :reuse] = prefix
         all_tokens[i, reuse:] = tail
     t = torch.from_numpy(all_tokens.reshape(-1)).to(device)
     offsets = torch.arange(0, (n_prompts + 1) * PROMPT_TOKENS, PROMPT_TOKENS,
@@ -158,6 +159,9 @@ def main():
     ap.add_argument("--blocks", type=int, default=NUM_BLOCKS)
     ap.add_argument("--pods", type=int, default=NUM_PODS,
                     help="fleet size (config 5 scale: 256+)")
+    ap.add_argument("--prefix-frac", type=float, default=0.5,
+                    help="fraction of each prompt that is an index-resident "
+                         "shared prefix (hit ratio of the workload)")
     ap.add_argument("--graph", action="store_true",
                     help="capture the read call in a hipGraph and replay")
     ap.add_argument("--force-sharded", action="store_true",
@@ -239,7 +243,8 @@ def main():
 
     n_prompts = args.batch * args.calls_per_step
     tokens, tok_offsets = build_prompts(chains, n_prompts, device,
-                                        seed=99 + (0 if sharded else rank))
+                                        seed=99 + (0 if sharded else rank),
+                                        prefix_frac=args.prefix_frac)
     tp_init = _to_i64(ChunkedTokenDatabase(
         TokenProcessorConfig(block_size=BLOCK_SIZE)).config.init_hash())
     parents = torch.full((args.batch,), tp_init, dtype=torch.int64,
@@ -486,6 +491,7 @@ def main():
                 "p50_single_prompt_ms": (round(single_ms, 3)
                                          if single_ms is not None else None),
                 "batch_per_call": args.batch,
+                "prefix_frac": args.prefix_frac,
                 "ingest_blocks_per_sec": round(ingest_rate, 1),
                 "timed_path": "hash-chain + probe + longest-prefix score "
                               "+ D2H + top-pod (tokenization excluded)",
