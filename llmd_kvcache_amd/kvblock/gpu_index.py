@@ -284,7 +284,7 @@ class TableIndex(Index):
         filt = self._filter_tensor(pod_identifier_set, num_pods)
         found, masks = self.table.lookup(hashes, model_id, filt, num_pods)
         found = found.cpu().tolist()
-        masks = masks.cpu()
+        masks_l = masks.cpu().tolist()  # one bulk D2H/convert, no per-item
         W = masks.shape[2]
 
         result: Dict[Key, List[PodEntry]] = {}
@@ -301,7 +301,7 @@ class TableIndex(Index):
                 if tier_name is None:
                     continue
                 for w in range(W):
-                    bits = _to_u64(int(masks[i, t, w]))
+                    bits = _to_u64(masks_l[i][t][w])
                     while bits:
                         b = (bits & -bits).bit_length() - 1
                         bits &= bits - 1
