@@ -165,3 +165,39 @@ class TestHttpService:
                 assert json.loads(resp.read())["status"] == "ok"
         finally:
             svc.stop()
+
+
+class TestGrpcErrorPath:
+    def test_internal_error_maps_to_grpc_status(self):
+        import grpc
+
+        from llmd_kvcache_amd.service.grpc_server import IndexerClient, serve
+
+        class BoomIndexer:
+            def get_pod_scores(self, *a, **kw):
+                raise RuntimeError("index exploded")
+
+        server = serve(BoomIndexer(), address="127.0.0.1:0")
+        try:
+            client = IndexerClient(f"127.0.0.1:{server._kvidx_port}")
+            with pytest.raises(grpc.RpcError) as ei:
+                client.get_pod_scores("x", "m")
+            assert ei.value.code() == grpc.StatusCode.INTERNAL
+            client.close()
+        finally:
+            server.stop(None)
+
+    def test_empty_scores_ok(self):
+        from llmd_kvcache_amd.service.grpc_server import IndexerClient, serve
+
+        class EmptyIndexer:
+            def get_pod_scores(self, *a, **kw):
+                return {}
+
+        server = serve(EmptyIndexer(), address="127.0.0.1:0")
+        try:
+            client = IndexerClient(f"127.0.0.1:{server._kvidx_port}")
+            assert client.get_pod_scores("x", "m") == {}
+            client.close()
+        finally:
+            server.stop(None)
