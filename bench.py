@@ -55,6 +55,7 @@ KEYS_PER_PROMPT = PROMPT_TOKENS // BLOCK_SIZE
 NUM_PODS = 64
 NUM_BLOCKS = 1 << 20     # ~1M blocks resident
 MODEL = "meta-llama/Llama-3.1-8B-Instruct"
+VOCAB = 128256           # Llama-3 vocabulary: token ids are < 128256
 BLOCKS_PER_EVENT = 64    # one BlockStored covers 64 blocks (1024 tokens)
 
 
@@ -86,7 +87,7 @@ def populate_index(index, device, rank):
     engine_hash = 1
     batch = []
     for e in range(n_events):
-        tokens = rng.integers(0, 1 << 31, size=BLOCKS_PER_EVENT * BLOCK_SIZE,
+        tokens = rng.integers(0, VOCAB, size=BLOCKS_PER_EVENT * BLOCK_SIZE,
                               dtype=np.int64)
         hashes = np.arange(engine_hash, engine_hash + BLOCKS_PER_EVENT,
                            dtype=np.uint64)
@@ -137,7 +138,7 @@ def build_prompts(chains, n_prompts, device, seed, prefix_frac=0.5):
         if reuse:
             prefix = np.tile(chain, reuse // len(chain) + 1)[:reuse]
             all_tokens[i, :reuse] = prefix
-        tail = rng.integers(0, 1 << 31, size=PROMPT_TOKENS - reuse,
+        tail = rng.integers(0, VOCAB, size=PROMPT_TOKENS - reuse,
                             dtype=np.int64)
         all_tokens[i, reuse:] = tail
     t = torch.from_numpy(all_tokens.reshape(-1)).to(device)
@@ -485,6 +486,7 @@ def main():
                                 else "single"),
                 "index_blocks": NUM_BLOCKS,
                 "num_pods": NUM_PODS,
+                "vocab": VOCAB,
                 "block_size": BLOCK_SIZE,
                 "keys_per_prompt": KEYS_PER_PROMPT,
                 "p50_batch_latency_ms": round(p50_ms, 3),
