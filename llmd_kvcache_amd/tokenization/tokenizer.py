@@ -53,40 +53,48 @@ class Tokenizer:
         raise NotImplementedError
 
 
+class _Call:
+    __slots__ = ("ev", "status", "payload")
+
+    def __init__(self) -> None:
+        self.ev = threading.Event()
+        self.status = "err"
+        self.payload: object = TokenizationError("load failed")
+
+
 class _SingleFlight:
-    """Deduplicates concurrent loads of the same key (tokenizer.go:350-371)."""
+    """Deduplicates concurrent loads of the same key (tokenizer.go:350-371).
+    Each in-flight key owns a _Call object that every waiter reads from
+    after its event fires - no shared-map pop races."""
 
     def __init__(self) -> None:
         self._mu = threading.Lock()
-        self._calls: Dict[str, threading.Event] = {}
-        self._results: Dict[str, object] = {}
+        self._calls: Dict[str, _Call] = {}
 
     def do(self, key: str, fn):
         with self._mu:
-            ev = self._calls.get(key)
-            if ev is None:
-                ev = threading.Event()
-                self._calls[key] = ev
-                leader = True
-            else:
-                leader = False
+            call = self._calls.get(key)
+            leader = call is None
+            if leader:
+                call = _Call()
+                self._calls[key] = call
         if leader:
             try:
-                result = fn()
-                self._results[key] = ("ok", result)
+                call.payload = fn()
+                call.status = "ok"
             except Exception as e:  # propagate to all waiters
-                self._results[key] = ("err", e)
+                call.payload = e
+                call.status = "err"
             finally:
-                ev.set()
                 with self._mu:
                     self._calls.pop(key, None)
-            status, payload = self._results.pop(key)
+                call.ev.set()
         else:
-            ev.wait()
-            status, payload = self._results.get(key, ("err", TokenizationError("load failed")))
-        if status == "err":
-            raise payload if isinstance(payload, Exception) else TokenizationError(str(payload))
-        return payload
+            call.ev.wait()
+        if call.status == "err":
+            e = call.payload
+            raise e if isinstance(e, Exception) else TokenizationError(str(e))
+        return call.payload
 
 
 def discover_local_tokenizers(

@@ -349,3 +349,43 @@ class TestAsyncRetry:
             assert tok.calls == 1
         finally:
             pool.shutdown()
+
+
+class TestSingleFlight:
+    def test_concurrent_loads_deduplicate(self):
+        from llmd_kvcache_amd.tokenization.tokenizer import _SingleFlight
+
+        sf = _SingleFlight()
+        calls = []
+        results = []
+        errors = []
+
+        def slow_load():
+            calls.append(1)
+            time.sleep(0.1)
+            return "loaded"
+
+        def worker():
+            try:
+                results.append(sf.do("k", slow_load))
+            except Exception as e:  # pragma: no cover
+                errors.append(e)
+
+        threads = [threading.Thread(target=worker) for _ in range(16)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        assert errors == []
+        assert results == ["loaded"] * 16
+        assert len(calls) == 1  # exactly one real load
+
+    def test_leader_failure_propagates_then_retries(self):
+        from llmd_kvcache_amd.tokenization.tokenizer import _SingleFlight
+
+        sf = _SingleFlight()
+        with pytest.raises(TokenizationError):
+            sf.do("k", lambda: (_ for _ in ()).throw(
+                TokenizationError("boom")))
+        # a later call is a fresh flight
+        assert sf.do("k", lambda: 42) == 42
