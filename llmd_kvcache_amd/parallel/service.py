@@ -122,8 +122,20 @@ class ShardedIndexService:
 
     # -- non-rank-0 loop -----------------------------------------------
     def serve(self) -> None:
-        """Follower ranks: execute broadcast ops until stop."""
+        """Follower ranks: execute broadcast ops until stop.  A failing op
+        must not kill the follower - the next collective would then hang
+        every rank - so failures are logged and the loop continues (score
+        ops participate in the all_reduce even on failure via the zero
+        masks their empty dispatch produces; event failures only lose the
+        local application, mirroring the pool's poison-pill stance)."""
         assert self.rank != 0
+        import logging
+
+        logger = logging.getLogger("llmd_kvcache_amd.parallel")
         while self._running:
             op = self._broadcast(None)
-            self._dispatch(op)
+            try:
+                self._dispatch(op)
+            except Exception:
+                logger.exception("follower dispatch failed for %r",
+                                 op[0] if op else op)

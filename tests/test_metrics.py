@@ -71,3 +71,38 @@ class TestInstrumentedIndex:
         idx, inner = self.make()
         # decorator exposes the inner backend's extra attributes
         assert idx._pod_cache_size == inner._pod_cache_size
+
+
+class TestInstrumentedOverTableIndex:
+    def test_decorator_transparent_for_fast_paths(self):
+        """InstrumentedIndex must forward the table-backend fast-path
+        surface (table, fused_scores, apply_event_batches detection) so
+        the events pool and Indexer keep their native paths."""
+        import pytest
+
+        torch = pytest.importorskip("torch")
+        from llmd_kvcache_amd.kvblock.gpu_index import (
+            NativeIndex,
+            TableIndexConfig,
+        )
+        from llmd_kvcache_amd.ops import cpu_ext
+
+        if cpu_ext.maybe_load() is None:
+            pytest.skip("native extension not built")
+        inner = NativeIndex(TableIndexConfig(capacity=1 << 10, pods_per_key=4))
+        idx = InstrumentedIndex(inner)
+        # counters move through the decorator
+        a0 = collector.admissions._value.get()
+        keys = [k(100), k(101)]
+        idx.add(keys, keys, [PodEntry("pod-a", "gpu")])
+        assert collector.admissions._value.get() == a0 + 2
+        # fast-path surface forwarded
+        assert idx.table is inner.table
+        assert idx.tier_weights() is not None
+        from llmd_kvcache_amd.kvblock.gpu_index import _to_i64
+
+        hashes = torch.tensor([_to_i64(x.chunk_hash) for x in keys],
+                              dtype=torch.int64)
+        counts = torch.tensor([2], dtype=torch.int32)
+        maps = idx.scores_to_map(idx.fused_scores(hashes, counts, "m", set()))
+        assert maps[0] == {"pod-a": 2.0}
