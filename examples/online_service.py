@@ -5,6 +5,8 @@ endpoints + Prometheus /metrics + the ZMQ events pool, configured via the
 same environment variables (main.go:41-58,167-225):
 
   HTTP_PORT              (default 8080)
+  SNAPSHOT_PATH          optional: load the index snapshot at boot if the
+                         file exists; save on SIGTERM (table backends)
   ZMQ_ENDPOINT           (default tcp://*:5557)
   ZMQ_TOPIC              (default kv@)
   POOL_CONCURRENCY       (default 4)
@@ -87,6 +89,13 @@ def main():
 
     collector.register()
     index = new_index(config.kv_block_index)
+    snapshot = os.environ.get("SNAPSHOT_PATH")
+    if snapshot and os.path.exists(snapshot) and hasattr(index, "load"):
+        try:
+            index.load(snapshot)
+            logger.info("loaded index snapshot from %s", snapshot)
+        except Exception:
+            logger.exception("snapshot load failed; starting empty")
     indexer = Indexer(config, kv_block_index=index)
     indexer.run()
     logger.info("indexer running (block_size=%d)", block_size)
@@ -121,6 +130,12 @@ def main():
             signal.pause()
     finally:
         logger.info("shutting down")
+        if snapshot and hasattr(index, "save"):
+            try:
+                index.save(snapshot)
+                logger.info("saved index snapshot to %s", snapshot)
+            except Exception:
+                logger.exception("snapshot save failed")
         http.stop()
         pool.shutdown()
         indexer.shutdown()

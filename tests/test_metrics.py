@@ -106,3 +106,36 @@ class TestInstrumentedOverTableIndex:
         counts = torch.tensor([2], dtype=torch.int32)
         maps = idx.scores_to_map(idx.fused_scores(hashes, counts, "m", set()))
         assert maps[0] == {"pod-a": 2.0}
+
+
+class TestTraceLogging:
+    def test_trace_level_and_helpers(self):
+        import logging
+
+        from llmd_kvcache_amd.utils.logging import (
+            TRACE,
+            enable_trace,
+            get_logger,
+            trace,
+        )
+
+        assert logging.getLevelName(TRACE) == "TRACE"
+        log = get_logger("test")
+        records = []
+
+        class Capture(logging.Handler):
+            def emit(self, record):
+                records.append(record.getMessage())
+
+        h = Capture()
+        log.addHandler(h)
+        try:
+            log.setLevel(TRACE + 1)
+            trace(log, "hidden %d", 1)
+            assert records == []
+            enable_trace()
+            log.setLevel(TRACE)
+            trace(log, "visible %d", 2)
+            assert records == ["visible 2"]
+        finally:
+            log.removeHandler(h)
