@@ -201,3 +201,38 @@ class TestGrpcErrorPath:
             client.close()
         finally:
             server.stop(None)
+
+
+class TestAsgiApp:
+    def test_fastapi_endpoints(self):
+        pytest.importorskip("fastapi")
+        from fastapi.testclient import TestClient
+
+        from llmd_kvcache_amd.service.asgi import build_app
+
+        idx, index = make_indexer()
+        prompt = "asgi test prompt payload!" * 10
+        tokens = FixedTokenizer().encode(prompt, "m")[0]
+        keys = idx.tokens_processor.tokens_to_kv_block_keys(None, tokens, "m")
+        index.add(keys, keys, [PodEntry("pod-f", "gpu")])
+
+        client = TestClient(build_app(idx))
+        r = client.post("/score_completions",
+                        json={"prompt": prompt, "model": "m"})
+        assert r.status_code == 200
+        assert r.json().get("pod-f", 0) > 0
+
+        r = client.post("/score_completions", json={"prompt": "", "model": "m"})
+        assert r.status_code == 400
+
+        r = client.post("/score_chat_completions", json={
+            "model": "m",
+            "messages": [{"role": "user", "content": "hi there you"}],
+            "chat_template": "{% for m in messages %}{{ m['content'] }}"
+                             "{% endfor %}",
+        })
+        assert r.status_code == 200
+        assert r.json()["templated_messages"] == "hi there you"
+
+        assert client.get("/health").json()["status"] == "ok"
+        assert client.get("/metrics").status_code == 200
