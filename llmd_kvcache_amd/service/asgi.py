@@ -19,39 +19,41 @@ from ..preprocessing import chat_completions as cc
 
 
 def build_app(indexer: Indexer):
-    from fastapi import FastAPI, HTTPException
+    # Raw-body handlers (no pydantic request models): keeps the endpoint
+    # payloads byte-identical to the reference's loosely-typed JSON and
+    # independent of the installed fastapi/pydantic major versions.
+    from fastapi import FastAPI, HTTPException, Request
     from fastapi.responses import PlainTextResponse
-    from pydantic import BaseModel
-
-    class ScoreRequest(BaseModel):
-        prompt: str
-        model: str = ""
-
-    class ChatScoreRequest(BaseModel):
-        model: str = ""
-        messages: List[Dict[str, Any]] = []
-        chat_template: Optional[str] = None
-        tools: Optional[List[Dict[str, Any]]] = None
-        documents: Optional[List[Dict[str, Any]]] = None
-        add_generation_prompt: bool = False
-        continue_final_message: bool = False
-        chat_template_kwargs: Dict[str, Any] = {}
 
     app = FastAPI(title="llmd_kvcache_amd indexer")
 
+    async def _json(request: Request) -> Dict[str, Any]:
+        try:
+            body = await request.json()
+        except Exception:
+            raise HTTPException(400, "invalid JSON body")
+        if not isinstance(body, dict):
+            raise HTTPException(400, "JSON object expected")
+        return body
+
     @app.post("/score_completions")
-    def score_completions(req: ScoreRequest):
-        if not req.prompt:
+    async def score_completions(request: Request):
+        body = await _json(request)
+        prompt = body.get("prompt", "")
+        if not prompt:
             raise HTTPException(400, "field 'prompt' required")
-        return indexer.get_pod_scores(None, req.prompt, req.model, []) or {}
+        model = body.get("model", "")
+        return indexer.get_pod_scores(None, prompt, model, []) or {}
 
     @app.post("/score_chat_completions")
-    def score_chat_completions(req: ChatScoreRequest):
-        template = req.chat_template
-        kwargs = dict(req.chat_template_kwargs)
+    async def score_chat_completions(request: Request):
+        body = await _json(request)
+        model = body.get("model", "")
+        template = body.get("chat_template")
+        kwargs = dict(body.get("chat_template_kwargs") or {})
         if template is None:
             fetched, tvars = cc.get_model_chat_template(
-                cc.FetchChatTemplateRequest(model=req.model)
+                cc.FetchChatTemplateRequest(model=model)
             )
             template = fetched
             merged = dict(tvars)
@@ -59,16 +61,18 @@ def build_app(indexer: Indexer):
             kwargs = merged
         rendered = cc.render_chat_template(
             cc.RenderJinjaTemplateRequest(
-                conversations=[req.messages],
+                conversations=[body.get("messages", [])],
                 chat_template=template,
-                tools=req.tools,
-                documents=req.documents,
-                add_generation_prompt=req.add_generation_prompt,
-                continue_final_message=req.continue_final_message,
+                tools=body.get("tools"),
+                documents=body.get("documents"),
+                add_generation_prompt=body.get("add_generation_prompt",
+                                               False),
+                continue_final_message=body.get("continue_final_message",
+                                                False),
                 chat_template_kwargs=kwargs,
             )
         )
-        pods = indexer.get_pod_scores(None, rendered, req.model, [])
+        pods = indexer.get_pod_scores(None, rendered, model, [])
         return {"podScores": pods or {}, "templated_messages": rendered}
 
     @app.get("/health")
