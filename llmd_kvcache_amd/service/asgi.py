@@ -1,53 +1,64 @@
-"""ASGI (FastAPI) service front - the production-grade alternative to the
-stdlib HTTP server, same endpoint surface as the reference's online
-binary (main.go:269-365):
+"""ASGI service front - a dependency-free ASGI 3.0 application with the
+same endpoint surface as the reference's online binary (main.go:269-365),
+deployable under uvicorn/gunicorn:
 
     uvicorn --factory llmd_kvcache_amd.service.asgi:create_default_app
 
-or embed:  app = build_app(indexer)
-
-Scoring runs in the ASGI thread pool (the indexer is thread-safe and the
-GPU fast path releases the GIL inside kernels/tokenizers).
+(Written as a raw ASGI callable rather than on a web framework: the
+payloads mirror the reference's loosely-typed JSON exactly and the
+module has zero extra dependencies.)
 """
 
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional
+import json
+from typing import Any, Dict
 
 from ..indexer import Indexer
 from ..preprocessing import chat_completions as cc
 
 
-def build_app(indexer: Indexer):
-    # Raw-body handlers (no pydantic request models): keeps the endpoint
-    # payloads byte-identical to the reference's loosely-typed JSON and
-    # independent of the installed fastapi/pydantic major versions.
-    from fastapi import FastAPI, HTTPException, Request
-    from fastapi.responses import PlainTextResponse
+class AsgiIndexerApp:
+    def __init__(self, indexer: Indexer):
+        self.indexer = indexer
 
-    app = FastAPI(title="llmd_kvcache_amd indexer")
-
-    async def _json(request: Request) -> Dict[str, Any]:
+    async def __call__(self, scope, receive, send):
+        if scope["type"] == "lifespan":
+            await self._lifespan(receive, send)
+            return
+        if scope["type"] != "http":
+            return
+        method = scope["method"]
+        path = scope["path"]
         try:
-            body = await request.json()
-        except Exception:
-            raise HTTPException(400, "invalid JSON body")
-        if not isinstance(body, dict):
-            raise HTTPException(400, "JSON object expected")
-        return body
+            if method == "POST" and path == "/score_completions":
+                body = await self._read_json(receive)
+                await self._score_completions(send, body)
+            elif method == "POST" and path == "/score_chat_completions":
+                body = await self._read_json(receive)
+                await self._score_chat(send, body)
+            elif method == "GET" and path == "/health":
+                await self._json(send, 200, {"status": "ok"})
+            elif method == "GET" and path == "/metrics":
+                await self._metrics(send)
+            else:
+                await self._text(send, 404, "not found")
+        except _BadRequest as e:
+            await self._text(send, 400, str(e))
+        except Exception as e:  # pragma: no cover - defensive
+            await self._text(send, 500, f"error: {e}")
 
-    @app.post("/score_completions")
-    async def score_completions(request: Request):
-        body = await _json(request)
+    # -- handlers ------------------------------------------------------
+    async def _score_completions(self, send, body: Dict[str, Any]):
         prompt = body.get("prompt", "")
         if not prompt:
-            raise HTTPException(400, "field 'prompt' required")
-        model = body.get("model", "")
-        return indexer.get_pod_scores(None, prompt, model, []) or {}
+            raise _BadRequest("field 'prompt' required")
+        scores = self.indexer.get_pod_scores(
+            None, prompt, body.get("model", ""), []
+        )
+        await self._json(send, 200, scores or {})
 
-    @app.post("/score_chat_completions")
-    async def score_chat_completions(request: Request):
-        body = await _json(request)
+    async def _score_chat(self, send, body: Dict[str, Any]):
         model = body.get("model", "")
         template = body.get("chat_template")
         kwargs = dict(body.get("chat_template_kwargs") or {})
@@ -65,31 +76,76 @@ def build_app(indexer: Indexer):
                 chat_template=template,
                 tools=body.get("tools"),
                 documents=body.get("documents"),
-                add_generation_prompt=body.get("add_generation_prompt",
-                                               False),
+                add_generation_prompt=body.get("add_generation_prompt", False),
                 continue_final_message=body.get("continue_final_message",
                                                 False),
                 chat_template_kwargs=kwargs,
             )
         )
-        pods = indexer.get_pod_scores(None, rendered, model, [])
-        return {"podScores": pods or {}, "templated_messages": rendered}
+        pods = self.indexer.get_pod_scores(None, rendered, model, [])
+        await self._json(send, 200, {"podScores": pods or {},
+                                     "templated_messages": rendered})
 
-    @app.get("/health")
-    def health():
-        return {"status": "ok"}
-
-    @app.get("/metrics")
-    def metrics():
+    async def _metrics(self, send):
         from prometheus_client import generate_latest
 
-        return PlainTextResponse(generate_latest(),
-                                 media_type="text/plain; version=0.0.4")
+        await self._raw(send, 200, generate_latest(),
+                        b"text/plain; version=0.0.4")
 
-    return app
+    # -- plumbing ------------------------------------------------------
+    @staticmethod
+    async def _lifespan(receive, send):
+        while True:
+            message = await receive()
+            if message["type"] == "lifespan.startup":
+                await send({"type": "lifespan.startup.complete"})
+            elif message["type"] == "lifespan.shutdown":
+                await send({"type": "lifespan.shutdown.complete"})
+                return
+
+    @staticmethod
+    async def _read_json(receive) -> Dict[str, Any]:
+        chunks = []
+        while True:
+            message = await receive()
+            chunks.append(message.get("body", b""))
+            if not message.get("more_body"):
+                break
+        try:
+            body = json.loads(b"".join(chunks))
+        except Exception:
+            raise _BadRequest("invalid JSON body")
+        if not isinstance(body, dict):
+            raise _BadRequest("JSON object expected")
+        return body
+
+    @staticmethod
+    async def _raw(send, status: int, payload: bytes, ctype: bytes):
+        await send({
+            "type": "http.response.start",
+            "status": status,
+            "headers": [(b"content-type", ctype),
+                        (b"content-length", str(len(payload)).encode())],
+        })
+        await send({"type": "http.response.body", "body": payload})
+
+    async def _json(self, send, status: int, obj) -> None:
+        await self._raw(send, status, json.dumps(obj).encode(),
+                        b"application/json")
+
+    async def _text(self, send, status: int, msg: str) -> None:
+        await self._raw(send, status, msg.encode(), b"text/plain")
 
 
-def create_default_app():
+class _BadRequest(Exception):
+    pass
+
+
+def build_app(indexer: Indexer) -> AsgiIndexerApp:
+    return AsgiIndexerApp(indexer)
+
+
+def create_default_app() -> AsgiIndexerApp:
     """uvicorn --factory entry point: builds an indexer from env (same
     surface as examples/online_service.py)."""
     import os

@@ -204,10 +204,36 @@ class TestGrpcErrorPath:
 
 
 class TestAsgiApp:
-    def test_fastapi_endpoints(self):
-        pytest.importorskip("fastapi")
-        from fastapi.testclient import TestClient
+    """Drives the raw ASGI 3.0 app directly (send/receive callables) -
+    no framework test client needed."""
 
+    @staticmethod
+    def call(app, method, path, payload=None):
+        import asyncio
+
+        messages = []
+        body = json.dumps(payload).encode() if payload is not None else b""
+        received = [False]
+
+        async def receive():
+            received[0] = True
+            return {"type": "http.request", "body": body, "more_body": False}
+
+        async def send(message):
+            messages.append(message)
+
+        scope = {"type": "http", "method": method, "path": path,
+                 "headers": []}
+        asyncio.new_event_loop().run_until_complete(
+            app(scope, receive, send))
+        status = messages[0]["status"]
+        raw = b"".join(m.get("body", b"") for m in messages[1:])
+        try:
+            return status, json.loads(raw)
+        except Exception:
+            return status, raw
+
+    def test_asgi_endpoints(self):
         from llmd_kvcache_amd.service.asgi import build_app
 
         idx, index = make_indexer()
@@ -215,24 +241,28 @@ class TestAsgiApp:
         tokens = FixedTokenizer().encode(prompt, "m")[0]
         keys = idx.tokens_processor.tokens_to_kv_block_keys(None, tokens, "m")
         index.add(keys, keys, [PodEntry("pod-f", "gpu")])
+        app = build_app(idx)
 
-        client = TestClient(build_app(idx))
-        r = client.post("/score_completions",
-                        json={"prompt": prompt, "model": "m"})
-        assert r.status_code == 200
-        assert r.json().get("pod-f", 0) > 0
+        status, out = self.call(app, "POST", "/score_completions",
+                                {"prompt": prompt, "model": "m"})
+        assert status == 200 and out.get("pod-f", 0) > 0
 
-        r = client.post("/score_completions", json={"prompt": "", "model": "m"})
-        assert r.status_code == 400
+        status, _ = self.call(app, "POST", "/score_completions",
+                              {"prompt": "", "model": "m"})
+        assert status == 400
 
-        r = client.post("/score_chat_completions", json={
+        status, out = self.call(app, "POST", "/score_chat_completions", {
             "model": "m",
             "messages": [{"role": "user", "content": "hi there you"}],
             "chat_template": "{% for m in messages %}{{ m['content'] }}"
                              "{% endfor %}",
         })
-        assert r.status_code == 200
-        assert r.json()["templated_messages"] == "hi there you"
+        assert status == 200
+        assert out["templated_messages"] == "hi there you"
 
-        assert client.get("/health").json()["status"] == "ok"
-        assert client.get("/metrics").status_code == 200
+        status, out = self.call(app, "GET", "/health")
+        assert status == 200 and out["status"] == "ok"
+        status, _ = self.call(app, "GET", "/metrics")
+        assert status == 200
+        status, _ = self.call(app, "GET", "/nope")
+        assert status == 404
