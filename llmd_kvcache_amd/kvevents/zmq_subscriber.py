@@ -61,9 +61,14 @@ class ZmqSubscriber:
                 self._sock = sock
                 self.port = sock.port
                 # bound; the socket's own threads handle accept/read.
-                while not self._stop.is_set():
+                # Watch the listener: if it dies (OS error), fall through
+                # to the retry loop and rebind - the reference's 5 s
+                # reconnect semantics (zmq_subscriber.go:55-77).
+                while not self._stop.is_set() and sock.alive:
                     self._stop.wait(0.25)
-                return
+                if self._stop.is_set():
+                    return
+                raise ConnectionError("listener died; rebinding")
             except Exception as e:
                 logger.warning(
                     "zmq-subscriber failed (%s); retrying in %.0fs",

@@ -172,6 +172,7 @@ class SubSocket:
         listener.settimeout(0.25)
         self._listener = listener
         self.port = listener.getsockname()[1]
+        self._accepting = True
         t = threading.Thread(target=self._accept_loop, daemon=True,
                              name="zmtp-sub-accept")
         t.start()
@@ -180,22 +181,31 @@ class SubSocket:
     def connect(self, endpoint: str) -> None:
         host, port = parse_endpoint(endpoint)
         sock = socket.create_connection((host, port), timeout=5.0)
+        self._accepting = True  # connected mode counts as alive
         self._setup_peer(sock)
 
     def _accept_loop(self) -> None:
-        while not self._stop.is_set():
-            try:
-                conn, _addr = self._listener.accept()
-            except socket.timeout:
-                continue
-            except OSError:
-                return
-            t = threading.Thread(
-                target=self._setup_peer_safe, args=(conn,), daemon=True,
-                name="zmtp-sub-conn",
-            )
-            t.start()
-            self._threads.append(t)
+        try:
+            while not self._stop.is_set():
+                try:
+                    conn, _addr = self._listener.accept()
+                except socket.timeout:
+                    continue
+                except OSError:
+                    return
+                t = threading.Thread(
+                    target=self._setup_peer_safe, args=(conn,), daemon=True,
+                    name="zmtp-sub-conn",
+                )
+                t.start()
+                self._threads.append(t)
+        finally:
+            self._accepting = False
+
+    @property
+    def alive(self) -> bool:
+        """True while the bound listener is still accepting peers."""
+        return getattr(self, "_accepting", False) and not self._stop.is_set()
 
     def _setup_peer_safe(self, sock: socket.socket) -> None:
         try:

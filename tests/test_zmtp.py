@@ -270,3 +270,55 @@ class TestReconnect:
             pub.close()
         finally:
             sub.close()
+
+
+class TestRebind:
+    def test_subscriber_rebinds_after_listener_death(self):
+        """Killing the bound listener must trigger the 5s-retry rebind
+        (reference zmq_subscriber.go:55-77); publishers can then
+        reconnect and flow resumes."""
+        import llmd_kvcache_amd.kvevents.zmq_subscriber as zs
+
+        received = []
+
+        class PoolStub:
+            def add_task(self, msg):
+                received.append(msg)
+
+        old_retry = zs.RETRY_INTERVAL_S
+        zs.RETRY_INTERVAL_S = 0.2
+        sub = zs.ZmqSubscriber(PoolStub(), "tcp://127.0.0.1:0", "kv@")
+        try:
+            sub.start()
+            deadline = time.monotonic() + 5
+            while sub.port is None and time.monotonic() < deadline:
+                time.sleep(0.02)
+            first_port = sub.port
+            assert first_port
+            old_sock = sub._sock
+
+            # kill the listener out from under it; wait for a NEW socket
+            # (the old one reports alive until its accept() times out)
+            old_sock._listener.close()
+            deadline = time.monotonic() + 10
+            rebound = False
+            while time.monotonic() < deadline:
+                s = sub._sock
+                if s is not None and s is not old_sock and s.alive:
+                    rebound = True
+                    break
+                time.sleep(0.05)
+            assert rebound
+
+            pub = PubSocket()
+            pub.connect(f"tcp://127.0.0.1:{sub.port}")
+            assert pub.wait_for_subscriber(5.0)
+            pub.send_multipart([b"kv@p@m", struct.pack(">Q", 1), b"x"])
+            deadline = time.monotonic() + 5
+            while not received and time.monotonic() < deadline:
+                time.sleep(0.02)
+            assert received
+            pub.close()
+        finally:
+            zs.RETRY_INTERVAL_S = old_retry
+            sub.stop()
