@@ -226,3 +226,42 @@ def test_high_load_factor_inserts_remain_findable():
     recent = keys[int(n * 0.9):]
     found = nat.lookup(recent, set())
     assert len(found) >= len(recent) * 0.95
+
+
+def test_custom_tier_weights_through_indexer_fused_path():
+    """A third tier (disk) configured at the Indexer level must flow into
+    the fused scorer's weight vector (backend.go-style tier config)."""
+    from llmd_kvcache_amd.indexer import Config, Indexer
+    from llmd_kvcache_amd.scorer import KVCacheBackendConfig
+    from llmd_kvcache_amd.tokenization.pool import TokenizationPool
+    from llmd_kvcache_amd.tokenization.tokenizer import Tokenizer
+
+    class T(Tokenizer):
+        @property
+        def type(self):
+            return "t"
+
+        def encode(self, p, m, a=True):
+            n = len(p) // 4
+            return list(range(n)), [(i * 4, (i + 1) * 4) for i in range(n)]
+
+    cfg = Config()
+    cfg.token_processor.block_size = 4
+    cfg.backend_configs = [
+        KVCacheBackendConfig("gpu", 1.0),
+        KVCacheBackendConfig("cpu", 0.8),
+        KVCacheBackendConfig("disk", 0.25),
+    ]
+    index = NativeIndex(TableIndexConfig(
+        capacity=1 << 10, pods_per_key=4,
+        tier_names=["gpu", "cpu", "disk"]))
+    idx = Indexer(cfg, tokenization_pool=TokenizationPool(
+        cfg.tokenizers_pool, tokenizer=T()), kv_block_index=index)
+
+    tokens = list(range(8))
+    keys = idx.tokens_processor.tokens_to_kv_block_keys(None, tokens, "m")
+    index.add(keys, keys, [PodEntry("pod-d", "disk")])
+    index.add(keys[:1], keys[:1], [PodEntry("pod-d", "cpu")])
+    scores = idx.get_pod_scores(None, "abcd" * 8, "m", [])
+    # key0: max(cpu .8, disk .25)=0.8; key1: disk 0.25 -> 1.05
+    assert scores["pod-d"] == pytest.approx(1.05, abs=1e-6)
