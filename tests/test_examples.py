@@ -42,3 +42,41 @@ class TestExamples:
 
     def test_scorer_sketch(self):
         run_example("kv_cache_aware_scorer.py", expect="ready")
+
+
+class TestOnlineServiceConfig:
+    """build_index_config honors every KVCACHE_INDEX_BACKEND value the
+    online binary documents (main.go env surface)."""
+
+    def _build(self, monkeypatch, backend, **env):
+        sys.path.insert(0, EXAMPLES)
+        try:
+            import online_service
+        finally:
+            sys.path.pop(0)
+        monkeypatch.setenv("KVCACHE_INDEX_BACKEND", backend)
+        for k, v in env.items():
+            monkeypatch.setenv(k, v)
+        return online_service.build_index_config()
+
+    def test_in_memory(self, monkeypatch):
+        cfg = self._build(monkeypatch, "in_memory")
+        assert cfg.in_memory is not None and cfg.enable_metrics
+
+    def test_native(self, monkeypatch):
+        cfg = self._build(monkeypatch, "native")
+        assert cfg.native is not None
+
+    def test_cost_aware(self, monkeypatch):
+        cfg = self._build(monkeypatch, "cost_aware")
+        assert cfg.cost_aware is not None
+
+    def test_valkey_addr(self, monkeypatch):
+        cfg = self._build(monkeypatch, "valkey",
+                          REDIS_ADDR="valkey://10.0.0.9:6380")
+        assert cfg.valkey is not None
+        assert cfg.valkey.address == "valkey://10.0.0.9:6380"
+
+    def test_unknown_backend_exits(self, monkeypatch):
+        with pytest.raises(SystemExit):
+            self._build(monkeypatch, "bogus")
