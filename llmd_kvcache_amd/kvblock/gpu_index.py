@@ -170,10 +170,15 @@ class KvTable:
         )
 
     # -- raw ops (tensors in the table's device) -----------------------
-    def insert(self, engine_hashes, request_hashes, model_id, pod_entries):
+    def insert(self, engine_hashes, request_hashes, model_id, pod_entries,
+               emap_write: bool = True):
+        """emap_write=False skips the engine-map write - the tier-
+        promotion path reuses request hashes as engine hashes and a
+        self-mapping there would corrupt get_request_key()."""
         fn = self.ops.gpu_insert if self.is_cuda else self.ops.cpu_insert
         fn(*self._t(), engine_hashes, request_hashes, model_id, pod_entries,
-           self.next_epoch(), self.cfg.shard_id, self.cfg.num_shards)
+           self.next_epoch(), self.cfg.shard_id, self.cfg.num_shards,
+           1 if emap_write else 0)
 
     def evict(self, engine_hashes, model_id, pod_entries):
         fn = self.ops.gpu_evict if self.is_cuda else self.ops.cpu_evict
@@ -314,7 +319,7 @@ class TableIndex(Index):
         return result
 
     def add(self, engine_keys: Sequence[Key], request_keys: Sequence[Key],
-            entries: Sequence[PodEntry]) -> None:
+            entries: Sequence[PodEntry], write_emap: bool = True) -> None:
         if not engine_keys or not request_keys or not entries:
             raise ValueError("no keys or entries provided for adding to index")
         if len(engine_keys) != len(request_keys):
@@ -324,7 +329,7 @@ class TableIndex(Index):
         rh = self._hashes_tensor(request_keys)
         pe = self._entries_tensor(entries)
         with self._write_lock:
-            self.table.insert(eh, rh, model_id, pe)
+            self.table.insert(eh, rh, model_id, pe, emap_write=write_emap)
 
     def evict(self, engine_key: Key, entries: Sequence[PodEntry]) -> None:
         if not entries:
@@ -430,6 +435,53 @@ class _nullcontext:
 
     def __exit__(self, *a):
         return False
+
+
+class _PinnedUploader:
+    """Double-buffered pinned staging for event uploads.
+
+    numpy -> pinned host buffer (CPU memcpy) -> true async H2D DMA,
+    instead of pageable `.to(non_blocking=True)` (which bounces through
+    the runtime's staging path and serializes).  Two slots rotate; a
+    slot's recorded event is synchronized before the slot is rewritten
+    so an in-flight DMA never reads a buffer being reused (roadmap #4).
+    """
+
+    SLOTS = 2
+
+    def __init__(self):
+        self._slots = [dict() for _ in range(self.SLOTS)]
+        self._events = [None] * self.SLOTS
+        self._i = 0
+
+    def begin(self) -> None:
+        self._i = (self._i + 1) % self.SLOTS
+        ev = self._events[self._i]
+        if ev is not None:
+            ev.synchronize()
+
+    def up(self, arr, name: str, device, dtype=None) -> torch.Tensor:
+        import numpy as np
+
+        if not isinstance(arr, np.ndarray):
+            arr = np.asarray(arr, dtype=dtype)
+        src = torch.from_numpy(np.ascontiguousarray(arr))
+        n = src.numel()
+        if n == 0:
+            return torch.zeros(0, dtype=src.dtype, device=device)
+        slot = self._slots[self._i]
+        buf = slot.get(name)
+        if buf is None or buf.numel() < n or buf.dtype != src.dtype:
+            cap = max(1024, 1 << (n - 1).bit_length())
+            buf = torch.empty(cap, dtype=src.dtype, pin_memory=True)
+            slot[name] = buf
+        buf[:n].copy_(src)
+        return buf[:n].to(device, non_blocking=True)
+
+    def end(self) -> None:
+        ev = torch.cuda.Event()
+        ev.record()
+        self._events[self._i] = ev
 
 
 class NativeIndex(TableIndex):
@@ -581,21 +633,24 @@ class GpuIndex(TableIndex):
             return
 
         d = self.device
-        i32 = torch.int32
         tokens_np = (np.concatenate(token_arrays) if token_arrays
                      else np.zeros(0, dtype=np.int64))
         hashes_np = (np.concatenate(hash_arrays) if hash_arrays
                      else np.zeros(0, dtype=np.int64))
+        up = getattr(self, "_pinned_up", None)
+        if up is None:
+            up = self._pinned_up = _PinnedUploader()
+        up.begin()
         common = (
-            torch.from_numpy(tokens_np).to(d, non_blocking=True),
-            torch.tensor(tok_off, dtype=i32, device=d),
-            torch.from_numpy(hashes_np).to(d, non_blocking=True),
-            torch.tensor(eh_off, dtype=i32, device=d),
-            torch.tensor(parents, dtype=torch.int64, device=d),
-            torch.tensor(has_parent, dtype=torch.uint8, device=d),
-            torch.tensor(ev_type, dtype=torch.uint8, device=d),
-            torch.tensor(pod_entry, dtype=i32, device=d),
-            torch.tensor(grp_off, dtype=i32, device=d),
+            up.up(tokens_np, "tok", d),
+            up.up(tok_off, "tok_off", d, dtype=np.int32),
+            up.up(hashes_np, "eh", d),
+            up.up(eh_off, "eh_off", d, dtype=np.int32),
+            up.up(parents, "par", d, dtype=np.int64),
+            up.up(has_parent, "haspar", d, dtype=np.uint8),
+            up.up(ev_type, "evt", d, dtype=np.uint8),
+            up.up(pod_entry, "pe", d, dtype=np.int32),
+            up.up(grp_off, "grp", d, dtype=np.int32),
         )
         # Phase-split fast path (chains lane-per-group, inserts
         # thread-per-block) unless this batch both stores AND removes some
@@ -614,7 +669,7 @@ class GpuIndex(TableIndex):
             ev_of = np.repeat(np.arange(len(counts), dtype=np.int32), counts)
             self.table.ops.gpu_apply_events_split(
                 *self.table._t(), *common,
-                torch.from_numpy(ev_of).to(d, non_blocking=True),
+                up.up(ev_of, "ev_of", d),
                 model_id, _to_i64(init_hash), block_size,
                 self.table.next_epoch(), self.cfg.shard_id,
                 self.cfg.num_shards,
@@ -626,3 +681,4 @@ class GpuIndex(TableIndex):
                 self.table.next_epoch(), self.cfg.shard_id,
                 self.cfg.num_shards,
             )
+        up.end()

@@ -66,7 +66,22 @@ class TieredIndex(Index):
         merged = dict(hot)
         for k, entries in cold.items():
             merged.setdefault(k, entries)
+        self._promote(cold)
         return merged
+
+    def _promote(self, cold_hits: Dict[Key, List[PodEntry]]) -> None:
+        """Re-warm the HBM tier with cold-tier hits so the next request
+        for this prefix is served by the single fused kernel again.
+        write_emap=False: the engine map already carries these mappings
+        (adds write both tiers) and a request-key self-mapping would
+        corrupt get_request_key()."""
+        from .gpu_index import TableIndex
+
+        if not cold_hits or not isinstance(self.hot, TableIndex):
+            return
+        for k, entries in cold_hits.items():
+            if entries:
+                self.hot.add([k], [k], entries, write_emap=False)
 
     def add(
         self,

@@ -16,7 +16,7 @@ std::vector<uint64_t> tokens_to_chunk_hashes_fast(std::vector<uint64_t>,
                                                   uint64_t, int64_t);
 void cpu_insert(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                 at::Tensor, at::Tensor, int64_t, at::Tensor, at::Tensor,
-                int64_t, at::Tensor, int64_t, int64_t, int64_t);
+                int64_t, at::Tensor, int64_t, int64_t, int64_t, int64_t);
 at::Tensor cpu_score_from_masks(at::Tensor, at::Tensor, at::Tensor, int64_t);
 void cpu_evict(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                at::Tensor, at::Tensor, int64_t, at::Tensor, int64_t,
@@ -40,7 +40,7 @@ std::vector<at::Tensor> cpu_get_request_keys(at::Tensor, at::Tensor,
 // hip_ops.hip
 void gpu_insert(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                 at::Tensor, at::Tensor, int64_t, at::Tensor, at::Tensor,
-                int64_t, at::Tensor, int64_t, int64_t, int64_t);
+                int64_t, at::Tensor, int64_t, int64_t, int64_t, int64_t);
 void gpu_evict(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                at::Tensor, at::Tensor, int64_t, at::Tensor, int64_t,
                at::Tensor);

@@ -267,7 +267,7 @@ void cpu_insert(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
                 at::Tensor e_vals, int64_t pods_per_key,
                 at::Tensor engine_hashes, at::Tensor request_hashes,
                 int64_t model_id, at::Tensor pod_entries, int64_t epoch,
-                int64_t shard_id, int64_t num_shards) {
+                int64_t shard_id, int64_t num_shards, int64_t emap_write) {
   auto v = make_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
                      (int)pods_per_key);
   auto eh = engine_hashes.contiguous();
@@ -280,7 +280,9 @@ void cpu_insert(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
   int64_t m = pe.numel();
   TORCH_CHECK(rh.numel() == n, "engine/request key length mismatch");
   for (int64_t i = 0; i < n; ++i) {
-    emap_put(v, ehp[i], (uint32_t)model_id, remap_hash(rhp[i]), (int32_t)epoch);
+    if (emap_write)
+      emap_put(v, ehp[i], (uint32_t)model_id, remap_hash(rhp[i]),
+               (int32_t)epoch);
     if (num_shards > 1 &&
         (int64_t)(remap_hash(rhp[i]) % (uint64_t)num_shards) != shard_id)
       continue;

@@ -231,10 +231,10 @@ __global__ void k_insert(DevTable v, const uint64_t* __restrict__ eh,
                          const uint64_t* __restrict__ rh, int64_t n,
                          uint32_t model, const uint32_t* __restrict__ entries,
                          int n_entries, int32_t epoch, int shard_id,
-                         int num_shards) {
+                         int num_shards, int emap_write) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
-  dev_emap_put(v, eh[i], model, remap_hash(rh[i]));
+  if (emap_write) dev_emap_put(v, eh[i], model, remap_hash(rh[i]));
   if (num_shards > 1 &&
       (int)(remap_hash(rh[i]) % (uint64_t)num_shards) != shard_id)
     return;  // engine map replicated; main table sharded by ownership
@@ -763,7 +763,7 @@ void gpu_insert(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
                 at::Tensor e_vals, int64_t pods_per_key,
                 at::Tensor engine_hashes, at::Tensor request_hashes,
                 int64_t model_id, at::Tensor pod_entries, int64_t epoch,
-                int64_t shard_id, int64_t num_shards) {
+                int64_t shard_id, int64_t num_shards, int64_t emap_write) {
   auto v = dev_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
                     pods_per_key);
   int64_t n = engine_hashes.numel();
@@ -776,7 +776,7 @@ void gpu_insert(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
                      reinterpret_cast<const uint32_t*>(
                          pod_entries.data_ptr<int32_t>()),
                      (int)pod_entries.numel(), (int32_t)epoch,
-                     (int)shard_id, (int)num_shards);
+                     (int)shard_id, (int)num_shards, (int)emap_write);
 }
 
 void gpu_evict(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
