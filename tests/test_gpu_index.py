@@ -325,10 +325,20 @@ class TestTieredGpu:
             all_keys.extend(ks)
         torch.cuda.synchronize()
         early = all_keys[:50]
+        before = t.hot.lookup(early, set())
+        assert len(before) < len(early)  # HBM tier evicted under pressure
         merged = t.lookup(early, set())
         assert len(merged) == len(early)
-        hot_only = t.hot.lookup(early + [Key(MODEL, 1)], set())
-        assert len(hot_only) < len(early)  # HBM tier evicted under pressure
+        # cold hits were promoted back into the HBM tier
+        after = t.hot.lookup(early, set())
+        assert len(after) == len(early)
+        # emap_write=0 on the GPU insert: a key known only to the cold
+        # tier gains no engine-map self-mapping when promoted
+        solo = Key(MODEL, 99_999)
+        cold.add([solo], [solo], [PodEntry("pod-a", "gpu")])
+        t.lookup([solo], set())
+        assert t.hot.lookup([solo], set())  # promoted
+        assert t.hot.get_request_key(solo) is None
 
 
 class TestManyPods:
