@@ -455,10 +455,18 @@ class _PinnedUploader:
         self._i = 0
 
     def begin(self) -> None:
+        import time as _time
+
         self._i = (self._i + 1) % self.SLOTS
         ev = self._events[self._i]
         if ev is not None:
-            ev.synchronize()
+            # NOT ev.synchronize(): hipEventSynchronize costs ~2.4 ms of
+            # interrupt-wakeup latency on this stack even for an already-
+            # complete event (measured: scripts/bench_upload.py), which
+            # would dominate the ~3 ms batch. query() is ~us and nearly
+            # always already true when a slot rotates back around.
+            while not ev.query():
+                _time.sleep(0)
 
     def up(self, arr, name: str, device, dtype=None) -> torch.Tensor:
         import numpy as np
@@ -479,9 +487,12 @@ class _PinnedUploader:
         return buf[:n].to(device, non_blocking=True)
 
     def end(self) -> None:
-        ev = torch.cuda.Event()
+        # reuse one event per slot: hipEventCreate/Destroy churn per call
+        # is measurable on this stack (scripts/bench_upload.py)
+        ev = self._events[self._i]
+        if ev is None:
+            ev = self._events[self._i] = torch.cuda.Event()
         ev.record()
-        self._events[self._i] = ev
 
 
 class NativeIndex(TableIndex):
@@ -637,21 +648,47 @@ class GpuIndex(TableIndex):
                      else np.zeros(0, dtype=np.int64))
         hashes_np = (np.concatenate(hash_arrays) if hash_arrays
                      else np.zeros(0, dtype=np.int64))
-        up = getattr(self, "_pinned_up", None)
-        if up is None:
-            up = self._pinned_up = _PinnedUploader()
-        up.begin()
-        common = (
-            up.up(tokens_np, "tok", d),
-            up.up(tok_off, "tok_off", d, dtype=np.int32),
-            up.up(hashes_np, "eh", d),
-            up.up(eh_off, "eh_off", d, dtype=np.int32),
-            up.up(parents, "par", d, dtype=np.int64),
-            up.up(has_parent, "haspar", d, dtype=np.uint8),
-            up.up(ev_type, "evt", d, dtype=np.uint8),
-            up.up(pod_entry, "pe", d, dtype=np.int32),
-            up.up(grp_off, "grp", d, dtype=np.int32),
-        )
+        import os as _os
+
+        # Default is PAGEABLE staging: measured on MI355X (ROCm 7.2,
+        # scripts/bench_upload.py) the pinned+async path is bimodal -
+        # ~0.1 ms typical for the 4 MB token upload but with sporadic
+        # 38-96 ms stalls in hipMemcpyAsync-from-pinned that crater
+        # ingest 9.9M -> 1.8M blocks/s, while pageable .to() is a steady
+        # ~0.1 ms (2.8 ms per 512-event batch end to end).  KVIDX_PINNED=1
+        # opts back in for stacks where the async path behaves.
+        if _os.environ.get("KVIDX_PINNED", "0") == "1":
+            up = getattr(self, "_pinned_up", None)
+            if up is None:
+                up = self._pinned_up = _PinnedUploader()
+        else:
+            up = None
+        if up is not None:
+            up.begin()
+            common = (
+                up.up(tokens_np, "tok", d),
+                up.up(tok_off, "tok_off", d, dtype=np.int32),
+                up.up(hashes_np, "eh", d),
+                up.up(eh_off, "eh_off", d, dtype=np.int32),
+                up.up(parents, "par", d, dtype=np.int64),
+                up.up(has_parent, "haspar", d, dtype=np.uint8),
+                up.up(ev_type, "evt", d, dtype=np.uint8),
+                up.up(pod_entry, "pe", d, dtype=np.int32),
+                up.up(grp_off, "grp", d, dtype=np.int32),
+            )
+        else:
+            i32 = torch.int32
+            common = (
+                torch.from_numpy(tokens_np).to(d, non_blocking=True),
+                torch.tensor(tok_off, dtype=i32, device=d),
+                torch.from_numpy(hashes_np).to(d, non_blocking=True),
+                torch.tensor(eh_off, dtype=i32, device=d),
+                torch.tensor(parents, dtype=torch.int64, device=d),
+                torch.tensor(has_parent, dtype=torch.uint8, device=d),
+                torch.tensor(ev_type, dtype=torch.uint8, device=d),
+                torch.tensor(pod_entry, dtype=i32, device=d),
+                torch.tensor(grp_off, dtype=i32, device=d),
+            )
         # Phase-split fast path (chains lane-per-group, inserts
         # thread-per-block) unless this batch both stores AND removes some
         # engine hash - then per-pod op ORDER matters and the serial
@@ -667,9 +704,10 @@ class GpuIndex(TableIndex):
         if use_split:
             counts = np.diff(np.asarray(eh_off, dtype=np.int64))
             ev_of = np.repeat(np.arange(len(counts), dtype=np.int32), counts)
+            ev_of_t = (up.up(ev_of, "ev_of", d) if up is not None
+                       else torch.from_numpy(ev_of).to(d, non_blocking=True))
             self.table.ops.gpu_apply_events_split(
-                *self.table._t(), *common,
-                up.up(ev_of, "ev_of", d),
+                *self.table._t(), *common, ev_of_t,
                 model_id, _to_i64(init_hash), block_size,
                 self.table.next_epoch(), self.cfg.shard_id,
                 self.cfg.num_shards,
@@ -681,4 +719,5 @@ class GpuIndex(TableIndex):
                 self.table.next_epoch(), self.cfg.shard_id,
                 self.cfg.num_shards,
             )
-        up.end()
+        if up is not None:
+            up.end()
