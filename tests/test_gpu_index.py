@@ -457,3 +457,37 @@ class TestHighLoadFactor:
         gpu.table.insert(dup_eh, dup_rh, gpu.registry.model_id(MODEL), pe)
         torch.cuda.synchronize()
         assert gpu.lookup([Key(MODEL, 5_000_000 + n)], set())
+
+
+class TestPinnedUploadOptIn:
+    def test_pinned_path_parity(self, monkeypatch):
+        """KVIDX_PINNED=1 routes uploads through the double-buffered
+        pinned stager (_PinnedUploader); results must match the default
+        pageable path exactly (opt-in path kept working; ROADMAP #4)."""
+        tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=4))
+        results = {}
+        for mode in ("0", "1"):
+            monkeypatch.setenv("KVIDX_PINNED", mode)
+            gpu = GpuIndex(GpuIndexConfig(capacity=1 << 12, pods_per_key=10))
+            batches = []
+            for e in range(8):
+                toks = [e * 100 + i for i in range(8)]
+                batches.append((f"pod-{e % 3}", MODEL,
+                                [BlockStored([1000 + 2 * e, 1001 + 2 * e],
+                                             None, toks, 4)]))
+            gpu.apply_event_batches(batches, tp)
+            # second call exercises slot rotation + event reuse
+            gpu.apply_event_batches(
+                [("pod-9", MODEL, [BlockStored([5000], None,
+                                               list(range(4)), 4)])], tp)
+            torch.cuda.synchronize()
+            if mode == "1":
+                assert getattr(gpu, "_pinned_up", None) is not None
+            snap = {}
+            for e in range(8):
+                keys = tp.tokens_to_kv_block_keys(
+                    None, [e * 100 + i for i in range(8)], MODEL)
+                snap[e] = {k: sorted(map(tuple, v)) for k, v in
+                           gpu.lookup(keys, set()).items()}
+            results[mode] = snap
+        assert results["0"] == results["1"]
