@@ -258,6 +258,7 @@ def main():
                          device=device)
     weights = index.tier_weights()
     model_id = index.registry.model_id(MODEL)
+    n_tiers = max(1, len(index.registry.id_to_tier))
     num_pods = index._num_pods_padded()
     no_filter = torch.zeros(0, dtype=torch.int64, device=device)
     key_offsets = torch.arange(0, (args.batch + 1) * KEYS_PER_PROMPT,
@@ -296,7 +297,7 @@ def main():
             scores = ops.gpu_fused_score(
                 *index.table._t(), hashes, key_offsets, model_id,
                 no_filter, weights, num_pods, index.table.next_epoch(),
-                KEYS_PER_PROMPT)
+                KEYS_PER_PROMPT, n_tiers)
         best = scores.argmax(dim=1)
         return best.cpu(), scores[:, 0].sum().item()  # forces D2H
 
@@ -338,7 +339,7 @@ def main():
             scores = ops.gpu_fused_score(
                 *index.table._t(), hashes, key_offsets, model_id,
                 no_filter, weights, num_pods, frozen_epoch_note,
-                KEYS_PER_PROMPT)
+                KEYS_PER_PROMPT, n_tiers)
             return scores.argmax(dim=1)
 
         try:
@@ -447,7 +448,7 @@ def main():
             sc = ops.gpu_fused_score(
                 *index.table._t(), hh.view(-1), one_off, model_id,
                 no_filter, weights, num_pods, index.table.next_epoch(),
-                KEYS_PER_PROMPT)
+                KEYS_PER_PROMPT, n_tiers)
             sc.argmax(dim=1).cpu()
             torch.cuda.synchronize()
             lat1.append(time.monotonic() - t0)

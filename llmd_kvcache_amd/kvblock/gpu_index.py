@@ -194,12 +194,15 @@ class KvTable:
                   num_pods, self.next_epoch(), shard_id, num_shards)
 
     def fused_score(self, hashes, counts_or_offsets, model_id, filter_words,
-                    weights, num_pods, max_k=None):
+                    weights, num_pods, max_k=None, n_tiers=MAX_TIERS):
         if self.is_cuda:
+            # n_tiers (tiers actually registered) sizes the kernel's LDS:
+            # 1-2 real tiers instead of MAX_TIERS=4 keeps 256-pod fleets
+            # at 16-32 KB/WG and full occupancy.
             return self.ops.gpu_fused_score(
                 *self._t(), hashes, counts_or_offsets, model_id, filter_words,
                 weights, num_pods, self.next_epoch(),
-                max_k if max_k is not None else 512)
+                max_k if max_k is not None else 512, n_tiers)
         return self.ops.cpu_fused_score(
             *self._t(), hashes, counts_or_offsets, model_id, filter_words,
             weights, num_pods, self.next_epoch())
@@ -366,7 +369,8 @@ class TableIndex(Index):
         if self.table.is_cuda:
             W = (num_pods + 63) // 64
             k = max_k if max_k is not None else 512
-            if k * MAX_TIERS * W * 8 > 64 * 1024:
+            n_tiers = max(1, len(self.registry.id_to_tier))
+            if k * n_tiers * W * 8 > 64 * 1024:
                 # LDS would overflow (huge fleet x long prompts): fall back
                 # to the two-kernel path - global-mask lookup + mask walk.
                 found, masks = self.table.lookup(hashes, model_id, filt,
@@ -375,7 +379,8 @@ class TableIndex(Index):
                 return self.table.ops.gpu_score_from_masks(
                     masks.contiguous(), counts_or_offsets, weights, num_pods)
             return self.table.fused_score(hashes, counts_or_offsets, model_id,
-                                          filt, weights, num_pods, max_k)
+                                          filt, weights, num_pods, max_k,
+                                          n_tiers=n_tiers)
         with self._write_lock:
             return self.table.fused_score(hashes, counts_or_offsets, model_id,
                                           filt, weights, num_pods, max_k)

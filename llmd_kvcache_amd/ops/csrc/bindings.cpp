@@ -57,7 +57,7 @@ std::vector<at::Tensor> gpu_lookup(at::Tensor, at::Tensor, at::Tensor,
 at::Tensor gpu_fused_score(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                            at::Tensor, at::Tensor, at::Tensor, int64_t,
                            at::Tensor, at::Tensor, int64_t, at::Tensor,
-                           at::Tensor, int64_t, int64_t, int64_t);
+                           at::Tensor, int64_t, int64_t, int64_t, int64_t);
 at::Tensor gpu_score_from_masks(at::Tensor, at::Tensor, at::Tensor, int64_t);
 std::vector<at::Tensor> gpu_hash_chain(at::Tensor, at::Tensor, at::Tensor,
                                        int64_t);

@@ -308,8 +308,11 @@ __global__ void __launch_bounds__(256) k_fused_score(
     const int32_t* __restrict__ offsets,  // [B+1]
     uint32_t model, const uint64_t* __restrict__ filter, int has_filter,
     const float* __restrict__ weights, int num_pods, int W, int32_t epoch,
-    float* __restrict__ scores) {
-  extern __shared__ unsigned long long lds_masks[];  // [K, T, W]
+    int n_tiers, float* __restrict__ scores) {
+  // n_tiers = tiers actually registered (usually 1-2, <= MAX_TIERS):
+  // sizing LDS by it instead of MAX_TIERS keeps 256-pod fleets (W=4)
+  // at 16-32 KB/WG instead of 64 KB, preserving occupancy.
+  extern __shared__ unsigned long long lds_masks[];  // [K, n_tiers, W]
   const int b = blockIdx.x;
   const int K = offsets[b + 1] - offsets[b];
   const uint64_t* h = hashes + offsets[b];
@@ -317,17 +320,18 @@ __global__ void __launch_bounds__(256) k_fused_score(
   const int wave = threadIdx.x >> 6;
 
   // zero LDS
-  for (int x = threadIdx.x; x < K * MAX_TIERS * W; x += blockDim.x)
+  for (int x = threadIdx.x; x < K * n_tiers * W; x += blockDim.x)
     lds_masks[x] = 0;
   __syncthreads();
 
   // phase 1: all probes in parallel (each is ~2 dependent HBM reads)
   for (int k = threadIdx.x; k < K; k += blockDim.x) {
-    unsigned long long* mk = lds_masks + (size_t)k * MAX_TIERS * W;
+    unsigned long long* mk = lds_masks + (size_t)k * n_tiers * W;
     dev_probe_collect(v, h[k], model, filter, has_filter, num_pods, W, epoch,
                       [&](uint32_t tier, uint32_t pid) {
-                        atomicOr(&mk[tier * W + pid / 64],
-                                 1ull << (pid % 64));
+                        if ((int)tier < n_tiers)  // interned => always true
+                          atomicOr(&mk[tier * W + pid / 64],
+                                   1ull << (pid % 64));
                       });
   }
   __syncthreads();
@@ -341,12 +345,12 @@ __global__ void __launch_bounds__(256) k_fused_score(
     active[w] = 0;
   }
   for (int k = 0; k < K; ++k) {
-    const unsigned long long* mk = lds_masks + (size_t)k * MAX_TIERS * W;
+    const unsigned long long* mk = lds_masks + (size_t)k * n_tiers * W;
     bool any = false;
     for (int w = 0; w < W; ++w) {
       int cur = 0;
       float wmax = 0.f;
-      for (int t = 0; t < MAX_TIERS; ++t) {
+      for (int t = 0; t < n_tiers; ++t) {
         if ((mk[t * W + w] >> lane) & 1) {
           cur = 1;
           wmax = fmaxf(wmax, weights[t]);
@@ -850,13 +854,16 @@ at::Tensor gpu_fused_score(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
                            int64_t pods_per_key, at::Tensor hashes,
                            at::Tensor offsets, int64_t model_id,
                            at::Tensor filter_words, at::Tensor weights,
-                           int64_t num_pods, int64_t epoch, int64_t max_k) {
+                           int64_t num_pods, int64_t epoch, int64_t max_k,
+                           int64_t n_tiers) {
   auto v = dev_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
                     pods_per_key);
   int64_t B = offsets.numel() - 1;
   int64_t W = (num_pods + 63) / 64;
   TORCH_CHECK(W <= 16, "fused score supports up to 1024 pods");
-  size_t lds = (size_t)max_k * MAX_TIERS * W * sizeof(uint64_t);
+  if (n_tiers < 1) n_tiers = 1;
+  if (n_tiers > MAX_TIERS) n_tiers = MAX_TIERS;
+  size_t lds = (size_t)max_k * n_tiers * W * sizeof(uint64_t);
   TORCH_CHECK(lds <= 64 * 1024,
               "fused score LDS overflow: reduce keys per prompt or pods");
   bool has_filter = filter_words.numel() > 0;
@@ -868,7 +875,7 @@ at::Tensor gpu_fused_score(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
       offsets.data_ptr<int32_t>(), (uint32_t)model_id,
       has_filter ? U64P(filter_words) : nullptr, has_filter ? 1 : 0,
       weights.data_ptr<float>(), (int)num_pods, (int)W, (int32_t)epoch,
-      scores.data_ptr<float>());
+      (int)n_tiers, scores.data_ptr<float>());
   return scores;
 }
 

@@ -83,7 +83,8 @@ def main():
         hashes = hashes.t().contiguous().view(-1)
         scores = ops.gpu_fused_score(
             *idx.table._t(), hashes, offs, model_id, no_filter, weights,
-            num_pods, idx.table.next_epoch(), K)
+            num_pods, idx.table.next_epoch(), K,
+            max(1, len(idx.registry.id_to_tier)))
         return scores.argmax(dim=1).cpu()
 
     # pre-stage event batches (host prep off the timed path to isolate
