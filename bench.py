@@ -260,6 +260,11 @@ def main():
     model_id = index.registry.model_id(MODEL)
     n_tiers = max(1, len(index.registry.id_to_tier))
     num_pods = index._num_pods_padded()
+    # direct fused-kernel calls must fit the 64 KB LDS budget; huge
+    # fleets (W*n_tiers too big) route through index.fused_scores, whose
+    # two-kernel global-mask fallback handles any size
+    fused_fits = (KEYS_PER_PROMPT * n_tiers *
+                  ((num_pods + 63) // 64) * 8 <= 64 * 1024)
     no_filter = torch.zeros(0, dtype=torch.int64, device=device)
     key_offsets = torch.arange(0, (args.batch + 1) * KEYS_PER_PROMPT,
                                KEYS_PER_PROMPT, dtype=torch.int32,
@@ -293,11 +298,14 @@ def main():
         if sharded is not None:
             scores = sharded.sharded_scores(hashes, key_offsets, MODEL,
                                             set(), weights)
-        else:
+        elif fused_fits:
             scores = ops.gpu_fused_score(
                 *index.table._t(), hashes, key_offsets, model_id,
                 no_filter, weights, num_pods, index.table.next_epoch(),
                 KEYS_PER_PROMPT, n_tiers)
+        else:
+            scores = index.fused_scores(hashes, key_offsets, MODEL, set(),
+                                        weights, max_k=KEYS_PER_PROMPT)
         best = scores.argmax(dim=1)
         return best.cpu(), scores[:, 0].sum().item()  # forces D2H
 
@@ -445,10 +453,14 @@ def main():
             t0 = time.monotonic()
             hh = ops.gpu_hash_chain_tr(one_tok, one_par, one_nch,
                                        BLOCK_SIZE, KEYS_PER_PROMPT, 0)
-            sc = ops.gpu_fused_score(
-                *index.table._t(), hh.view(-1), one_off, model_id,
-                no_filter, weights, num_pods, index.table.next_epoch(),
-                KEYS_PER_PROMPT, n_tiers)
+            if fused_fits:
+                sc = ops.gpu_fused_score(
+                    *index.table._t(), hh.view(-1), one_off, model_id,
+                    no_filter, weights, num_pods, index.table.next_epoch(),
+                    KEYS_PER_PROMPT, n_tiers)
+            else:
+                sc = index.fused_scores(hh.view(-1), one_off, MODEL, set(),
+                                        weights, max_k=KEYS_PER_PROMPT)
             sc.argmax(dim=1).cpu()
             torch.cuda.synchronize()
             lat1.append(time.monotonic() - t0)
