@@ -336,36 +336,32 @@ __global__ void __launch_bounds__(256) k_fused_score(
   }
   __syncthreads();
 
-  // phase 2: wave 0 walks the chain; lane l owns pod w*64+l per word w.
-  if (wave != 0) return;
-  float score[16];  // W <= 16 in the fused path (host-enforced)
-  int active[16];
-  for (int w = 0; w < W; ++w) {
-    score[w] = 0.f;
-    active[w] = 0;
-  }
-  for (int k = 0; k < K; ++k) {
-    const unsigned long long* mk = lds_masks + (size_t)k * n_tiers * W;
-    bool any = false;
-    for (int w = 0; w < W; ++w) {
+  // phase 2: the walk. Words (64-pod groups) are independent prefix
+  // walks, so they are distributed across the workgroup's 4 waves
+  // (wave v takes words v, v+4, ...); lane l owns pod w*64+l. The
+  // per-word early break is exact (a word stops when ITS active set
+  // empties - the old all-words break was only an optimization).
+  for (int w = wave; w < W; w += 4) {
+    float score = 0.f;
+    int active = 0;
+    for (int k = 0; k < K; ++k) {
+      const unsigned long long* mk =
+          lds_masks + (size_t)k * n_tiers * W + w;
       int cur = 0;
       float wmax = 0.f;
       for (int t = 0; t < n_tiers; ++t) {
-        if ((mk[t * W + w] >> lane) & 1) {
+        if ((mk[t * W] >> lane) & 1) {
           cur = 1;
           wmax = fmaxf(wmax, weights[t]);
         }
       }
-      int act = (k == 0) ? cur : (active[w] & cur);
-      active[w] = act;
-      if (act) score[w] += wmax;
-      if (__any(act)) any = true;
+      int act = (k == 0) ? cur : (active & cur);
+      active = act;
+      if (act) score += wmax;
+      if (!__any(act)) break;  // this word's active set is empty
     }
-    if (!any) break;  // active set empty: longest prefix ended everywhere
-  }
-  for (int w = 0; w < W; ++w) {
     int pid = w * 64 + lane;
-    if (pid < num_pods) scores[(size_t)b * num_pods + pid] = score[w];
+    if (pid < num_pods) scores[(size_t)b * num_pods + pid] = score;
   }
 }
 
