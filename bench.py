@@ -291,8 +291,8 @@ def main():
         request hashes."""
         hashes_t = ops.gpu_hash_chain_tr(
             call_tokens[call_idx % len(call_tokens)], parents, nchunks_t,
-            BLOCK_SIZE, KEYS_PER_PROMPT, 0)
-        return hashes_t.t().contiguous().view(-1)
+            BLOCK_SIZE, KEYS_PER_PROMPT, 0, 1)  # row_major: no transpose
+        return hashes_t.view(-1)
 
     def probe_score(hashes):
         if sharded is not None:
@@ -342,8 +342,8 @@ def main():
         def graph_body():
             hashes_t = ops.gpu_hash_chain_tr(
                 static_tok, parents, nchunks_t, BLOCK_SIZE,
-                KEYS_PER_PROMPT, 0)
-            hashes = hashes_t.t().contiguous().view(-1)
+                KEYS_PER_PROMPT, 0, 1)
+            hashes = hashes_t.view(-1)
             scores = ops.gpu_fused_score(
                 *index.table._t(), hashes, key_offsets, model_id,
                 no_filter, weights, num_pods, frozen_epoch_note,
@@ -452,7 +452,7 @@ def main():
         for _ in range(30):
             t0 = time.monotonic()
             hh = ops.gpu_hash_chain_tr(one_tok, one_par, one_nch,
-                                       BLOCK_SIZE, KEYS_PER_PROMPT, 0)
+                                       BLOCK_SIZE, KEYS_PER_PROMPT, 0, 1)
             if fused_fits:
                 sc = ops.gpu_fused_score(
                     *index.table._t(), hh.view(-1), one_off, model_id,

@@ -62,7 +62,7 @@ at::Tensor gpu_score_from_masks(at::Tensor, at::Tensor, at::Tensor, int64_t);
 std::vector<at::Tensor> gpu_hash_chain(at::Tensor, at::Tensor, at::Tensor,
                                        int64_t);
 at::Tensor gpu_hash_chain_tr(at::Tensor, at::Tensor, at::Tensor, int64_t,
-                             int64_t, int64_t);
+                             int64_t, int64_t, int64_t);
 void gpu_apply_events(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, int64_t, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, at::Tensor,
@@ -100,7 +100,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gpu_hash_chain", &kvidx::gpu_hash_chain);
   m.def("gpu_hash_chain_tr", &kvidx::gpu_hash_chain_tr,
         py::arg("tokens_t"), py::arg("parents"), py::arg("n_chunks"),
-        py::arg("block_size"), py::arg("max_chunks"), py::arg("ilp") = 0);
+        py::arg("block_size"), py::arg("max_chunks"), py::arg("ilp") = 0,
+        py::arg("row_major") = 0);
   m.def("gpu_apply_events", &kvidx::gpu_apply_events);
   m.def("gpu_apply_events_split", &kvidx::gpu_apply_events_split);
 #else

@@ -454,7 +454,8 @@ __global__ void k_hash_chain_tr(const int32_t* __restrict__ tokens_t,  // [T,B]
                                 const uint64_t* __restrict__ parents,  // [B]
                                 const int32_t* __restrict__ n_chunks,  // [B]
                                 int64_t B, int64_t L, int max_chunks,
-                                uint64_t* __restrict__ out) {  // [maxC,B]
+                                int row_major,
+                                uint64_t* __restrict__ out) {  // [maxC,B] or [B,maxC]
   const int64_t lane = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (lane >= L) return;
   int64_t b[ILP];
@@ -485,7 +486,13 @@ __global__ void k_hash_chain_tr(const int32_t* __restrict__ tokens_t,  // [T,B]
       const bool active = live[i] && c < mc[i];
       const uint64_t h2 = chunk_hash_fast(h[i], tok[i], BS);
       h[i] = active ? h2 : h[i];
-      if (active) out[(int64_t)c * B + b[i]] = h[i];
+      // row_major writes land scattered (stride maxC per lane) but the
+      // kernel is ALU-latency bound with idle bandwidth - writing the
+      // layout the fused-score kernel consumes saves a 128 MB/call
+      // transpose pass downstream.
+      if (active)
+        out[row_major ? (int64_t)b[i] * max_chunks + c
+                      : (int64_t)c * B + b[i]] = h[i];
     }
   }
 }
@@ -500,6 +507,7 @@ __global__ void k_hash_chain_tr_pf(const int32_t* __restrict__ tokens_t,
                                    const uint64_t* __restrict__ parents,
                                    const int32_t* __restrict__ n_chunks,
                                    int64_t B, int64_t L, int max_chunks,
+                                   int row_major,
                                    uint64_t* __restrict__ out) {
   const int64_t lane = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (lane >= L) return;
@@ -518,7 +526,9 @@ __global__ void k_hash_chain_tr_pf(const int32_t* __restrict__ tokens_t,
     const bool active = c < mc;
     const uint64_t h2 = chunk_hash_fast(h, cur, BS);
     h = active ? h2 : h;
-    if (active) out[(int64_t)c * B + lane] = h;
+    if (active)
+      out[row_major ? (int64_t)lane * max_chunks + c
+                    : (int64_t)c * B + lane] = h;
 #pragma unroll
     for (int j = 0; j < BS; ++j) cur[j] = nxt[j];
   }
@@ -922,12 +932,17 @@ std::vector<at::Tensor> gpu_hash_chain(at::Tensor tokens, at::Tensor tok_off,
 // int32 [B].  Returns out int64 [max_chunks, B] (transposed hashes).
 at::Tensor gpu_hash_chain_tr(at::Tensor tokens_t, at::Tensor parents,
                              at::Tensor n_chunks, int64_t block_size,
-                             int64_t max_chunks, int64_t ilp) {
+                             int64_t max_chunks, int64_t ilp,
+                             int64_t row_major) {
   TORCH_CHECK(tokens_t.is_cuda() && tokens_t.dtype() == at::kInt);
   TORCH_CHECK(tokens_t.dim() == 2, "tokens_t must be [T, B]");
   int64_t B = tokens_t.size(1);
   TORCH_CHECK(parents.numel() == B && n_chunks.numel() == B);
-  auto out = at::zeros({max_chunks, B}, parents.options());
+  // row_major=1: out is [B, max_chunks] (what the fused-score kernel
+  // consumes directly - no transpose pass); default [max_chunks, B].
+  auto out = at::zeros(row_major ? std::initializer_list<int64_t>{B, max_chunks}
+                                 : std::initializer_list<int64_t>{max_chunks, B},
+                       parents.options());
   if (B == 0 || max_chunks == 0) return out;
   // ILP trades wave count for per-lane chain interleave: only pay the
   // wave reduction when there are waves to spare (tuned by
@@ -944,6 +959,7 @@ at::Tensor gpu_hash_chain_tr(at::Tensor tokens_t, at::Tensor parents,
     hipLaunchKernelGGL(kern, dim3(blocks), dim3(threads), 0, STREAM,
                        tokens_t.data_ptr<int32_t>(), U64P(parents),
                        n_chunks.data_ptr<int32_t>(), B, L, (int)max_chunks,
+                       (int)row_major,
                        reinterpret_cast<uint64_t*>(out.data_ptr<int64_t>()));
   };
   bool done = false;

@@ -79,8 +79,8 @@ def main():
     ingest_stream = torch.cuda.Stream()
 
     def read_call():
-        hashes = ops.gpu_hash_chain_tr(toks, parents, nch, BS, K, 0)
-        hashes = hashes.t().contiguous().view(-1)
+        hashes = ops.gpu_hash_chain_tr(toks, parents, nch, BS, K, 0, 1)
+        hashes = hashes.view(-1)
         scores = ops.gpu_fused_score(
             *idx.table._t(), hashes, offs, model_id, no_filter, weights,
             num_pods, idx.table.next_epoch(), K,

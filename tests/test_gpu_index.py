@@ -491,3 +491,30 @@ class TestPinnedUploadOptIn:
                            gpu.lookup(keys, set()).items()}
             results[mode] = snap
         assert results["0"] == results["1"]
+
+
+class TestHashChainTrRowMajor:
+    def test_row_major_output_matches_transpose(self):
+        """row_major=1 writes [B, maxC] directly - must equal the
+        transposed [maxC, B] default (saves the 128 MB/call transpose
+        pass the fused-score consumer otherwise needs)."""
+        from llmd_kvcache_amd.ops import cpu_ext
+        from llmd_kvcache_amd.utils import hashing
+
+        mod = cpu_ext.require()
+        rng = random.Random(31)
+        B, C, BS = 97, 7, 16
+        tok = torch.tensor(
+            [[rng.randrange(0, 1 << 31) for _ in range(B)]
+             for _ in range(C * BS)], dtype=torch.int32, device="cuda")
+        parents = torch.tensor([_to_i64(hashing.init_hash(""))] * B,
+                               dtype=torch.int64, device="cuda")
+        nch = torch.tensor([rng.randrange(1, C + 1) for _ in range(B)],
+                           dtype=torch.int32, device="cuda")
+        col = mod.gpu_hash_chain_tr(tok, parents, nch, BS, C, 0, 0)
+        row = mod.gpu_hash_chain_tr(tok, parents, nch, BS, C, 0, 1)
+        assert col.shape == (C, B) and row.shape == (B, C)
+        assert torch.equal(col.t().contiguous(), row)
+        # prefetch A/B variant honors the flag too
+        row_pf = mod.gpu_hash_chain_tr(tok, parents, nch, BS, C, 8, 1)
+        assert torch.equal(row_pf, row)
