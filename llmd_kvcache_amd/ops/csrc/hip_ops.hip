@@ -967,12 +967,18 @@ at::Tensor gpu_hash_chain_tr(at::Tensor tokens_t, at::Tensor parents,
     done = true;
     // Measured (profiles/r01_chain_sweep.md): the plain single-chain
     // kernel with the force-unrolled token loop wins at every batch
-    // (0.95 ms/call vs 1.12-1.24 ms for the explicit double-buffer
-    // variant - the prefetch's register moves cost more than the load
-    // latency they hide once the unroll removed the gpr_idx waits).
+    // vs the explicit double-buffer prefetch variant - the prefetch's
+    // register moves cost more than the load latency they hide once the
+    // unroll removed the gpr_idx waits. (The original pf A/B runs also
+    // had a dispatch bug that gave pf 1/8 the lanes, flattering it 8x;
+    // the verdict only strengthens with the fix.)
     switch (want_ilp) {
       case 2: launch(k_hash_chain_tr<16, 2>); break;
-      case 8: launch(k_hash_chain_tr_pf<16>); break;  // A/B: prefetch
+      case 8:  // A/B: prefetch variant is one lane per prompt (no ILP)
+        L = B;
+        blocks = (int)((L + threads - 1) / threads);
+        launch(k_hash_chain_tr_pf<16>);
+        break;
       case 4: launch(k_hash_chain_tr<16, 4>); break;
       default: launch(k_hash_chain_tr<16, 1>); break;
     }
