@@ -185,13 +185,14 @@ class KvTable:
         fn(*self._t(), engine_hashes, model_id, pod_entries)
 
     def lookup(self, request_hashes, model_id, filter_words, num_pods,
-               sharded=True):
+               sharded=True, n_tiers=MAX_TIERS):
         fn = self.ops.gpu_lookup if self.is_cuda else self.ops.cpu_lookup
         cfg = self.cfg
         shard_id = cfg.shard_id if sharded else 0
         num_shards = cfg.num_shards if sharded else 1
         return fn(*self._t(), request_hashes, model_id, filter_words,
-                  num_pods, self.next_epoch(), shard_id, num_shards)
+                  num_pods, self.next_epoch(), shard_id, num_shards,
+                  n_tiers)
 
     def fused_score(self, hashes, counts_or_offsets, model_id, filter_words,
                     weights, num_pods, max_k=None, n_tiers=MAX_TIERS):
@@ -290,10 +291,13 @@ class TableIndex(Index):
         num_pods = self._num_pods_padded()
         hashes = self._hashes_tensor(request_keys)
         filt = self._filter_tensor(pod_identifier_set, num_pods)
-        found, masks = self.table.lookup(hashes, model_id, filt, num_pods)
+        found, masks = self.table.lookup(
+            hashes, model_id, filt, num_pods,
+            n_tiers=max(1, len(self.registry.id_to_tier)))
         found = found.cpu().tolist()
         masks_l = masks.cpu().tolist()  # one bulk D2H/convert, no per-item
         W = masks.shape[2]
+        n_tiers = masks.shape[1]
 
         result: Dict[Key, List[PodEntry]] = {}
         for i, key in enumerate(request_keys):
@@ -303,7 +307,7 @@ class TableIndex(Index):
             if f == 2:
                 break  # present-but-empty: chain cut (in_memory.go:118-121)
             entries: List[PodEntry] = []
-            for t in range(MAX_TIERS):
+            for t in range(n_tiers):
                 tier_name = (self.registry.id_to_tier[t]
                              if t < len(self.registry.id_to_tier) else None)
                 if tier_name is None:
@@ -374,7 +378,7 @@ class TableIndex(Index):
                 # LDS would overflow (huge fleet x long prompts): fall back
                 # to the two-kernel path - global-mask lookup + mask walk.
                 found, masks = self.table.lookup(hashes, model_id, filt,
-                                                 num_pods)
+                                                 num_pods, n_tiers=n_tiers)
                 del found
                 return self.table.ops.gpu_score_from_masks(
                     masks.contiguous(), counts_or_offsets, weights, num_pods)

@@ -25,7 +25,7 @@ std::vector<at::Tensor> cpu_lookup(at::Tensor, at::Tensor, at::Tensor,
                                    at::Tensor, at::Tensor, at::Tensor,
                                    at::Tensor, int64_t, at::Tensor, int64_t,
                                    at::Tensor, int64_t, int64_t, int64_t,
-                                   int64_t);
+                                   int64_t, int64_t);
 at::Tensor cpu_fused_score(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                            at::Tensor, at::Tensor, at::Tensor, int64_t,
                            at::Tensor, at::Tensor, int64_t, at::Tensor,
@@ -53,7 +53,7 @@ std::vector<at::Tensor> gpu_lookup(at::Tensor, at::Tensor, at::Tensor,
                                    at::Tensor, at::Tensor, at::Tensor,
                                    at::Tensor, int64_t, at::Tensor, int64_t,
                                    at::Tensor, int64_t, int64_t, int64_t,
-                                   int64_t);
+                                   int64_t, int64_t);
 at::Tensor gpu_fused_score(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                            at::Tensor, at::Tensor, at::Tensor, int64_t,
                            at::Tensor, at::Tensor, int64_t, at::Tensor,

@@ -298,6 +298,8 @@ at::Tensor cpu_score_from_masks(at::Tensor masks, at::Tensor offsets,
                                 at::Tensor weights, int64_t num_pods) {
   auto m = masks.contiguous();
   auto off = offsets.contiguous();
+  TORCH_CHECK(m.dim() == 3, "masks must be [Ktot, T, W]");
+  const int64_t T = m.size(1);  // tier planes (sized by caller)
   const uint64_t* mp = reinterpret_cast<uint64_t*>(m.data_ptr<int64_t>());
   const int32_t* op = off.data_ptr<int32_t>();
   const float* wts = weights.data_ptr<float>();
@@ -312,11 +314,11 @@ at::Tensor cpu_score_from_masks(at::Tensor masks, at::Tensor offsets,
       float* out = sp + b * num_pods;
       int64_t K = op[b + 1] - op[b];
       for (int64_t k = 0; k < K; ++k) {
-        const uint64_t* mk = mp + (size_t)(op[b] + k) * MAX_TIERS * W;
+        const uint64_t* mk = mp + (size_t)(op[b] + k) * T * W;
         bool any = false;
         for (int64_t w = 0; w < W; ++w) {
           uint64_t cur = 0;
-          for (int t = 0; t < MAX_TIERS; ++t) cur |= mk[t * W + w];
+          for (int t = 0; t < T; ++t) cur |= mk[t * W + w];
           uint64_t act = (k == 0) ? cur : (active[w] & cur);
           active[w] = act;
           if (act) any = true;
@@ -325,7 +327,7 @@ at::Tensor cpu_score_from_masks(at::Tensor masks, at::Tensor offsets,
             int bit = __builtin_ctzll(bits);
             bits &= bits - 1;
             float wmax = 0.f;
-            for (int t = 0; t < MAX_TIERS; ++t)
+            for (int t = 0; t < T; ++t)
               if ((mk[t * W + w] >> bit) & 1) wmax = std::max(wmax, wts[t]);
             out[w * 64 + bit] += wmax;
           }
@@ -382,7 +384,8 @@ std::vector<at::Tensor> cpu_lookup(at::Tensor keys, at::Tensor meta,
                                    at::Tensor request_hashes, int64_t model_id,
                                    at::Tensor filter_words, int64_t num_pods,
                                    int64_t epoch, int64_t shard_id,
-                                   int64_t num_shards) {
+                                   int64_t num_shards,
+                                   int64_t n_tiers) {
   auto v = make_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
                      (int)pods_per_key);
   auto rh = request_hashes.contiguous();
@@ -394,8 +397,13 @@ std::vector<at::Tensor> cpu_lookup(at::Tensor keys, at::Tensor meta,
                                         filter_words.data_ptr<int64_t>())
                                   : nullptr;
 
+  if (n_tiers < 1) n_tiers = 1;
+  if (n_tiers > MAX_TIERS) n_tiers = MAX_TIERS;
   auto found = at::zeros({K}, at::kByte);
-  auto masks = at::zeros({K, MAX_TIERS, W}, at::kLong);
+  // masks sized by the REGISTERED tier count (usually 1-2): halves-to-
+  // quarters the fallback path's mask traffic and the sharded
+  // all_reduce bytes vs a fixed MAX_TIERS=4 plane count.
+  auto masks = at::zeros({K, n_tiers, W}, at::kLong);
   uint8_t* fp = found.data_ptr<uint8_t>();
   uint64_t* mp = reinterpret_cast<uint64_t*>(masks.data_ptr<int64_t>());
 
@@ -414,9 +422,9 @@ std::vector<at::Tensor> cpu_lookup(at::Tensor keys, at::Tensor meta,
       any_raw = true;
       uint32_t pid = pod_entry_id(e);
       uint32_t tier = pod_entry_tier(e);
-      if (pid >= (uint32_t)num_pods || tier >= MAX_TIERS) continue;
+      if (pid >= (uint32_t)num_pods || tier >= (uint32_t)n_tiers) continue;
       if (has_filter && !((fw[pid / 64] >> (pid % 64)) & 1)) continue;
-      mp[(k * MAX_TIERS + tier) * W + pid / 64] |= 1ull << (pid % 64);
+      mp[(k * n_tiers + tier) * W + pid / 64] |= 1ull << (pid % 64);
       any_visible = true;
     }
     (void)any_visible;

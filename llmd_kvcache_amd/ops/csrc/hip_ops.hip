@@ -284,6 +284,7 @@ __global__ void k_lookup_masks(DevTable v, const uint64_t* __restrict__ rh,
                                const uint64_t* __restrict__ filter,
                                int has_filter, int num_pods, int W,
                                int32_t epoch, int shard_id, int num_shards,
+                               int n_tiers,
                                uint8_t* __restrict__ found,
                                unsigned long long* __restrict__ masks) {
   int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -291,11 +292,12 @@ __global__ void k_lookup_masks(DevTable v, const uint64_t* __restrict__ rh,
   if (num_shards > 1 &&
       (int)(remap_hash(rh[k]) % (uint64_t)num_shards) != shard_id)
     return;  // unowned key: contributes zero masks on this shard
-  unsigned long long* mk = masks + (size_t)k * MAX_TIERS * W;
+  unsigned long long* mk = masks + (size_t)k * n_tiers * W;
   int f = dev_probe_collect(
       v, rh[k], model, filter, has_filter, num_pods, W, epoch,
       [&](uint32_t tier, uint32_t pid) {
-        atomicOr(&mk[tier * W + pid / 64], 1ull << (pid % 64));
+        if ((int)tier < n_tiers)  // interned => always true
+          atomicOr(&mk[tier * W + pid / 64], 1ull << (pid % 64));
       });
   found[k] = (uint8_t)f;
 }
@@ -373,7 +375,7 @@ __global__ void __launch_bounds__(256) k_fused_score(
 __global__ void __launch_bounds__(64) k_score_from_masks(
     const unsigned long long* __restrict__ masks,  // [Ktot, T, W]
     const int32_t* __restrict__ offsets,           // [B+1]
-    const float* __restrict__ weights, int num_pods, int W,
+    const float* __restrict__ weights, int T, int num_pods, int W,
     float* __restrict__ scores) {
   const int b = blockIdx.x;
   const int w_lo = blockIdx.y * 16;
@@ -389,12 +391,12 @@ __global__ void __launch_bounds__(64) k_score_from_masks(
   }
   for (int k = 0; k < K; ++k) {
     const unsigned long long* mk =
-        masks + (size_t)(offsets[b] + k) * MAX_TIERS * W;
+        masks + (size_t)(offsets[b] + k) * T * W;
     bool any = false;
     for (int w = 0; w < WG; ++w) {
       int cur = 0;
       float wmax = 0.f;
-      for (int t = 0; t < MAX_TIERS; ++t) {
+      for (int t = 0; t < T; ++t) {
         if ((mk[t * W + w_lo + w] >> lane) & 1) {
           cur = 1;
           wmax = fmaxf(wmax, weights[t]);
@@ -833,14 +835,17 @@ std::vector<at::Tensor> gpu_lookup(at::Tensor keys, at::Tensor meta,
                                    at::Tensor request_hashes, int64_t model_id,
                                    at::Tensor filter_words, int64_t num_pods,
                                    int64_t epoch, int64_t shard_id,
-                                   int64_t num_shards) {
+                                   int64_t num_shards, int64_t n_tiers) {
   auto v = dev_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
                     pods_per_key);
   int64_t K = request_hashes.numel();
   int64_t W = (num_pods + 63) / 64;
   bool has_filter = filter_words.numel() > 0;
+  if (n_tiers < 1) n_tiers = 1;
+  if (n_tiers > MAX_TIERS) n_tiers = MAX_TIERS;
   auto found = at::zeros({K}, request_hashes.options().dtype(at::kByte));
-  auto masks = at::zeros({K, MAX_TIERS, W}, request_hashes.options());
+  // tier planes sized by the registered tier count (see cpu_lookup)
+  auto masks = at::zeros({K, n_tiers, W}, request_hashes.options());
   if (K == 0) return {found, masks};
   int threads = 256;
   int blocks = (int)((K + threads - 1) / threads);
@@ -849,7 +854,7 @@ std::vector<at::Tensor> gpu_lookup(at::Tensor keys, at::Tensor meta,
       U64P(request_hashes), K, (uint32_t)model_id,
       has_filter ? U64P(filter_words) : nullptr, has_filter ? 1 : 0,
       (int)num_pods, (int)W, (int32_t)epoch, (int)shard_id, (int)num_shards,
-      found.data_ptr<uint8_t>(),
+      (int)n_tiers, found.data_ptr<uint8_t>(),
       reinterpret_cast<unsigned long long*>(masks.data_ptr<int64_t>()));
   return {found, masks};
 }
@@ -890,14 +895,16 @@ at::Tensor gpu_score_from_masks(at::Tensor masks, at::Tensor offsets,
   int64_t B = offsets.numel() - 1;
   int64_t W = (num_pods + 63) / 64;
   TORCH_CHECK(W <= 64, "score_from_masks supports up to 4096 pods");
+  TORCH_CHECK(masks.dim() == 3, "masks must be [Ktot, T, W]");
+  int64_t T = masks.size(1);  // tier planes, sized by the lookup
   auto scores = at::zeros({B, num_pods}, masks.options().dtype(at::kFloat));
   if (B == 0) return scores;
   int wgroups = (int)((W + 15) / 16);
   hipLaunchKernelGGL(
       k_score_from_masks, dim3((int)B, wgroups), dim3(64), 0, STREAM,
       reinterpret_cast<const unsigned long long*>(masks.data_ptr<int64_t>()),
-      offsets.data_ptr<int32_t>(), weights.data_ptr<float>(), (int)num_pods,
-      (int)W, scores.data_ptr<float>());
+      offsets.data_ptr<int32_t>(), weights.data_ptr<float>(), (int)T,
+      (int)num_pods, (int)W, scores.data_ptr<float>());
   return scores;
 }
 

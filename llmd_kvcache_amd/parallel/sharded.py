@@ -87,7 +87,9 @@ class ShardedIndex:
         if weights is None:
             weights = self.local.tier_weights()
 
-        found, masks = self.local.table.lookup(hashes, model_id, filt, num_pods)
+        found, masks = self.local.table.lookup(
+            hashes, model_id, filt, num_pods,
+            n_tiers=max(1, len(self.local.registry.id_to_tier)))
         del found  # chain-cut semantics handled by the scoring walk
         dist.all_reduce(masks, op=dist.ReduceOp.SUM, group=self.group)
 
