@@ -27,7 +27,9 @@ import torch.distributed as dist
 from ..kvblock.gpu_index import _to_i64
 from ..kvblock.token_processor import ChunkedTokenDatabase
 from ..kvevents.events import decode_event_batch
-from .sharded import ShardedIndex
+from .sharded import ShardedIndex, check_registry_sync
+
+SYNC_CHECK_INTERVAL = 64  # apply_messages calls between registry audits
 
 
 class ShardedIndexService:
@@ -41,6 +43,7 @@ class ShardedIndexService:
         self.rank = sharded.rank
         self.group = sharded.group
         self._running = True
+        self._applies = 0
 
     # -- collective plumbing -------------------------------------------
     def _broadcast(self, op):
@@ -67,6 +70,11 @@ class ShardedIndexService:
             offs = torch.tensor(offsets, dtype=torch.int32)
             scores = self.sharded.sharded_scores(h, offs, model, set(pods))
             return scores
+        if kind == "sync_check":
+            # ROADMAP #8: loud detection of registry divergence before it
+            # can misattribute merged-mask scores
+            check_registry_sync(self.sharded.local, self.group)
+            return None
         if kind == "stop":
             self._running = False
             return None
@@ -95,6 +103,18 @@ class ShardedIndexService:
         exactly what the ZMQ subscriber hands the events pool."""
         assert self.rank == 0
         op = ("events", list(messages))
+        self._broadcast(op)
+        self._dispatch(op)
+        self._applies += 1
+        if self._applies % SYNC_CHECK_INTERVAL == 0:
+            self.check_sync()
+
+    def check_sync(self) -> None:
+        """Audit that pod/model/tier id spaces still agree on every rank
+        (auto-run every SYNC_CHECK_INTERVAL applies; callable any time).
+        Raises RuntimeError on divergence - on every rank."""
+        assert self.rank == 0
+        op = ("sync_check",)
         self._broadcast(op)
         self._dispatch(op)
 

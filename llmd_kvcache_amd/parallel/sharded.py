@@ -119,3 +119,37 @@ class ShardedIndex:
             hashes, offsets, request_keys[0].model_name, pod_identifier_set
         )
         return self.local.scores_to_map(scores)[0]
+
+
+def registry_fingerprint(registry) -> int:
+    """Order-sensitive FNV-1a fingerprint of the interned id spaces.
+
+    Mask merging assumes pod/model/tier ids agree on every rank (the
+    replicated event stream assigns them identically).  If a rank ever
+    diverged - an error path that dropped one event, a locally-registered
+    extra pod - merged masks would silently attribute hits to the wrong
+    pods.  ROADMAP #8: detect loudly instead."""
+    from ..utils.hashing import fnv1a_64
+
+    acc = b"\x00".join(
+        "\x1f".join(ns).encode("utf-8")
+        for ns in (registry.id_to_pod, registry.id_to_model,
+                   registry.id_to_tier)
+    )
+    return fnv1a_64(acc)
+
+
+def check_registry_sync(index, group=None) -> None:
+    """Raise on cross-rank registry divergence (call between batches -
+    it is one tiny all_gather, latency-bound)."""
+    fp = registry_fingerprint(index.registry)
+    world = dist.get_world_size(group)
+    fps = [None] * world
+    dist.all_gather_object(fps, fp, group=group)
+    if any(f != fps[0] for f in fps):
+        raise RuntimeError(
+            f"registry divergence across ranks (fingerprints {fps}): "
+            "pod/model/tier id spaces no longer agree - scores from "
+            "merged masks would be misattributed. A rank dropped or "
+            "reordered events; rebuild the index from the stream."
+        )

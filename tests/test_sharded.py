@@ -228,3 +228,36 @@ def test_sharded_service_bridge():
     assert scores == {"vllm-pod-1": 8.0}
     assert filtered == {}
     assert results[1] > 0  # rank 1 holds part of the index
+
+
+def _body_registry_sync(rank, world_size):
+    from llmd_kvcache_amd.kvblock.gpu_index import TableIndexConfig
+    from llmd_kvcache_amd.parallel.sharded import (ShardedIndex,
+                                                   check_registry_sync,
+                                                   registry_fingerprint)
+
+    sharded = ShardedIndex(TableIndexConfig(capacity=1 << 10,
+                                            pods_per_key=4))
+    # identical interning order on every rank -> fingerprints agree
+    for i in range(5):
+        sharded.registry.pod_id(f"pod-{i}")
+    sharded.registry.model_id("m")
+    check_registry_sync(sharded.local)  # must not raise
+    fp_before = registry_fingerprint(sharded.registry)
+
+    # rank 1 interns an extra pod (simulating a dropped/extra event):
+    # EVERY rank must now raise, not just the divergent one
+    if rank == 1:
+        sharded.registry.pod_id("pod-ghost")
+        assert registry_fingerprint(sharded.registry) != fp_before
+    try:
+        check_registry_sync(sharded.local)
+        return "no-raise"
+    except RuntimeError as e:
+        assert "registry divergence" in str(e)
+        return "raised"
+
+
+def test_registry_sync_guard():
+    results = run_distributed("_body_registry_sync")
+    assert set(results.values()) == {"raised"}
