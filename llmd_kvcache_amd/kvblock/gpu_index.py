@@ -208,6 +208,27 @@ class KvTable:
             *self._t(), hashes, counts_or_offsets, model_id, filter_words,
             weights, num_pods, self.next_epoch())
 
+    def compact(self, new_capacity: Optional[int] = None) -> None:
+        """Rehash every live entry into a fresh tensor bundle, dropping
+        tombstones and dead probe windows (ROADMAP #9 - long-running
+        tables). Stamps (approximate-LRU) and pod rows carry over;
+        new_capacity (power of two) allows growing/shrinking. Costs 2x
+        table memory transiently; callers serialize against writes."""
+        import copy
+
+        cfg = copy.copy(self.cfg)
+        if new_capacity is not None:
+            if new_capacity & (new_capacity - 1):
+                raise ValueError("capacity must be a power of two")
+            cfg.capacity = new_capacity
+        fresh = KvTable(cfg)
+        fn = self.ops.gpu_compact if self.is_cuda else self.ops.cpu_compact
+        fn(*self._t(), *fresh._t()[:-1])
+        self.cfg = cfg
+        for name in ("keys", "meta", "stamp", "pods",
+                     "e_keys", "e_meta", "e_vals"):
+            setattr(self, name, getattr(fresh, name))
+
     def get_request_keys(self, engine_hashes, model_id):
         fn = (self.ops.gpu_get_request_keys if self.is_cuda
               else self.ops.cpu_get_request_keys)
@@ -400,6 +421,12 @@ class TableIndex(Index):
         for i, name in enumerate(self.registry.id_to_tier):
             w[i] = weight_map.get(name, 1.0)
         return torch.tensor(w, dtype=torch.float32, device=self.device)
+
+    def compact(self, new_capacity: Optional[int] = None) -> None:
+        """Reclaim tombstoned slots (and optionally resize) - see
+        KvTable.compact. Takes the write lock."""
+        with self._write_lock:
+            self.table.compact(new_capacity)
 
     def save(self, path: str) -> None:
         """Checkpoint the table + string registries to a file."""

@@ -18,6 +18,9 @@ void cpu_insert(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                 at::Tensor, at::Tensor, int64_t, at::Tensor, at::Tensor,
                 int64_t, at::Tensor, int64_t, int64_t, int64_t, int64_t);
 at::Tensor cpu_score_from_masks(at::Tensor, at::Tensor, at::Tensor, int64_t);
+void cpu_compact(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                 at::Tensor, at::Tensor, int64_t, at::Tensor, at::Tensor,
+                 at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor);
 void cpu_evict(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                at::Tensor, at::Tensor, int64_t, at::Tensor, int64_t,
                at::Tensor);
@@ -41,6 +44,9 @@ std::vector<at::Tensor> cpu_get_request_keys(at::Tensor, at::Tensor,
 void gpu_insert(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                 at::Tensor, at::Tensor, int64_t, at::Tensor, at::Tensor,
                 int64_t, at::Tensor, int64_t, int64_t, int64_t, int64_t);
+void gpu_compact(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                 at::Tensor, at::Tensor, int64_t, at::Tensor, at::Tensor,
+                 at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor);
 void gpu_evict(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                at::Tensor, at::Tensor, int64_t, at::Tensor, int64_t,
                at::Tensor);
@@ -84,6 +90,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hash_chain_batch", &kvidx::hash_chain_batch);
   m.def("tokens_to_chunk_hashes_fast", &kvidx::tokens_to_chunk_hashes_fast);
   m.def("cpu_insert", &kvidx::cpu_insert);
+  m.def("cpu_compact", &kvidx::cpu_compact);
   m.def("cpu_evict", &kvidx::cpu_evict);
   m.def("cpu_lookup", &kvidx::cpu_lookup);
   m.def("cpu_fused_score", &kvidx::cpu_fused_score);
@@ -92,6 +99,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
 #ifdef KVIDX_WITH_HIP
   m.attr("HAS_HIP") = true;
   m.def("gpu_insert", &kvidx::gpu_insert);
+  m.def("gpu_compact", &kvidx::gpu_compact);
   m.def("gpu_evict", &kvidx::gpu_evict);
   m.def("gpu_get_request_keys", &kvidx::gpu_get_request_keys);
   m.def("gpu_lookup", &kvidx::gpu_lookup);

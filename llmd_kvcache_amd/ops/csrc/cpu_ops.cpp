@@ -258,6 +258,38 @@ std::vector<at::Tensor> hash_chain_batch(at::Tensor tokens, at::Tensor offsets,
   return {out, chunk_off};
 }
 
+// Tombstone compaction (ROADMAP #9): rehash every live entry of `old`
+// into the freshly-zeroed `nw` bundle (optionally a different power-of-
+// two capacity). Stamps (approximate-LRU epochs) and pod rows are
+// carried over verbatim; tombstones and dead probe windows disappear.
+void cpu_compact(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
+                 at::Tensor pods, at::Tensor e_keys, at::Tensor e_meta,
+                 at::Tensor e_vals, int64_t pods_per_key,
+                 at::Tensor n_keys, at::Tensor n_meta, at::Tensor n_stamp,
+                 at::Tensor n_pods, at::Tensor n_e_keys, at::Tensor n_e_meta,
+                 at::Tensor n_e_vals) {
+  auto ov = make_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
+                      (int)pods_per_key);
+  auto nv = make_view(n_keys, n_meta, n_stamp, n_pods, n_e_keys, n_e_meta,
+                      n_e_vals, (int)pods_per_key);
+  int64_t cap = keys.numel();
+  for (int64_t i = 0; i < cap; ++i) {
+    uint32_t m = ov.meta[i];
+    if (!(m & META_OCC) || (m & META_TOMB)) continue;
+    int64_t s2 = table_put(nv, (uint64_t)ov.keys[i], m & META_MODEL_MASK,
+                           ov.stamp[i]);
+    std::memcpy(nv.pods + s2 * nv.pods_per_key,
+                ov.pods + i * ov.pods_per_key,
+                sizeof(uint32_t) * nv.pods_per_key);
+  }
+  for (int64_t i = 0; i < cap; ++i) {
+    uint32_t m = ov.e_meta[i];
+    if (!(m & META_OCC) || (m & META_TOMB)) continue;
+    emap_put(nv, (uint64_t)ov.e_keys[i], m & META_MODEL_MASK,
+             (uint64_t)ov.e_vals[i], 0);
+  }
+}
+
 // shard_id/num_shards: the main table stores only keys it owns
 // (request_hash % num_shards == shard_id); the engine->request map is
 // replicated on every shard so parent-chain stitching never needs a

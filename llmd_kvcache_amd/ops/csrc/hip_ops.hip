@@ -243,6 +243,29 @@ __global__ void k_insert(DevTable v, const uint64_t* __restrict__ eh,
     dev_pod_set_add(v, slot, entries[j], epoch);
 }
 
+// Tombstone compaction (ROADMAP #9): thread per OLD slot rehashes live
+// entries into a fresh bundle (lock-free via the usual CAS claims; old
+// (key,model) pairs are unique so racers only contend on free slots).
+__global__ void k_compact_main(DevTable ov, DevTable nv, int64_t cap) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= cap) return;
+  uint32_t m = ov.meta[i];
+  if (!(m & META_OCC) || (m & META_TOMB)) return;
+  int64_t s2 = dev_table_put(nv, ov.keys[i], m & META_MODEL_MASK,
+                             ov.stamp[i]);
+  uint32_t* dst = nv.pods + s2 * nv.pods_per_key;
+  const uint32_t* src = ov.pods + i * ov.pods_per_key;
+  for (int j = 0; j < nv.pods_per_key; ++j) dst[j] = src[j];
+}
+
+__global__ void k_compact_emap(DevTable ov, DevTable nv, int64_t cap) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= cap) return;
+  uint32_t m = ov.e_meta[i];
+  if (!(m & META_OCC) || (m & META_TOMB)) return;
+  dev_emap_put(nv, ov.e_keys[i], m & META_MODEL_MASK, ov.e_vals[i]);
+}
+
 __global__ void k_evict(DevTable v, const uint64_t* __restrict__ eh,
                         int64_t n, uint32_t model,
                         const uint32_t* __restrict__ entries, int n_entries) {
@@ -789,6 +812,25 @@ void gpu_insert(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
                          pod_entries.data_ptr<int32_t>()),
                      (int)pod_entries.numel(), (int32_t)epoch,
                      (int)shard_id, (int)num_shards, (int)emap_write);
+}
+
+void gpu_compact(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
+                 at::Tensor pods, at::Tensor e_keys, at::Tensor e_meta,
+                 at::Tensor e_vals, int64_t pods_per_key,
+                 at::Tensor n_keys, at::Tensor n_meta, at::Tensor n_stamp,
+                 at::Tensor n_pods, at::Tensor n_e_keys, at::Tensor n_e_meta,
+                 at::Tensor n_e_vals) {
+  auto ov = dev_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
+                     pods_per_key);
+  auto nv = dev_view(n_keys, n_meta, n_stamp, n_pods, n_e_keys, n_e_meta,
+                     n_e_vals, pods_per_key);
+  int64_t cap = keys.numel();
+  int threads = 256;
+  int blocks = (int)((cap + threads - 1) / threads);
+  hipLaunchKernelGGL(k_compact_main, dim3(blocks), dim3(threads), 0, STREAM,
+                     ov, nv, cap);
+  hipLaunchKernelGGL(k_compact_emap, dim3(blocks), dim3(threads), 0, STREAM,
+                     ov, nv, cap);
 }
 
 void gpu_evict(at::Tensor keys, at::Tensor meta, at::Tensor stamp,

@@ -518,3 +518,38 @@ class TestHashChainTrRowMajor:
         # prefetch A/B variant honors the flag too
         row_pf = mod.gpu_hash_chain_tr(tok, parents, nch, BS, C, 8, 1)
         assert torch.equal(row_pf, row)
+
+
+class TestCompactionGpu:
+    def test_gpu_compact_matches_cpu(self):
+        """Differential: same churn workload on GPU and CPU tables,
+        compact both, lookups must agree (and match pre-compact)."""
+        rng = random.Random(77)
+        cpu = NativeIndex(TableIndexConfig(capacity=1 << 11, pods_per_key=4))
+        gpu = GpuIndex(GpuIndexConfig(capacity=1 << 11, pods_per_key=4))
+        keys = [Key(MODEL, 3000 + i) for i in range(300)]
+        for idx_, k in enumerate(keys):
+            e = [Key(MODEL, 600000 + idx_)]
+            pods = [PodEntry(f"pod-{rng.randrange(9)}", "gpu")]
+            cpu.add(e, [k], pods)
+            gpu.add(e, [k], pods)
+        for i in range(0, 300, 4):
+            for t in (cpu, gpu):
+                t.evict(Key(MODEL, 600000 + i),
+                        [PodEntry(f"pod-{i % 9}", "gpu")])
+        torch.cuda.synchronize()
+
+        def snap(t):
+            return {k: sorted(map(tuple, v))
+                    for k, v in t.lookup(keys, set()).items()}
+
+        before = snap(gpu)
+        assert before == snap(cpu)
+        cpu.compact(new_capacity=1 << 12)
+        gpu.compact(new_capacity=1 << 12)
+        torch.cuda.synchronize()
+        assert snap(gpu) == before
+        assert snap(cpu) == before
+        tombs = int(((gpu.table.meta.cpu() & 0x40000000) != 0).sum())
+        assert tombs == 0
+        assert gpu.get_request_key(Key(MODEL, 600001)) == keys[1]

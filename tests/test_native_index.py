@@ -265,3 +265,46 @@ def test_custom_tier_weights_through_indexer_fused_path():
     scores = idx.get_pod_scores(None, "abcd" * 8, "m", [])
     # key0: max(cpu .8, disk .25)=0.8; key1: disk 0.25 -> 1.05
     assert scores["pod-d"] == pytest.approx(1.05, abs=1e-6)
+
+
+class TestCompaction:
+    def _fill_and_churn(self, nat, n=400, evict_every=3):
+        keys = [Key(MODEL, 1000 + i) for i in range(n)]
+        for i, k in enumerate(keys):
+            nat.add([Key(MODEL, 500000 + i)], [k],
+                    [PodEntry(f"pod-{i % 7}", "gpu")])
+        evicted = []
+        for i in range(0, n, evict_every):
+            nat.evict(Key(MODEL, 500000 + i), [PodEntry(f"pod-{i % 7}", "gpu")])
+            evicted.append(i)
+        return keys, set(evicted)
+
+    def _snapshot(self, nat, keys):
+        live = nat.lookup(keys, set())
+        return {k: sorted(map(tuple, v)) for k, v in live.items()}
+
+    def test_compact_preserves_state_and_drops_tombstones(self):
+        nat = NativeIndex(TableIndexConfig(capacity=1 << 11, pods_per_key=4))
+        keys, evicted = self._fill_and_churn(nat)
+        before = self._snapshot(nat, keys)
+        tombs_before = int(((nat.table.meta & 0x40000000) != 0).sum())
+        assert tombs_before > 0  # churn really created tombstones
+        nat.compact()
+        assert self._snapshot(nat, keys) == before
+        tombs_after = int(((nat.table.meta & 0x40000000) != 0).sum())
+        assert tombs_after == 0
+        # engine map survives (retained semantics incl. evicted mappings)
+        assert nat.get_request_key(Key(MODEL, 500001)) == Key(MODEL, 1001)
+
+    def test_compact_resize(self):
+        nat = NativeIndex(TableIndexConfig(capacity=1 << 11, pods_per_key=4))
+        keys, _ = self._fill_and_churn(nat, n=300)
+        before = self._snapshot(nat, keys)
+        nat.compact(new_capacity=1 << 12)  # grow
+        assert nat.table.keys.numel() == 1 << 12
+        assert self._snapshot(nat, keys) == before
+        nat.compact(new_capacity=1 << 11)  # shrink back
+        assert nat.table.keys.numel() == 1 << 11
+        assert self._snapshot(nat, keys) == before
+        with pytest.raises(ValueError):
+            nat.compact(new_capacity=3000)
