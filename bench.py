@@ -156,11 +156,11 @@ def main():
     ap.add_argument("--batch", type=int, default=32768,
                     help="prompts per scoring call (sub-batch). The chain "
                          "kernel's wall time is latency-bound and near-"
-                         "constant in batch (idle SIMDs absorb extra "
-                         "prompts), so large batches raise QPS almost "
-                         "linearly: 4096->3.7M, 8192->7.2M, 16384->13.5M, "
-                         "32768->22.5M scores/s (p50 1.10->1.37 ms); 65536 "
-                         "adds only ~14%% at 2x the latency.")
+                         "constant up to ~512 waves, so QPS rises almost "
+                         "linearly to the default: 4096->3.7M, 8192->7.2M, "
+                         "16384->13.5M, 32768->22.5M scores/s (p50 "
+                         "1.10->1.37 ms). Beyond that wall grows with "
+                         "batch: 131072 peaks ~30M at p50 4.5 ms.")
     ap.add_argument("--calls-per-step", type=int, default=2)
     ap.add_argument("--device", default=None, help="cpu to force CPU tables")
     ap.add_argument("--blocks", type=int, default=NUM_BLOCKS)
