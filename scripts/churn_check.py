@@ -33,6 +33,11 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--events", type=int, default=20000)
     ap.add_argument("--seed", type=int, default=1234)
+    ap.add_argument("--compact-every", type=int, default=0,
+                    help="run TableIndex.compact() on the GPU index every "
+                         "N events mid-stream (0 = never): proves "
+                         "compaction preserves semantics under the live "
+                         "write path")
     args = ap.parse_args()
     assert torch.cuda.is_available()
 
@@ -78,6 +83,11 @@ def main():
             for p, m, evs in batch:
                 digest_events(ref, tp, p, m, evs)
             batch = []
+            if args.compact_every and e // args.compact_every != \
+                    (e - 63) // args.compact_every:
+                torch.cuda.synchronize()
+                gpu.compact()
+                print(f"compacted at event {e}", flush=True)
     torch.cuda.synchronize()
     print(f"applied {args.events} events ({next_hash - 1} blocks stored, "
           f"{n_removed} removed)", flush=True)
