@@ -16,7 +16,7 @@ Bit-compatible with the reference schema (pkg/kvcache/kvevents/events.go):
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, List, Optional
 
 import msgpack
