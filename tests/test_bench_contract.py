@@ -1,0 +1,54 @@
+"""Driver-contract smoke for bench.py: tiny CPU configurations run as
+subprocesses; the final stdout line must be ONE JSON object carrying the
+fields the round driver parses (BASELINE.json metric shape)."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+TINY = ["--device", "cpu", "--batch", "32", "--blocks", "2048",
+        "--steps", "2", "--warmup", "1", "--calls-per-step", "1"]
+
+REQUIRED = {"metric", "value", "unit", "n_gpus", "steps", "warmup",
+            "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+            "dtype", "data", "config"}
+REQUIRED_CONFIG = {"model", "global_batch", "seq_len", "parallelism"}
+
+
+def run_bench(extra):
+    proc = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "bench.py")] + TINY + extra,
+        capture_output=True, text=True, timeout=300, cwd=ROOT)
+    assert proc.returncode == 0, proc.stderr[-800:]
+    line = proc.stdout.strip().splitlines()[-1]
+    return json.loads(line)
+
+
+@pytest.mark.timeout(600)
+class TestBenchContract:
+    def test_default_mode_json_shape(self):
+        d = run_bench([])
+        assert REQUIRED <= set(d)
+        assert REQUIRED_CONFIG <= set(d["config"])
+        assert d["metric"].startswith("Score() QPS")
+        assert isinstance(d["value"], (int, float)) and d["value"] > 0
+        assert d["higher_is_better"] is True
+        assert d["scaling"] == "weak"
+        assert d["data"] == "synthetic"
+        assert d["config"]["seq_len"] == 8192
+        assert d["config"]["block_size"] == 16
+
+    def test_sharded_mode(self):
+        d = run_bench(["--force-sharded"])
+        assert d["scaling"] == "strong"
+        assert d["config"]["parallelism"] == "shard1"
+        assert d["value"] > 0
+
+    def test_prefix_frac_and_pods_flags(self):
+        d = run_bench(["--prefix-frac", "1.0", "--pods", "8"])
+        assert d["config"]["prefix_frac"] == 1.0
+        assert d["config"]["num_pods"] == 8
