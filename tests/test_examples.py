@@ -80,3 +80,17 @@ class TestOnlineServiceConfig:
     def test_unknown_backend_exits(self, monkeypatch):
         with pytest.raises(SystemExit):
             self._build(monkeypatch, "bogus")
+
+
+def test_sharded_service_main_importable():
+    """The torchrun entry point from docs/deployment.md must stay
+    importable (its collective wiring is exercised by test_sharded's
+    service-bridge test)."""
+    import importlib.util
+
+    spec = importlib.util.spec_from_file_location(
+        "sharded_service_main",
+        os.path.join(EXAMPLES, "sharded_service_main.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    assert callable(mod.main)
