@@ -32,11 +32,16 @@ class TokenProcessorConfig:
     block_size: int = DEFAULT_BLOCK_SIZE
     # Aligned with vLLM's PYTHONHASHSEED; see token_processor.go:36-40.
     hash_seed: str = ""
+    # Pluggable chain hash (utils.hashing.CHAIN_ALGOS): "fnv-64a" is
+    # current vLLM/reference behavior and the only algo with C++/HIP
+    # fast paths; alternatives run on the Python path.
+    hash_algo: str = "fnv-64a"
     _init_hash: Optional[int] = field(default=None, repr=False)
 
     def init_hash(self) -> int:
         if self._init_hash is None:
-            self._init_hash = hashing.init_hash(self.hash_seed)
+            self._init_hash = hashing.CHAIN_ALGOS[self.hash_algo][1](
+                self.hash_seed)
         return self._init_hash
 
 
@@ -63,14 +68,15 @@ class ChunkedTokenDatabase:
         n_chunks = len(tokens) // bs
         if n_chunks == 0:
             return []
-        if self._native is not None:
+        if self._native is not None and self.config.hash_algo == "fnv-64a":
             return self._native.tokens_to_chunk_hashes(
                 list(tokens[: n_chunks * bs]), parent_hash, bs
             )
+        link = hashing.CHAIN_ALGOS[self.config.hash_algo][0]
         hashes = []
         h = parent_hash
         for i in range(n_chunks):
-            h = hashing.chunk_hash(h, tokens[i * bs : (i + 1) * bs])
+            h = link(h, tokens[i * bs : (i + 1) * bs])
             hashes.append(h)
         return hashes
 

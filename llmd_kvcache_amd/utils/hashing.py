@@ -77,3 +77,31 @@ def init_hash(seed: str = "") -> int:
     """Root parent hash: FNV-64a of the seed string bytes
     (token_processor.go:81-90). Empty seed -> FNV offset basis."""
     return fnv1a_64(seed.encode("utf-8"))
+
+
+# -- pluggable hash algorithms (SURVEY section 5: vLLM changed SHA-256 ->
+# FNV-64a once already; keep the chain function swappable) --------------
+
+def sha256_cbor_64(parent_hash: int, tokens) -> int:
+    """Alternative chain link: top 8 bytes (big-endian) of SHA-256 over
+    the same canonical-CBOR payload - the shape of vLLM's historical
+    sha256 content addressing. NOTE: gated to host paths until a vLLM
+    golden fixture pins the exact byte convention (ROADMAP #5); the
+    HIP/C++ fast paths implement fnv-64a only."""
+    import hashlib
+
+    digest = hashlib.sha256(cbor_chunk_payload(parent_hash, tokens)).digest()
+    return int.from_bytes(digest[:8], "big")
+
+
+def sha256_init_hash(seed: str = "") -> int:
+    import hashlib
+
+    return int.from_bytes(
+        hashlib.sha256(seed.encode("utf-8")).digest()[:8], "big")
+
+
+CHAIN_ALGOS = {
+    "fnv-64a": (chunk_hash, init_hash),
+    "sha256-cbor-64": (sha256_cbor_64, sha256_init_hash),
+}

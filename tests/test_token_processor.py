@@ -5,6 +5,8 @@ chunking with dropped partial tails, chain continuation from a parent key,
 seed handling.
 """
 
+import pytest
+
 from llmd_kvcache_amd.kvblock.keys import Key
 from llmd_kvcache_amd.kvblock.token_processor import (
     ChunkedTokenDatabase,
@@ -138,3 +140,30 @@ class TestBranchlessParity:
         a = mod.tokens_to_chunk_hashes(toks, hashing.init_hash(""), 16)
         b = mod.tokens_to_chunk_hashes_fast(toks, hashing.init_hash(""), 16)
         assert list(a) == list(b)
+
+
+class TestPluggableHash:
+    def test_sha256_algo_chains_differently_but_deterministically(self):
+        from llmd_kvcache_amd.utils import hashing as hh
+
+        fnv = ChunkedTokenDatabase(TokenProcessorConfig(block_size=4))
+        sha = ChunkedTokenDatabase(TokenProcessorConfig(
+            block_size=4, hash_algo="sha256-cbor-64"))
+        toks = list(range(12))
+        a = sha.tokens_to_kv_block_keys(None, toks, "m")
+        b = sha.tokens_to_kv_block_keys(None, toks, "m")
+        assert a == b and len(a) == 3
+        assert a != fnv.tokens_to_kv_block_keys(None, toks, "m")
+        # chain property: prefix determines prefix
+        c = sha.tokens_to_kv_block_keys(None, toks[:8], "m")
+        assert a[:2] == c
+        # link function is SHA-256 over the SAME canonical CBOR payload
+        import hashlib
+        payload = hh.cbor_chunk_payload(sha.config.init_hash(), toks[:4])
+        expect = int.from_bytes(hashlib.sha256(payload).digest()[:8], "big")
+        assert a[0].chunk_hash == expect
+
+    def test_unknown_algo_raises(self):
+        cfg = TokenProcessorConfig(hash_algo="md5")
+        with pytest.raises(KeyError):
+            cfg.init_hash()
