@@ -52,3 +52,34 @@ class TestBenchContract:
         d = run_bench(["--prefix-frac", "1.0", "--pods", "8"])
         assert d["config"]["prefix_frac"] == 1.0
         assert d["config"]["num_pods"] == 8
+
+
+@pytest.mark.timeout(600)
+class TestBenchTorchrun:
+    """The driver launches N>1 exactly like this (one rank per GPU over
+    RCCL; gloo here): shape must keep working."""
+
+    def run_torchrun(self, extra, port):
+        proc = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run",
+             "--nnodes=1", "--nproc-per-node", "2",
+             "--master-addr", "127.0.0.1", "--master-port", str(port),
+             os.path.join(ROOT, "bench.py")] + TINY +
+            ["--batch", "16", "--blocks", "1024"] + extra,
+            capture_output=True, text=True, timeout=300, cwd=ROOT)
+        assert proc.returncode == 0, proc.stderr[-800:]
+        json_lines = [l for l in proc.stdout.splitlines()
+                      if l.startswith("{")]
+        assert len(json_lines) == 1, "exactly one rank prints the result"
+        return json.loads(json_lines[0])
+
+    def test_replicated_world2(self):
+        d = self.run_torchrun([], 29541)
+        assert d["config"]["parallelism"] == "replicated2"
+        assert d["scaling"] == "weak"
+        assert d["config"]["global_batch"] == 32  # whole-job aggregate
+
+    def test_sharded_world2(self):
+        d = self.run_torchrun(["--sharded"], 29542)
+        assert d["config"]["parallelism"] == "shard2"
+        assert d["scaling"] == "strong"
