@@ -180,3 +180,32 @@ class TestLruModel:
                 assert lru.remove(k) == (model.pop(k, None) is not None)
             assert len(lru) == len(model)
             assert list(lru.keys()) == list(model.keys())
+
+
+@given(st.integers(0, 2**31), st.integers(10, 120))
+@settings(max_examples=25, deadline=None)
+def test_compact_is_lookup_invariant(seed, n_ops):
+    """Property: for ANY op sequence, TableIndex.compact() (including a
+    grow-then-shrink cycle) preserves every lookup result."""
+    import random as _random
+
+    from llmd_kvcache_amd.kvblock.gpu_index import (NativeIndex,
+                                                    TableIndexConfig)
+    from llmd_kvcache_amd.kvblock.keys import Key
+    from llmd_kvcache_amd.ops import cpu_ext
+    from tests.test_native_index import random_workload, run_op
+
+    if cpu_ext.maybe_load() is None:
+        return  # extension not built in this env
+    rng = _random.Random(seed)
+    nat = NativeIndex(TableIndexConfig(capacity=1 << 10, pods_per_key=4))
+    for op in random_workload(rng, n_ops=n_ops, key_space=64, n_pods=5):
+        run_op(nat, op)
+    keys = [Key("m", 1000 + i) for i in range(64)]
+    before = {k: sorted(map(tuple, v))
+              for k, v in nat.lookup(keys, set()).items()}
+    nat.compact(new_capacity=1 << 11)
+    nat.compact(new_capacity=1 << 10)
+    after = {k: sorted(map(tuple, v))
+             for k, v in nat.lookup(keys, set()).items()}
+    assert after == before
