@@ -11,10 +11,11 @@ Write path: adds go to BOTH tiers (the CPU tier write is cheap relative
 to event decode; this keeps the cold tier a superset without needing
 eviction callbacks out of the GPU kernel).
 
-Read path: fused score on the hot tier first; if the result shows early
-chain termination (a prompt whose keys fell out of the hot tier), the
-generic lookup merges cold-tier entries for the missing keys.  The
-common case (hot working set) costs exactly one kernel.
+Read path: one hot-tier lookup kernel yields per-key masks and found
+flags; fully hot-resident batches (the common case) score straight from
+the masks on-device. Any hot MISS routes through the generic merged
+lookup - which serves the cold entries AND promotes them back into HBM -
+so cold-tier data is never invisible to the fast path.
 """
 
 from __future__ import annotations
@@ -123,8 +124,60 @@ class TieredIndex(Index):
         for pod, model, events in batches:
             digest_events(self.cold, token_processor, pod, model, events)
 
-    def fused_scores(self, *args, **kwargs):
-        return self.hot.fused_scores(*args, **kwargs)
+    def fused_scores(self, hashes, counts_or_offsets, model_name,
+                     pod_identifier_set, weights=None, max_k=None):
+        """Read fast path with cold-tier correctness: one hot-tier lookup
+        kernel produces per-key masks AND found flags; when every key is
+        hot-resident (the common case) the scores come straight from the
+        masks on-device. Any hot miss routes the batch through the
+        generic merged lookup (which also promotes the cold hits back
+        into HBM) and the Python scorer."""
+        import torch
+
+        from ..scorer import LongestPrefixScorer
+        from .gpu_index import _to_u64
+
+        hot = self.hot
+        model_id = hot.registry.model_id(model_name)
+        num_pods = hot._num_pods_padded()
+        filt = hot._filter_tensor(pod_identifier_set, num_pods)
+        if weights is None:
+            weights = hot.tier_weights()
+        n_tiers = max(1, len(hot.registry.id_to_tier))
+        found, masks = hot.table.lookup(hashes, model_id, filt, num_pods,
+                                        n_tiers=n_tiers)
+        if hot.table.is_cuda:
+            offsets = counts_or_offsets
+        else:  # CPU fused convention passes per-prompt counts
+            counts = counts_or_offsets.to(torch.int32)
+            offsets = torch.zeros(counts.numel() + 1, dtype=torch.int32)
+            torch.cumsum(counts, 0, out=offsets[1:].view(counts.numel()))
+        if int((found == 0).sum()) == 0:
+            ops = hot.table.ops
+            fn = (ops.gpu_score_from_masks if hot.table.is_cuda
+                  else ops.cpu_score_from_masks)
+            return fn(masks.contiguous(), offsets.to(found.device)
+                      if hot.table.is_cuda else offsets.cpu(),
+                      weights, num_pods)
+        # hot miss: merged lookup (promotes) + reference scorer
+        hashes_l = hashes.cpu().tolist()
+        offs_l = offsets.cpu().tolist()
+        scorer = LongestPrefixScorer(
+            medium_weights={name: float(weights[i])
+                            for i, name in enumerate(hot.registry.id_to_tier)})
+        out = torch.zeros((len(offs_l) - 1, num_pods), dtype=torch.float32,
+                          device=hot.device)
+        for b in range(len(offs_l) - 1):
+            keys = [Key(model_name, _to_u64(h))
+                    for h in hashes_l[offs_l[b]:offs_l[b + 1]]]
+            if not keys:
+                continue
+            merged = self.lookup(keys, pod_identifier_set)
+            for pod, score in scorer.score(keys, merged).items():
+                pid = hot.registry.pod_to_id.get(pod)
+                if pid is not None and pid < num_pods:
+                    out[b, pid] = score
+        return out
 
     def tier_weights(self, *args, **kwargs):
         return self.hot.tier_weights(*args, **kwargs)

@@ -155,3 +155,56 @@ class TestPromotion:
         ek, rk = Key(MODEL, 900), Key(MODEL, 901)
         t.add([ek], [rk], [PodEntry("pod-q", "gpu")])
         assert t.get_request_key(ek) == rk
+
+
+class TestTieredFusedColdCorrectness:
+    def _hashes(self, keys, device="cpu"):
+        from llmd_kvcache_amd.kvblock.gpu_index import _to_i64
+
+        return torch.tensor([_to_i64(k.chunk_hash) for k in keys],
+                            dtype=torch.int64, device=device)
+
+    def test_fused_scores_sees_cold_only_keys(self):
+        """The read fast path must NOT lose cold-tier hits: keys present
+        only in the capacity tier score correctly and get promoted."""
+        t = make_tiered()
+        keys = [Key(MODEL, 900 + i) for i in range(4)]
+        entries = [PodEntry("pod-cold", "gpu")]
+        t.cold.add(keys, keys, entries)  # cold-only (hot never saw them)
+        counts = torch.tensor([len(keys)], dtype=torch.int32)
+        scores = t.fused_scores(self._hashes(keys), counts, MODEL, set())
+        m = t.scores_to_map(scores)[0]
+        assert m == {"pod-cold": pytest.approx(4.0)}
+        # and the lookup promoted them: next call is pure hot fast path
+        assert len(t.hot.lookup(keys, set())) == 4
+        scores2 = t.fused_scores(self._hashes(keys), counts, MODEL, set())
+        assert t.scores_to_map(scores2)[0] == {"pod-cold": pytest.approx(4.0)}
+
+    def test_fused_scores_mixed_hot_cold_prefix(self):
+        """Prefix split across tiers: hot has key0, only cold has key1 -
+        the merged walk must still credit both."""
+        t = make_tiered()
+        k0, k1 = Key(MODEL, 70), Key(MODEL, 71)
+        t.add([k0], [k0], [PodEntry("pod-m", "gpu")])      # both tiers
+        t.cold.add([k1], [k1], [PodEntry("pod-m", "gpu")])  # cold only
+        counts = torch.tensor([2], dtype=torch.int32)
+        scores = t.fused_scores(self._hashes([k0, k1]), counts, MODEL, set())
+        assert t.scores_to_map(scores)[0] == {"pod-m": pytest.approx(2.0)}
+
+    def test_indexer_fast_path_through_tiered(self):
+        """End to end: Indexer._score_keys fused path on a TieredIndex
+        backend reaches cold-tier data."""
+        from llmd_kvcache_amd.indexer import Config, Indexer
+        from llmd_kvcache_amd.kvblock.token_processor import (
+            ChunkedTokenDatabase, TokenProcessorConfig)
+
+        t = make_tiered()
+        cfg = Config()
+        cfg.token_processor = TokenProcessorConfig(block_size=4)
+        tp = ChunkedTokenDatabase(cfg.token_processor)
+        tokens = list(range(8))
+        keys = tp.tokens_to_kv_block_keys(None, tokens, MODEL)
+        t.cold.add(keys, keys, [PodEntry("pod-deep", "cpu")])
+        idx = Indexer(cfg, kv_block_index=t)
+        scores = idx.score_tokens(tokens, MODEL, [])
+        assert scores == {"pod-deep": pytest.approx(2 * 0.8)}
