@@ -340,6 +340,33 @@ class TestTieredGpu:
         assert t.hot.lookup([solo], set())  # promoted
         assert t.hot.get_request_key(solo) is None
 
+    def test_fused_scores_cold_only_on_gpu_hot(self):
+        """GPU hot tier: the fused read path must serve keys resident
+        only in the CPU capacity tier (and promote them)."""
+        from llmd_kvcache_amd.kvblock.gpu_index import (NativeIndex,
+                                                        TableIndexConfig,
+                                                        _to_i64)
+        from llmd_kvcache_amd.kvblock.tiered import TieredIndex
+
+        hot = GpuIndex(GpuIndexConfig(capacity=1 << 12, pods_per_key=4))
+        cold = NativeIndex(TableIndexConfig(capacity=1 << 14,
+                                            pods_per_key=10),
+                           registry=hot.registry)
+        t = TieredIndex(hot=hot, cold=cold)
+        keys = [Key(MODEL, 40_000 + i) for i in range(3)]
+        cold.add(keys, keys, [PodEntry("pod-cc", "cpu")])
+        hashes = torch.tensor([_to_i64(k.chunk_hash) for k in keys],
+                              dtype=torch.int64, device="cuda")
+        offs = torch.tensor([0, 3], dtype=torch.int32, device="cuda")
+        scores = t.fused_scores(hashes, offs, MODEL, set())
+        m = t.scores_to_map(scores)[0]
+        assert m == {"pod-cc": pytest.approx(3 * 0.8)}
+        torch.cuda.synchronize()
+        assert len(t.hot.lookup(keys, set())) == 3  # promoted to HBM
+        scores2 = t.fused_scores(hashes, offs, MODEL, set())  # pure hot now
+        assert t.scores_to_map(scores2)[0] == {"pod-cc":
+                                               pytest.approx(3 * 0.8)}
+
 
 class TestManyPods:
     def test_fused_score_256_pods(self):
