@@ -34,6 +34,9 @@ class AsgiIndexerApp:
             if method == "POST" and path == "/score_completions":
                 body = await self._read_json(receive)
                 await self._score_completions(send, body)
+            elif method == "POST" and path == "/score_batch":
+                body = await self._read_json(receive)
+                await self._score_batch(send, body)
             elif method == "POST" and path == "/score_chat_completions":
                 body = await self._read_json(receive)
                 await self._score_chat(send, body)
@@ -57,6 +60,19 @@ class AsgiIndexerApp:
             None, prompt, body.get("model", ""), []
         )
         await self._json(send, 200, scores or {})
+
+    async def _score_batch(self, send, body: Dict[str, Any]):
+        prompts = body.get("prompts")
+        if not isinstance(prompts, list) or not prompts:
+            raise _BadRequest("field 'prompts' (non-empty list) required")
+        model = body.get("model", "")
+        token_lists = [
+            self.indexer.tokenizers_pool.tokenize(None, p, model)
+            for p in prompts
+        ]
+        scores = self.indexer.score_tokens_batch(
+            token_lists, model, body.get("pods", []))
+        await self._json(send, 200, {"scores": scores})
 
     async def _score_chat(self, send, body: Dict[str, Any]):
         model = body.get("model", "")

@@ -2,6 +2,8 @@
 (examples/kv_events/online/main.go:260-389):
 
  - POST /score_completions        {"prompt", "model"} -> {pod: score}
+ - POST /score_batch              {"prompts": [...], "model"} ->
+                                  {"scores": [{pod: score}, ...]}  (one fused kernel)
  - POST /score_chat_completions   chat-completions request -> {"podScores",
    "templated_messages"}  (template fetched for the model when absent)
  - GET  /metrics                  Prometheus exposition
@@ -75,6 +77,8 @@ def make_handler(indexer: Indexer):
             try:
                 if self.path == "/score_completions":
                     self._score_completions()
+                elif self.path == "/score_batch":
+                    self._score_batch()
                 elif self.path == "/score_chat_completions":
                     self._score_chat_completions()
                 else:
@@ -94,6 +98,24 @@ def make_handler(indexer: Indexer):
                 return
             pods = indexer.get_pod_scores(None, prompt, model, [])
             self._send_json(200, pods or {})
+
+        def _score_batch(self):
+            # MI355X-native addition: N prompts -> ONE fused scoring
+            # kernel (the batched hot interface the bench measures)
+            req = self._read_json()
+            prompts = req.get("prompts")
+            model = req.get("model", "")
+            if not isinstance(prompts, list) or not prompts:
+                self._send_error(400, "field 'prompts' (non-empty list) "
+                                      "required")
+                return
+            token_lists = [
+                indexer.tokenizers_pool.tokenize(None, p, model)
+                for p in prompts
+            ]
+            scores = indexer.score_tokens_batch(
+                token_lists, model, req.get("pods", []))
+            self._send_json(200, {"scores": scores})
 
         def _score_chat_completions(self):
             req = self._read_json()

@@ -321,3 +321,61 @@ class TestLiveUvicorn:
         finally:
             server.should_exit = True
             t.join(timeout=5)
+
+
+class TestScoreBatch:
+    def test_batch_matches_per_prompt(self):
+        idx, index = make_indexer()
+        prompts = [f"batch prompt number {i} " * 8 for i in range(4)]
+        for i, p in enumerate(prompts[:2]):  # seed index for first two
+            toks = FixedTokenizer().encode(p, "m")[0]
+            keys = idx.tokens_processor.tokens_to_kv_block_keys(None, toks, "m")
+            index.add(keys, keys, [PodEntry(f"pod-{i}", "gpu")])
+
+        token_lists = [FixedTokenizer().encode(p, "m")[0] for p in prompts]
+        batch = idx.score_tokens_batch(token_lists, "m", [])
+        single = [idx.score_tokens(t, "m", []) for t in token_lists]
+        assert len(batch) == 4
+        for b, s in zip(batch, single):
+            assert set(b) == set(s)
+            for pod in s:
+                assert b[pod] == pytest.approx(s[pod])
+        assert batch[0] and batch[1]  # the seeded prompts actually hit
+
+    def test_http_and_asgi_score_batch(self):
+        idx, index = make_indexer()
+        prompt = "served batch prompt! " * 10
+        toks = FixedTokenizer().encode(prompt, "m")[0]
+        keys = idx.tokens_processor.tokens_to_kv_block_keys(None, toks, "m")
+        index.add(keys, keys, [PodEntry("pod-b", "gpu")])
+
+        # ASGI
+        from llmd_kvcache_amd.service.asgi import build_app
+        app = build_app(idx)
+        status, out = TestAsgiApp.call(
+            app, "POST", "/score_batch",
+            {"prompts": [prompt, "never seen text"], "model": "m"})
+        assert status == 200
+        assert out["scores"][0].get("pod-b", 0) > 0
+        assert out["scores"][1] == {}
+        status, _ = TestAsgiApp.call(app, "POST", "/score_batch",
+                                     {"prompts": [], "model": "m"})
+        assert status == 400
+
+        # stdlib HTTP
+        import http.client
+        from llmd_kvcache_amd.service.http_server import HttpService
+        http_svc = HttpService(idx, host="127.0.0.1", port=0)
+        http_svc.start()
+        try:
+            conn = http.client.HTTPConnection("127.0.0.1", http_svc.port,
+                                              timeout=5)
+            conn.request("POST", "/score_batch",
+                         json.dumps({"prompts": [prompt], "model": "m"}),
+                         {"Content-Type": "application/json"})
+            r = conn.getresponse()
+            out = json.loads(r.read())
+            assert r.status == 200 and out["scores"][0].get("pod-b", 0) > 0
+            conn.close()
+        finally:
+            http_svc.stop()
