@@ -110,6 +110,18 @@ def test_full_service_stack_on_gpu(tokenizer_dir):
             f"http://127.0.0.1:{http.port}/metrics", timeout=5
         ) as resp:
             assert resp.status == 200
+
+        # batched scoring surface (one fused kernel for the batch)
+        body = json.dumps({"prompts": [prompt, "w0 w1 w2"],
+                           "model": MODEL}).encode()
+        req = urllib.request.Request(
+            f"http://127.0.0.1:{http.port}/score_batch",
+            data=body, headers={"Content-Type": "application/json"},
+        )
+        with urllib.request.urlopen(req, timeout=5) as resp:
+            out = json.loads(resp.read())
+        assert out["scores"][0] == {"vllm-gpu-pod": 4.0}
+        assert out["scores"][1] == {}
     finally:
         pub.close()
         http.stop()
