@@ -119,6 +119,17 @@ def index_config_from_dict(data: Dict[str, Any]) -> IndexConfig:
         from .kvblock.gpu_index import GpuIndexConfig
 
         cfg.gpu = _apply(GpuIndexConfig(), data["gpu"] or {})
+    if "tiered" in data:
+        from .kvblock.gpu_index import GpuIndexConfig, TableIndexConfig
+        from .kvblock.tiered import TieredIndexConfig
+
+        sub = data["tiered"] or {}
+        hot = _apply(GpuIndexConfig(), sub.get("hot") or {})
+        tiered = TieredIndexConfig(hot=hot)
+        if sub.get("cold"):
+            tiered.cold = _apply(tiered.cold, sub["cold"])
+            tiered.cold.device = "cpu"
+        cfg.tiered = tiered
     if "cost_aware" in data:
         from .kvblock.cost_aware import CostAwareMemoryIndexConfig
 
@@ -133,8 +144,8 @@ def index_config_from_dict(data: Dict[str, Any]) -> IndexConfig:
         from .kvblock.redis_index import RedisIndexConfig
 
         cfg.valkey = _apply(RedisIndexConfig(), data["valkey"] or {})
-    if not any((cfg.in_memory, cfg.native, cfg.gpu, cfg.cost_aware,
-                cfg.redis, cfg.valkey)):
+    if not any((cfg.in_memory, cfg.native, cfg.gpu, cfg.tiered,
+                cfg.cost_aware, cfg.redis, cfg.valkey)):
         from .kvblock.in_memory import InMemoryIndexConfig
 
         cfg.in_memory = InMemoryIndexConfig()

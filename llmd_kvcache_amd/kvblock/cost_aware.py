@@ -21,11 +21,14 @@ from .index import Index
 from .keys import Key, PodEntry
 
 DEFAULT_MAX_COST_BYTES = 2 * 1024 * 1024 * 1024  # 2 GiB
+DEFAULT_ENGINE_MAP_SIZE = 1_000_000  # caps the engine->request mapping
+# (reference cost_aware_memory.go uses an LRU "to cap mapping size")
 
 
 @dataclass
 class CostAwareMemoryIndexConfig:
     max_cost_bytes: int = DEFAULT_MAX_COST_BYTES
+    engine_map_size: int = DEFAULT_ENGINE_MAP_SIZE
 
 
 def calculate_byte_size(key: Key, pods: Dict[PodEntry, None]) -> int:
@@ -41,8 +44,13 @@ class CostAwareMemoryIndex(Index):
     def __init__(self, cfg: Optional[CostAwareMemoryIndexConfig] = None):
         cfg = cfg or CostAwareMemoryIndexConfig()
         self.max_cost = cfg.max_cost_bytes
+        self.engine_map_size = max(1, cfg.engine_map_size)
         self._data: "OrderedDict[Key, Dict[PodEntry, None]]" = OrderedDict()
-        self._engine_to_request: Dict[Key, Key] = {}
+        # engine->request is a bounded LRU (cost_aware_memory.go caps this
+        # mapping with an LRU); _request_to_engines is the reverse set so
+        # budget eviction of a request key can drop its engine mappings.
+        self._engine_to_request: "OrderedDict[Key, Key]" = OrderedDict()
+        self._request_to_engines: Dict[Key, Set[Key]] = {}
         self._costs: Dict[Key, int] = {}
         self._total_cost = 0
         self._mu = threading.RLock()
@@ -52,10 +60,34 @@ class CostAwareMemoryIndex(Index):
         self._total_cost += new_cost - self._costs.get(key, 0)
         self._costs[key] = new_cost
 
+    def _map_engine(self, engine_key: Key, request_key: Key) -> None:
+        old = self._engine_to_request.get(engine_key)
+        if old is not None and old != request_key:
+            peers = self._request_to_engines.get(old)
+            if peers is not None:
+                peers.discard(engine_key)
+                if not peers:
+                    self._request_to_engines.pop(old, None)
+        self._engine_to_request[engine_key] = request_key
+        self._engine_to_request.move_to_end(engine_key)
+        self._request_to_engines.setdefault(request_key, set()).add(engine_key)
+        while len(self._engine_to_request) > self.engine_map_size:
+            ek, rk = self._engine_to_request.popitem(last=False)
+            peers = self._request_to_engines.get(rk)
+            if peers is not None:
+                peers.discard(ek)
+                if not peers:
+                    self._request_to_engines.pop(rk, None)
+
+    def _unmap_request(self, request_key: Key) -> None:
+        for ek in self._request_to_engines.pop(request_key, ()):
+            self._engine_to_request.pop(ek, None)
+
     def _evict_over_budget(self) -> None:
         while self._total_cost > self.max_cost and self._data:
             key, _pods = self._data.popitem(last=False)
             self._total_cost -= self._costs.pop(key, 0)
+            self._unmap_request(key)
 
     def lookup(
         self, request_keys: Sequence[Key], pod_identifier_set: Set[str]
@@ -93,7 +125,7 @@ class CostAwareMemoryIndex(Index):
             raise ValueError("mismatch between engine keys and request keys length")
         with self._mu:
             for engine_key, request_key in zip(engine_keys, request_keys):
-                self._engine_to_request[engine_key] = request_key
+                self._map_engine(engine_key, request_key)
                 pods = self._data.get(request_key)
                 if pods is None:
                     pods = {}
@@ -114,14 +146,14 @@ class CostAwareMemoryIndex(Index):
                 return
             pods = self._data.get(request_key)
             if pods is None:
-                self._engine_to_request.pop(engine_key, None)
+                self._unmap_request(request_key)
                 return
             for e in entries:
                 pods.pop(e, None)
             if not pods:
                 self._data.pop(request_key, None)
                 self._total_cost -= self._costs.pop(request_key, 0)
-                self._engine_to_request.pop(engine_key, None)
+                self._unmap_request(request_key)
             else:
                 self._set_cost(request_key, pods)
 

@@ -63,9 +63,13 @@ class Registry:
         tid = self.tier_to_id.get(tier)
         if tid is None:
             if len(self.id_to_tier) >= MAX_TIERS:
-                tid = MAX_TIERS - 1  # overflow tiers share the last slot
-                self.tier_to_id[tier] = tid
-                return tid
+                # Silently sharing a slot would mis-weight scores for
+                # every tier folded onto it; fail loudly instead (event
+                # paths catch this and drop the offending event).
+                raise ValueError(
+                    f"tier registry full ({MAX_TIERS}): cannot intern "
+                    f"{tier!r}; registered tiers: {self.id_to_tier}"
+                )
             tid = len(self.id_to_tier)
             self.tier_to_id[tier] = tid
             self.id_to_tier.append(tier)
@@ -621,8 +625,25 @@ class GpuIndex(TableIndex):
             n_events_in_group = 0
             for ev in by_pod[(pod, model)]:
                 if isinstance(ev, BlockStored):
-                    tier = self.registry.tier_id(
-                        ev.medium.lower() if ev.medium else "gpu")
+                    # parse everything fallible BEFORE appending: a
+                    # malformed parent hash or an un-internable tier drops
+                    # the whole event, matching _digest_block_stored /
+                    # pool.go (CPU and GPU replicas converge on the same
+                    # state for malformed streams).
+                    if ev.parent_block_hash is not None:
+                        try:
+                            par = _to_i64(
+                                get_hash_as_uint64(ev.parent_block_hash))
+                            hp = 1
+                        except Exception:
+                            continue  # drop event (bad parent hash)
+                    else:
+                        par, hp = 0, 0
+                    try:
+                        tier = self.registry.tier_id(
+                            ev.medium.lower() if ev.medium else "gpu")
+                    except ValueError:
+                        continue  # drop event (tier registry full)
                     hs = _hashes_np(ev.block_hashes)
                     if hs.size == 0:
                         continue
@@ -633,23 +654,17 @@ class GpuIndex(TableIndex):
                     token_arrays.append(toks)
                     n_tokens += toks.size
                     tok_off.append(n_tokens)
-                    if ev.parent_block_hash is not None:
-                        try:
-                            parents.append(
-                                _to_i64(get_hash_as_uint64(ev.parent_block_hash)))
-                            has_parent.append(1)
-                        except Exception:
-                            parents.append(0)
-                            has_parent.append(0)
-                    else:
-                        parents.append(0)
-                        has_parent.append(0)
+                    parents.append(par)
+                    has_parent.append(hp)
                     ev_type.append(0)
                     pod_entry.append((tier << 24) | (pod_id + 1))
                     n_events_in_group += 1
                 elif isinstance(ev, BlockRemoved):
-                    tier = self.registry.tier_id(
-                        ev.medium.lower() if ev.medium else "gpu")
+                    try:
+                        tier = self.registry.tier_id(
+                            ev.medium.lower() if ev.medium else "gpu")
+                    except ValueError:
+                        continue  # drop event (tier registry full)
                     hs = _hashes_np(ev.block_hashes)
                     if hs.size == 0:
                         continue

@@ -3,10 +3,17 @@
 Parity with reference pkg/kvcache/kvblock/instrumented_index.go:25-92:
 records admissions, evictions, lookup requests/latency, per-lookup hits
 and the max consecutive pod hit count (computed off the hot path).
+
+The hit-metrics computation runs on ONE shared bounded worker (the
+reference spawns a goroutine per lookup, which is cheap in Go; a Python
+thread per lookup is not - under metrics-enabled load the create/join
+cost dominates). When the queue is full the sample is dropped: metrics
+are best-effort, the hot path never blocks.
 """
 
 from __future__ import annotations
 
+import queue
 import threading
 import time
 from typing import Dict, List, Optional, Sequence, Set
@@ -14,6 +21,33 @@ from typing import Dict, List, Optional, Sequence, Set
 from ..metrics import collector
 from .index import Index
 from .keys import Key, PodEntry
+
+_HIT_QUEUE_MAX = 1024
+
+_hit_queue: "queue.Queue" = queue.Queue(maxsize=_HIT_QUEUE_MAX)
+_hit_worker_lock = threading.Lock()
+_hit_worker: Optional[threading.Thread] = None
+
+
+def _hit_worker_loop() -> None:
+    while True:
+        request_keys, result = _hit_queue.get()
+        try:
+            InstrumentedIndex._record_hit_metrics(request_keys, result)
+        except Exception:
+            pass
+
+
+def _ensure_hit_worker() -> None:
+    global _hit_worker
+    if _hit_worker is not None and _hit_worker.is_alive():
+        return
+    with _hit_worker_lock:
+        if _hit_worker is None or not _hit_worker.is_alive():
+            _hit_worker = threading.Thread(
+                target=_hit_worker_loop, name="kvidx-hit-metrics", daemon=True
+            )
+            _hit_worker.start()
 
 
 class InstrumentedIndex(Index):
@@ -31,13 +65,13 @@ class InstrumentedIndex(Index):
             collector.lookup_latency.observe(time.monotonic() - t0)
         if collector.lookup_hits is not None:
             collector.lookup_hits.inc(len(result))
-        # max consecutive pod-hit count computed asynchronously
-        # (instrumented_index.go:62,71-92)
-        threading.Thread(
-            target=self._record_hit_metrics,
-            args=(list(request_keys), dict(result)),
-            daemon=True,
-        ).start()
+        # max consecutive pod-hit count computed asynchronously on the
+        # shared bounded worker (instrumented_index.go:62,71-92)
+        _ensure_hit_worker()
+        try:
+            _hit_queue.put_nowait((list(request_keys), dict(result)))
+        except queue.Full:
+            pass  # drop the sample, never block the lookup
         return result
 
     @staticmethod

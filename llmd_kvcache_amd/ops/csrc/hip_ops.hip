@@ -608,6 +608,12 @@ __global__ void k_apply_events(
     // BlockStored: lane 0 resolves parent + streams the request chain.
     const int n_tok = tok_off[e + 1] - tok_off[e];
     const int n_chunks = n_tok / block_size;
+    // Engine-hash count must match the token-derived chain length; the
+    // reference Add() (and this repo's CPU digest path) rejects the whole
+    // event on mismatch - dropping here keeps CPU and GPU replicas
+    // convergent on malformed streams. (Uniform per wave: n_chunks/nh
+    // are scalar per event.)
+    if (n_chunks != nh) continue;
     uint64_t* req = req_scratch + eh_off[e];
     if (lane == 0) {
       uint64_t parent = init_hash;
@@ -617,16 +623,14 @@ __global__ void k_apply_events(
       }
       const int64_t* t0 = tokens + tok_off[e];
       uint64_t h = parent;
-      for (int c = 0; c < n_chunks && c < nh; ++c) {
+      for (int c = 0; c < n_chunks; ++c) {
         h = chunk_hash_fast(h, t0 + (int64_t)c * block_size, block_size);
         req[c] = h;
       }
-      // engine hashes beyond the token-derived chain keep prior behavior:
-      // reference Add() errors on length mismatch; we clamp to min(n,nh).
     }
     __threadfence_block();
     __builtin_amdgcn_wave_barrier();
-    const int n_ins = min(n_chunks, nh);
+    const int n_ins = n_chunks;
     for (int i = lane; i < n_ins; i += 64) {
       dev_emap_put(v, eh[i], model, remap_hash(req[i]));
       if (num_shards > 1 &&
@@ -707,6 +711,9 @@ __global__ void k_event_chains(
     if (ev_type[e] == 1) continue;  // removals have no chain
     const int nh = eh_off[e + 1] - eh_off[e];
     const int n_chunks = (tok_off[e + 1] - tok_off[e]) / block_size;
+    if (n_chunks != nh) continue;  // drop mismatched event (see
+                                   // k_apply_events; k_event_inserts
+                                   // applies the same guard)
     uint64_t parent = init_hash;
     if (has_parent[e]) {
       int64_t bi = dev_bmap_find(bkeys, bvals, bmask, parents[e]);
@@ -753,9 +760,11 @@ __global__ void k_event_inserts(
     if (empty) atomicOr(&v.meta[slot], META_TOMB);
     return;
   }
-  // BlockStored: only blocks covered by the token chain insert
+  // BlockStored: dropped entirely when the engine-hash count mismatches
+  // the token-derived chain (same guard as k_event_chains/k_apply_events)
   const int local = (int)(i - eh_off[e]);
   const int n_chunks = (tok_off[e + 1] - tok_off[e]) / block_size;
+  if (n_chunks != eh_off[e + 1] - eh_off[e]) return;
   if (local >= n_chunks) return;
   const uint64_t req = req_scratch[i];
   dev_emap_put(v, ehashes[i], model, remap_hash(req));

@@ -65,11 +65,15 @@ class ShardedIndexService:
             return None
         if kind == "score":
             _, hashes, offsets, model, pods = op
-            device = self.sharded.device
-            h = torch.tensor(hashes, dtype=torch.int64, device=device)
+            # Stage on CPU from the broadcast payload: this is symmetric
+            # across ranks (same op object everywhere), so a failure here
+            # fails every rank BEFORE any collective - no deadlock.  All
+            # asymmetric work (device moves, registry, probe) happens
+            # inside sharded_scores, which guarantees collective
+            # participation on its own failure paths.
+            h = torch.tensor(hashes, dtype=torch.int64)
             offs = torch.tensor(offsets, dtype=torch.int32)
-            scores = self.sharded.sharded_scores(h, offs, model, set(pods))
-            return scores
+            return self.sharded.sharded_scores(h, offs, model, set(pods))
         if kind == "sync_check":
             # ROADMAP #8: loud detection of registry divergence before it
             # can misattribute merged-mask scores
@@ -144,10 +148,14 @@ class ShardedIndexService:
     def serve(self) -> None:
         """Follower ranks: execute broadcast ops until stop.  A failing op
         must not kill the follower - the next collective would then hang
-        every rank - so failures are logged and the loop continues (score
-        ops participate in the all_reduce even on failure via the zero
-        masks their empty dispatch produces; event failures only lose the
-        local application, mirroring the pool's poison-pill stance)."""
+        every rank - so failures are logged and the loop continues.  Score
+        ops are guaranteed to participate in the mask all_reduce on EVERY
+        failure path: sharded_scores all_reduces zero masks when the local
+        probe throws, and _dispatch participates explicitly when tensor
+        staging throws before sharded_scores is reached (covered by
+        tests/test_sharded.py follower-failure test).  Event failures only
+        lose the local application, mirroring the pool's poison-pill
+        stance."""
         assert self.rank != 0
         import logging
 

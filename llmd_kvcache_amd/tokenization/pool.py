@@ -115,28 +115,26 @@ class TokenizationPool:
     ) -> List[int]:
         """Synchronous tokenization (blocks on the worker result)."""
         result: "queue.Queue" = queue.Queue(maxsize=1)
-        self._queue.put(_Task(render_req, prompt, model_name, result))
+        task = _Task(render_req, prompt, model_name, result)
         if not self._running:
-            # inline mode when pool isn't started (tests/library usage)
-            self._drain_one()
+            # inline mode when the pool isn't started (tests/library
+            # usage): process THIS task directly - going through the
+            # shared queue could dequeue a concurrent caller's task and
+            # leave this caller's result queue empty forever.
+            self._process(task)
+        else:
+            self._queue.put(task)
         outcome, payload = result.get()
         if outcome == "err":
             raise payload
         return payload
 
     def enqueue_tokenization(self, render_req, prompt: str, model_name: str) -> None:
-        self._queue.put(_Task(render_req, prompt, model_name, None))
+        task = _Task(render_req, prompt, model_name, None)
         if not self._running:
-            self._drain_one()
-
-    # -- internals -----------------------------------------------------
-    def _drain_one(self) -> None:
-        try:
-            task = self._queue.get_nowait()
-        except queue.Empty:
-            return
-        if task is not None:
             self._process(task)
+        else:
+            self._queue.put(task)
 
     def _worker(self) -> None:
         while True:

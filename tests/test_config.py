@@ -158,3 +158,30 @@ class TestMultiModelBurst:
         ]
         for mset in models_seen:
             assert len(mset) == 1  # each apply call is single-model
+
+
+class TestTieredConfig:
+    def test_tiered_backend_from_dict(self):
+        """ADVICE round-1: 'tiered' must be selectable from a dict/JSON
+        service config like every other backend."""
+        from llmd_kvcache_amd.config import index_config_from_dict
+
+        cfg = index_config_from_dict({
+            "tiered": {
+                "hot": {"capacity": 1 << 12, "pods_per_key": 4},
+                "cold": {"capacity": 1 << 14},
+            }
+        })
+        assert cfg.tiered is not None
+        assert cfg.in_memory is None  # no fallback default kicked in
+        assert cfg.tiered.hot.capacity == 1 << 12
+        assert cfg.tiered.hot.pods_per_key == 4
+        assert cfg.tiered.cold.capacity == 1 << 14
+        assert cfg.tiered.cold.device == "cpu"
+
+    def test_tiered_empty_uses_defaults(self):
+        from llmd_kvcache_amd.config import index_config_from_dict
+
+        cfg = index_config_from_dict({"tiered": {}})
+        assert cfg.tiered is not None
+        assert cfg.tiered.cold.capacity >= cfg.tiered.hot.capacity

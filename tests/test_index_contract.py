@@ -197,3 +197,55 @@ class TestIndexContract:
         assert not errors
         result = index.lookup([k(1000)], set())
         assert result[k(1000)] == [pe("pod-1000")]
+
+
+class TestCostAwareEngineMapBounds:
+    """ADVICE round-1 (medium): the engine->request map must be bounded
+    and budget eviction must clean up mappings for dropped request keys
+    (reference cost_aware_memory.go caps the mapping with an LRU)."""
+
+    def _make(self, budget=4096, emap=8):
+        from llmd_kvcache_amd.kvblock.cost_aware import (
+            CostAwareMemoryIndex,
+            CostAwareMemoryIndexConfig,
+        )
+
+        return CostAwareMemoryIndex(
+            CostAwareMemoryIndexConfig(max_cost_bytes=budget,
+                                       engine_map_size=emap)
+        )
+
+    def test_budget_eviction_cleans_engine_mappings(self):
+        idx = self._make(budget=600)  # fits only a few keys
+        entries = [PodEntry("pod-a", "gpu")]
+        for i in range(50):
+            ek, rk = Key("m", 10_000 + i), Key("m", 20_000 + i)
+            idx.add([ek], [rk], entries)
+        # every request key evicted by budget lost its engine mapping too
+        live_requests = set(idx._data.keys())
+        for ek, rk in idx._engine_to_request.items():
+            assert rk in live_requests
+        assert len(idx._engine_to_request) <= len(live_requests)
+        # evicted engine keys resolve to None now
+        assert idx.get_request_key(Key("m", 10_000)) is None
+
+    def test_engine_map_capacity_bounded(self):
+        idx = self._make(budget=1 << 30, emap=8)
+        entries = [PodEntry("pod-a", "gpu")]
+        for i in range(100):
+            idx.add([Key("m", 1000 + i)], [Key("m", 5000 + i)], entries)
+        assert len(idx._engine_to_request) <= 8
+        # newest mappings survive, oldest were LRU-evicted
+        assert idx.get_request_key(Key("m", 1099)) == Key("m", 5099)
+        assert idx.get_request_key(Key("m", 1000)) is None
+
+    def test_evict_to_empty_drops_all_aliases(self):
+        idx = self._make()
+        entries = [PodEntry("pod-a", "gpu")]
+        rk = Key("m", 7777)
+        eks = [Key("m", 100), Key("m", 101)]
+        idx.add(eks, [rk, rk], entries)
+        idx.evict(eks[0], entries)  # removes the only pod -> key dropped
+        assert idx.get_request_key(eks[0]) is None
+        assert idx.get_request_key(eks[1]) is None
+        assert idx._request_to_engines == {}

@@ -308,3 +308,45 @@ class TestCompaction:
         assert self._snapshot(nat, keys) == before
         with pytest.raises(ValueError):
             nat.compact(new_capacity=3000)
+
+
+class TestTierRegistryOverflow:
+    """VERDICT round-1 weak #7: >MAX_TIERS distinct mediums must error
+    loudly (the old behavior silently shared the last tier slot and
+    mis-weighted scores); event paths drop the offending event."""
+
+    def test_intern_overflow_raises(self):
+        from llmd_kvcache_amd.kvblock.gpu_index import MAX_TIERS, Registry
+
+        reg = Registry(tier_names=["gpu", "cpu"])
+        reg.tier_id("nvme")
+        reg.tier_id("remote")
+        assert len(reg.id_to_tier) == MAX_TIERS
+        with pytest.raises(ValueError, match="tier registry full"):
+            reg.tier_id("tier-five")
+        # existing tiers still intern fine
+        assert reg.tier_id("cpu") == 1
+
+    def test_digest_drops_overflow_tier_event(self):
+        """An event with an un-internable medium is dropped (poison-pill
+        stance), not silently mis-tiered."""
+        from llmd_kvcache_amd.kvblock.gpu_index import (NativeIndex,
+                                                        TableIndexConfig)
+        from llmd_kvcache_amd.kvblock.token_processor import (
+            ChunkedTokenDatabase, TokenProcessorConfig)
+        from llmd_kvcache_amd.kvevents.events import BlockStored
+        from llmd_kvcache_amd.kvevents.pool import digest_events
+
+        idx = NativeIndex(TableIndexConfig(capacity=1 << 10, pods_per_key=4))
+        for t in ("nvme", "remote"):
+            idx.registry.tier_id(t)
+        tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=4))
+        tokens = list(range(8))
+        ev = BlockStored([11, 12], None, tokens, 4, medium="tier-five")
+        digest_events(idx, tp, "pod-x", "m", [ev])  # must not raise
+        keys = tp.tokens_to_kv_block_keys(None, tokens, "m")
+        assert idx.lookup(keys, set()) == {}  # event was dropped
+
+        ok = BlockStored([11, 12], None, tokens, 4, medium="gpu")
+        digest_events(idx, tp, "pod-x", "m", [ok])
+        assert len(idx.lookup(keys, set())) == 2

@@ -261,3 +261,90 @@ def _body_registry_sync(rank, world_size):
 def test_registry_sync_guard():
     results = run_distributed("_body_registry_sync")
     assert set(results.values()) == {"raised"}
+
+
+def test_sharded_matches_single_process_world4():
+    """Same differential as the world-2 test at world_size 4 (VERDICT
+    round-1 item 3: extend gloo coverage to 4 ranks)."""
+    results = run_distributed("_body_matches_single_process", world_size=4)
+    for rank, mismatches in results.items():
+        assert mismatches == [], f"rank {rank}: {mismatches[:3]}"
+
+
+def _body_follower_failure(rank, world_size):
+    """A follower whose local probe (or device staging) throws must STILL
+    participate in the mask-merge collective - otherwise rank 0 blocks
+    there forever.  With the error flag riding in the merge tensor, the
+    peers raise instead of returning silently-degraded scores, and the
+    follower keeps serving (parallel/service.py serve docstring)."""
+    import time as _time
+
+    from llmd_kvcache_amd.kvblock.gpu_index import TableIndexConfig
+    from llmd_kvcache_amd.kvblock.token_processor import (
+        ChunkedTokenDatabase,
+        TokenProcessorConfig,
+    )
+    from llmd_kvcache_amd.kvevents.events import BlockStored, EventBatch
+    from llmd_kvcache_amd.parallel.service import ShardedIndexService
+    from llmd_kvcache_amd.parallel.sharded import ShardedIndex
+
+    tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=4))
+    sharded = ShardedIndex(TableIndexConfig(capacity=1 << 12, pods_per_key=10))
+    svc = ShardedIndexService(sharded, tp)
+    tokens = list(range(32))
+    keys = tp.tokens_to_kv_block_keys(None, tokens, MODEL)
+
+    if rank == 0:
+        batch = EventBatch(
+            ts=_time.time(),
+            events=[BlockStored(list(range(100, 108)), None, tokens, 4)],
+        )
+        svc.apply_messages([("vllm-pod-1", MODEL, batch.encode())])
+        healthy = svc.score(keys, set())          # both shards contribute
+        outcomes = []
+        for _ in range(2):  # follower probe fails, then staging fails
+            try:
+                svc.score(keys, set())
+                outcomes.append("no-raise")
+            except RuntimeError as e:
+                assert "peer rank" in str(e)
+                outcomes.append("raised")
+        recovered = svc.score(keys, set())        # follower healthy again
+        svc.stop()
+        return (healthy, outcomes, recovered)
+
+    # Follower: break the local probe for exactly the 2nd score op, and
+    # the device move for exactly the 3rd (both asymmetric failures that
+    # happen inside sharded_scores' guarded region).
+    real_lookup = sharded.local.table.lookup
+    real_device = sharded.device
+    state = {"scores_seen": 0}
+
+    def flaky_lookup(*a, **kw):
+        if state["scores_seen"] == 2:
+            raise RuntimeError("injected probe failure")
+        return real_lookup(*a, **kw)
+
+    sharded.local.table.lookup = flaky_lookup
+    orig_dispatch = svc._dispatch
+
+    def counting_dispatch(op):
+        if op[0] == "score":
+            state["scores_seen"] += 1
+            if state["scores_seen"] == 3:
+                sharded.device = "not-a-device"  # .to() raises in-guard
+            else:
+                sharded.device = real_device
+        return orig_dispatch(op)
+
+    svc._dispatch = counting_dispatch
+    svc.serve()  # must reach "stop" without hanging despite 2 failures
+    return "served"
+
+
+def test_follower_failure_keeps_collective_alive():
+    results = run_distributed("_body_follower_failure")
+    healthy, outcomes, recovered = results[0]
+    assert results[1] == "served"
+    assert outcomes == ["raised", "raised"]  # explicit error, not bad data
+    assert recovered == healthy              # follower recovered fully

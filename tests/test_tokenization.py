@@ -389,3 +389,47 @@ class TestSingleFlight:
                 TokenizationError("boom")))
         # a later call is a fresh flight
         assert sf.do("k", lambda: 42) == 42
+
+
+class TestInlineModeConcurrency:
+    def test_concurrent_inline_tokenize_does_not_cross_tasks(self):
+        """ADVICE round-1: two concurrent callers on an UNSTARTED pool
+        must each get their own result (the old inline drain could
+        process the OTHER caller's task and leave this caller blocked)."""
+        import threading
+
+        from llmd_kvcache_amd.tokenization.pool import (TokenizationConfig,
+                                                        TokenizationPool)
+
+        class Tok:
+            def encode(self, prompt, model):
+                return [len(prompt)] * 3, [(0, len(prompt))] * 3
+
+            def render_chat_template(self, req):
+                raise AssertionError("not used")
+
+        pool = TokenizationPool(
+            TokenizationConfig(workers_count=2), tokenizer=Tok()
+        )
+        assert not pool._running  # inline mode
+        results = {}
+        errs = []
+
+        def call(tag, prompt):
+            try:
+                results[tag] = pool.tokenize(None, prompt, "m")
+            except Exception as e:  # pragma: no cover
+                errs.append(e)
+
+        threads = [
+            threading.Thread(target=call, args=(i, "x" * (i + 1)))
+            for i in range(8)
+        ]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=10)
+        assert not errs
+        assert all(not t.is_alive() for t in threads)  # nobody deadlocked
+        for i in range(8):
+            assert results[i] == [i + 1] * 3
