@@ -36,6 +36,12 @@ class TokenProcessorConfig:
     # current vLLM/reference behavior and the only algo with C++/HIP
     # fast paths; alternatives run on the Python path.
     hash_algo: str = "fnv-64a"
+    # Host-side session/prefix chain cache (kvblock/chain_cache.py):
+    # warm shared prefixes skip the serial FNV/CBOR chain entirely.
+    # 0 disables.  Applies only to root-parented chains (the read path);
+    # event chains with explicit parents bypass it.
+    chain_cache_entries: int = 1 << 16
+    chain_cache_seg_chunks: int = 32
     _init_hash: Optional[int] = field(default=None, repr=False)
 
     def init_hash(self) -> int:
@@ -57,13 +63,50 @@ class ChunkedTokenDatabase:
             self._native = cpu_ext.maybe_load()
         except Exception:  # pragma: no cover - ops package always importable
             self._native = None
+        self.chain_cache = None
+        if self.config.chain_cache_entries > 0:
+            from .chain_cache import ChainCache
+
+            self.chain_cache = ChainCache(
+                max_entries=self.config.chain_cache_entries,
+                seg_chunks=self.config.chain_cache_seg_chunks,
+            )
 
     @property
     def block_size(self) -> int:
         return self.config.block_size
 
     def chunk_hashes(self, parent_hash: int, tokens: Sequence[int]) -> List[int]:
-        """Full-chunk chain hashes starting from parent_hash."""
+        """Full-chunk chain hashes starting from parent_hash.  Warm
+        root-parented prefixes are served from the session chain cache
+        (only the uncached tail is computed, seeded by the last cached
+        hash)."""
+        bs = self.config.block_size
+        n_chunks = len(tokens) // bs
+        if n_chunks == 0:
+            return []
+        cache = self.chain_cache
+        if (cache is not None
+                and parent_hash == self.config.init_hash()
+                and n_chunks >= cache.seg_chunks):
+            from .chain_cache import ChainCache
+
+            tokens_b = ChainCache._token_bytes(tokens[: n_chunks * bs])
+            covered, chain = cache.lookup(tokens_b, bs, n_chunks)
+            if covered >= n_chunks:
+                return list(chain[:n_chunks])
+            if covered:
+                tail = self._chain(chain[covered - 1],
+                                   tokens[covered * bs: n_chunks * bs])
+                full = list(chain[:covered]) + tail
+            else:
+                full = self._chain(parent_hash, tokens[: n_chunks * bs])
+            cache.store(tokens_b, bs, full)
+            return full
+        return self._chain(parent_hash, tokens[: n_chunks * bs])
+
+    def _chain(self, parent_hash: int, tokens: Sequence[int]) -> List[int]:
+        """Uncached chain over full chunks (C++ fast path or Python)."""
         bs = self.config.block_size
         n_chunks = len(tokens) // bs
         if n_chunks == 0:

@@ -167,3 +167,77 @@ class TestPluggableHash:
         cfg = TokenProcessorConfig(hash_algo="md5")
         with pytest.raises(KeyError):
             cfg.init_hash()
+
+
+class TestChainCache:
+    """Host-side session chain cache (kvblock/chain_cache.py): warm
+    prefixes skip the serial chain, results stay bit-identical."""
+
+    def _tp(self, **kw):
+        from llmd_kvcache_amd.kvblock.token_processor import (
+            ChunkedTokenDatabase, TokenProcessorConfig)
+
+        return ChunkedTokenDatabase(TokenProcessorConfig(**kw))
+
+    def test_warm_hit_bit_identical(self):
+        tp = self._tp(chain_cache_seg_chunks=4)
+        cold = self._tp(chain_cache_entries=0)
+        toks = list(range(16 * 16))  # 16 chunks
+        a = tp.chunk_hashes(tp.config.init_hash(), toks)
+        assert tp.chain_cache.misses == 1
+        b = tp.chunk_hashes(tp.config.init_hash(), toks)
+        assert tp.chain_cache.hits >= 1
+        assert a == b == cold.chunk_hashes(cold.config.init_hash(), toks)
+
+    def test_partial_prefix_hit(self):
+        tp = self._tp(chain_cache_seg_chunks=4)
+        cold = self._tp(chain_cache_entries=0)
+        base = list(range(16 * 16))
+        tp.chunk_hashes(tp.config.init_hash(), base)
+        # shares the first 8 chunks (2 segments), then diverges
+        variant = base[: 8 * 16] + [99999 + i for i in range(8 * 16)]
+        got = tp.chunk_hashes(tp.config.init_hash(), variant)
+        assert tp.chain_cache.hits == 1
+        assert got == cold.chunk_hashes(cold.config.init_hash(), variant)
+        # the diverged tail differs from the base chain
+        assert got[8:] != tp.chunk_hashes(tp.config.init_hash(), base)[8:]
+
+    def test_growing_session_extends_chain(self):
+        tp = self._tp(chain_cache_seg_chunks=4)
+        cold = self._tp(chain_cache_entries=0)
+        s1 = list(range(8 * 16))
+        s2 = s1 + list(range(1000, 1000 + 8 * 16))  # session grows
+        tp.chunk_hashes(tp.config.init_hash(), s1)
+        got = tp.chunk_hashes(tp.config.init_hash(), s2)
+        assert got == cold.chunk_hashes(cold.config.init_hash(), s2)
+        # and the longer chain is now cached in full
+        tp.chain_cache.hits = 0
+        tp.chunk_hashes(tp.config.init_hash(), s2)
+        assert tp.chain_cache.hits == 1
+
+    def test_non_root_parent_bypasses_cache(self):
+        tp = self._tp(chain_cache_seg_chunks=4)
+        toks = list(range(8 * 16))
+        tp.chunk_hashes(12345, toks)  # event-style explicit parent
+        assert tp.chain_cache.hits == 0
+        assert tp.chain_cache.misses == 0
+        assert len(tp.chain_cache) == 0
+
+    def test_eviction_bounded(self):
+        tp = self._tp(chain_cache_seg_chunks=4, chain_cache_entries=8)
+        for i in range(20):
+            toks = [i * 1000 + j for j in range(4 * 16)]
+            tp.chunk_hashes(tp.config.init_hash(), toks)
+        assert len(tp.chain_cache) <= 8
+
+    def test_disabled(self):
+        tp = self._tp(chain_cache_entries=0)
+        assert tp.chain_cache is None
+        toks = list(range(4 * 16))
+        assert len(tp.chunk_hashes(tp.config.init_hash(), toks)) == 4
+
+    def test_short_prompt_bypasses(self):
+        tp = self._tp(chain_cache_seg_chunks=32)
+        toks = list(range(4 * 16))  # 4 chunks < seg_chunks
+        tp.chunk_hashes(tp.config.init_hash(), toks)
+        assert len(tp.chain_cache) == 0
