@@ -198,11 +198,11 @@ def main():
                          "16384->13.5M, 32768->22.5M scores/s (p50 "
                          "1.10->1.37 ms). Beyond that wall grows with "
                          "batch: 131072 peaks ~30M at p50 4.5 ms.")
-    ap.add_argument("--calls-per-step", type=int, default=16,
+    ap.add_argument("--calls-per-step", type=int, default=32,
                     help="scoring calls per timed step; with the default "
-                         "batch each call is ~3 ms, so 16 calls x 20 steps "
-                         "gives a ~1 s timed region (round-1 verdict: a "
-                         "0.06 s window was too thin)")
+                         "batch each call is ~1.5 ms, so 32 calls x 20 "
+                         "steps gives a ~1 s timed region (round-1 "
+                         "verdict: a 0.06 s window was too thin)")
     ap.add_argument("--distinct-calls", type=int, default=2,
                     help="distinct pre-staged prompt batches cycled "
                          "through (bounds host/GPU prompt memory; every "
