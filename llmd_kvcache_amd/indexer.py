@@ -170,11 +170,69 @@ class Indexer:
         """Batched scoring: N prompts through ONE fused kernel call
         (the designed hot interface - bench.py measures this shape at
         22M scores/s). Falls back to per-prompt scoring on generic
-        backends."""
-        keys_per_prompt = [
-            self.tokens_processor.tokens_to_kv_block_keys(None, t, model_name)
-            for t in token_lists
-        ]
+        backends.
+
+        The chain step runs through the PARALLEL C++ op
+        (ops.hash_chain_batch, at::parallel_for across prompts) rather
+        than a per-prompt Python loop - at wire-level request rates the
+        chains are the dominant host cost (round-2: the coalescing
+        service front feeds this method)."""
+        fused = getattr(self._kv_block_index, "fused_scores", None)
+        tp = self.tokens_processor
+        native = (fused is not None and tp._native is not None
+                  and tp.config.hash_algo == "fnv-64a")
+        if not native:
+            keys_per_prompt = [
+                tp.tokens_to_kv_block_keys(None, t, model_name)
+                for t in token_lists
+            ]
+            return self._score_keys_batch(keys_per_prompt, model_name,
+                                          pod_identifiers)
+
+        import numpy as np
+        import torch
+
+        from .kvblock.gpu_index import _to_i64
+
+        idx = self._kv_block_index
+        bs = tp.block_size
+        init = _to_i64(tp.config.init_hash())
+        B = len(token_lists)
+        lens = [len(t) for t in token_lists]
+        flat = np.empty(sum(lens), dtype=np.int64)
+        off = np.zeros(B + 1, dtype=np.int64)
+        pos = 0
+        for i, t in enumerate(token_lists):
+            n = lens[i]
+            flat[pos:pos + n] = np.asarray(t, dtype=np.int64)
+            pos += n
+            off[i + 1] = pos
+        parents = torch.full((B,), init, dtype=torch.int64)
+        hashes, chunk_off = idx.table.ops.hash_chain_batch(
+            torch.from_numpy(flat), torch.from_numpy(off), parents, bs)
+        counts = (chunk_off[1:] - chunk_off[:-1]).to(torch.int32)
+        max_k = int(counts.max()) if B else 0
+        if max_k == 0:
+            return [{} for _ in range(B)]
+        weights = idx.tier_weights(
+            {b.name: b.weight for b in self.config.backend_configs}
+        )
+        if idx.table.is_cuda:
+            device = idx.device
+            scores = fused(hashes.to(device),
+                           chunk_off.to(torch.int32).to(device),
+                           model_name, set(pod_identifiers), weights,
+                           max_k=max_k)
+        else:
+            scores = fused(hashes, counts, model_name,
+                           set(pod_identifiers), weights, max_k=max_k)
+        return idx.scores_to_map(scores)
+
+    def _score_keys_batch(
+        self, keys_per_prompt, model_name: str,
+        pod_identifiers: Sequence[str],
+    ) -> List[Dict[str, float]]:
+        """Generic-backend batched scoring (per-prompt calls)."""
         fused = getattr(self._kv_block_index, "fused_scores", None)
         if fused is None or not any(keys_per_prompt):
             return [self._score_keys(ks, pod_identifiers) if ks else {}

@@ -19,8 +19,23 @@ from ..preprocessing import chat_completions as cc
 
 
 class AsgiIndexerApp:
-    def __init__(self, indexer: Indexer):
+    def __init__(self, indexer: Indexer, coalesce: bool = True):
+        """coalesce=True (default) batches concurrent /score_completions
+        requests into shared fused kernel launches (service/coalesce.py);
+        the blocking tokenize+score runs on the default executor so the
+        event loop keeps accepting requests while a batch is in flight."""
         self.indexer = indexer
+        self.coalescer = None
+        if (coalesce and hasattr(indexer, "tokenizers_pool")
+                and hasattr(indexer, "score_tokens_batch")):
+            from .coalesce import CoalescingScorer
+
+            self.coalescer = CoalescingScorer(indexer)
+            self.coalescer.start()
+
+    def _score_one_blocking(self, prompt: str, model: str, pods):
+        tokens = self.indexer.tokenizers_pool.tokenize(None, prompt, model)
+        return self.coalescer.score(tokens, model, pods)
 
     async def __call__(self, scope, receive, send):
         if scope["type"] == "lifespan":
@@ -56,9 +71,16 @@ class AsgiIndexerApp:
         prompt = body.get("prompt", "")
         if not prompt:
             raise _BadRequest("field 'prompt' required")
-        scores = self.indexer.get_pod_scores(
-            None, prompt, body.get("model", ""), []
-        )
+        model = body.get("model", "")
+        if self.coalescer is not None:
+            import asyncio
+
+            loop = asyncio.get_running_loop()
+            scores = await loop.run_in_executor(
+                None, self._score_one_blocking, prompt, model,
+                body.get("pods", []))
+        else:
+            scores = self.indexer.get_pod_scores(None, prompt, model, [])
         await self._json(send, 200, scores or {})
 
     async def _score_batch(self, send, body: Dict[str, Any]):

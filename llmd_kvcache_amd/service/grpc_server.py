@@ -29,15 +29,25 @@ logger = logging.getLogger("llmd_kvcache_amd.grpc")
 
 
 class IndexerServicer:
-    def __init__(self, indexer: Indexer):
+    def __init__(self, indexer: Indexer, coalescer=None):
+        """coalescer: optional service.coalesce.CoalescingScorer - when
+        set (the default in serve()), concurrent RPCs share one fused
+        kernel launch per micro-batch instead of one launch each."""
         self.indexer = indexer
+        self.coalescer = coalescer
 
     def GetPodScores(self, request: GetPodScoresRequest, context):
         try:
-            scores: Dict[str, float] = self.indexer.get_pod_scores(
-                None, request.prompt, request.model_name,
-                request.pod_identifiers,
-            )
+            if self.coalescer is not None:
+                tokens = self.indexer.tokenizers_pool.tokenize(
+                    None, request.prompt, request.model_name)
+                scores: Dict[str, float] = self.coalescer.score(
+                    tokens, request.model_name, request.pod_identifiers)
+            else:
+                scores = self.indexer.get_pod_scores(
+                    None, request.prompt, request.model_name,
+                    request.pod_identifiers,
+                )
         except Exception as e:
             logger.exception("GetPodScores failed")
             context.abort(grpc.StatusCode.INTERNAL, str(e))
@@ -51,10 +61,21 @@ def serve(
     indexer: Indexer,
     address: str = "0.0.0.0:50051",
     max_workers: int = 16,
+    coalesce: bool = True,
 ) -> grpc.Server:
-    """Starts a non-blocking gRPC server; returns the server handle."""
+    """Starts a non-blocking gRPC server; returns the server handle.
+    coalesce=True (default) batches concurrent RPCs into shared fused
+    kernel launches (service/coalesce.py)."""
+    coalescer = None
+    if (coalesce and hasattr(indexer, "tokenizers_pool")
+            and hasattr(indexer, "score_tokens_batch")):
+        from .coalesce import CoalescingScorer
+
+        coalescer = CoalescingScorer(indexer)
+        coalescer.start()
     server = grpc.server(futures.ThreadPoolExecutor(max_workers=max_workers))
-    servicer = IndexerServicer(indexer)
+    servicer = IndexerServicer(indexer, coalescer)
+    server._kvidx_coalescer = coalescer  # stopped with the server by callers
     handlers = {
         "GetPodScores": grpc.unary_unary_rpc_method_handler(
             servicer.GetPodScores,

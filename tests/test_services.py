@@ -379,3 +379,90 @@ class TestScoreBatch:
             conn.close()
         finally:
             http_svc.stop()
+
+
+class TestCoalescingScorer:
+    """service/coalesce.py: concurrent single-prompt calls share fused
+    kernel launches (round-1 verdict item 1)."""
+
+    def _indexer(self):
+        from llmd_kvcache_amd.indexer import Config, Indexer
+        from llmd_kvcache_amd.kvblock.gpu_index import (NativeIndex,
+                                                        TableIndexConfig)
+        from llmd_kvcache_amd.kvblock.keys import PodEntry
+        from llmd_kvcache_amd.kvblock.token_processor import (
+            ChunkedTokenDatabase, TokenProcessorConfig)
+
+        cfg = Config(token_processor=TokenProcessorConfig(block_size=4))
+        idx = NativeIndex(TableIndexConfig(capacity=1 << 12, pods_per_key=4))
+        indexer = Indexer(cfg, kv_block_index=idx)
+        tp = ChunkedTokenDatabase(cfg.token_processor)
+        tokens = list(range(32))
+        keys = tp.tokens_to_kv_block_keys(None, tokens, "m")
+        idx.add(keys, keys, [PodEntry("pod-a", "gpu")])
+        return indexer, tokens
+
+    def test_concurrent_calls_coalesce(self):
+        import threading
+
+        from llmd_kvcache_amd.service.coalesce import CoalescingScorer
+
+        indexer, tokens = self._indexer()
+        sc = CoalescingScorer(indexer)
+        sc.start()
+        try:
+            results = [None] * 64
+            barrier = threading.Barrier(16)
+
+            def call(i):
+                barrier.wait()
+                for j in range(4):
+                    results[i * 4 + j] = sc.score(tokens, "m", [])
+
+            threads = [threading.Thread(target=call, args=(i,))
+                       for i in range(16)]
+            for t in threads:
+                t.start()
+            for t in threads:
+                t.join(timeout=20)
+            assert all(r == {"pod-a": 8.0} for r in results)
+            assert sc.requests == 64
+            assert sc.batches < 64  # concurrency actually coalesced
+        finally:
+            sc.stop()
+
+    def test_unstarted_falls_back_direct(self):
+        from llmd_kvcache_amd.service.coalesce import CoalescingScorer
+
+        indexer, tokens = self._indexer()
+        sc = CoalescingScorer(indexer)
+        assert sc.score(tokens, "m", []) == {"pod-a": 8.0}
+
+    def test_error_propagates_to_caller(self):
+        from llmd_kvcache_amd.service.coalesce import CoalescingScorer
+
+        indexer, tokens = self._indexer()
+
+        def boom(*a, **kw):
+            raise RuntimeError("injected")
+
+        indexer.score_tokens_batch = boom
+        sc = CoalescingScorer(indexer)
+        sc.start()
+        try:
+            with pytest.raises(RuntimeError, match="injected"):
+                sc.score(tokens, "m", [])
+        finally:
+            sc.stop()
+
+    def test_pod_filter_groups(self):
+        from llmd_kvcache_amd.service.coalesce import CoalescingScorer
+
+        indexer, tokens = self._indexer()
+        sc = CoalescingScorer(indexer)
+        sc.start()
+        try:
+            assert sc.score(tokens, "m", ["pod-a"]) == {"pod-a": 8.0}
+            assert sc.score(tokens, "m", ["nobody"]) == {}
+        finally:
+            sc.stop()
