@@ -192,11 +192,6 @@ class Indexer:
         import numpy as np
         import torch
 
-        from .kvblock.gpu_index import _to_i64
-
-        idx = self._kv_block_index
-        bs = tp.block_size
-        init = _to_i64(tp.config.init_hash())
         B = len(token_lists)
         lens = [len(t) for t in token_lists]
         flat = np.empty(sum(lens), dtype=np.int64)
@@ -207,26 +202,51 @@ class Indexer:
             flat[pos:pos + n] = np.asarray(t, dtype=np.int64)
             pos += n
             off[i + 1] = pos
+        scores = self.score_flat_tokens(
+            torch.from_numpy(flat), torch.from_numpy(off), model_name,
+            pod_identifiers)
+        return self._kv_block_index.scores_to_map(scores)
+
+    def score_flat_tokens(
+        self,
+        tokens_flat,
+        offsets,
+        model_name: str,
+        pod_identifiers: Sequence[str],
+    ):
+        """Core batched scorer over flat int64 token tensors: parallel
+        C++ chain -> one fused probe/score.  Returns the [B, num_pods]
+        float32 score tensor (CPU or device).  This is the entry the C++
+        wirefront calls once per micro-batch."""
+        import torch
+
+        from .kvblock.gpu_index import _to_i64
+
+        tp = self.tokens_processor
+        idx = self._kv_block_index
+        bs = tp.block_size
+        B = int(offsets.numel()) - 1
+        init = _to_i64(tp.config.init_hash())
         parents = torch.full((B,), init, dtype=torch.int64)
         hashes, chunk_off = idx.table.ops.hash_chain_batch(
-            torch.from_numpy(flat), torch.from_numpy(off), parents, bs)
+            tokens_flat, offsets, parents, bs)
         counts = (chunk_off[1:] - chunk_off[:-1]).to(torch.int32)
         max_k = int(counts.max()) if B else 0
+        num_pods = idx._num_pods_padded()
         if max_k == 0:
-            return [{} for _ in range(B)]
+            return torch.zeros((B, num_pods), dtype=torch.float32)
         weights = idx.tier_weights(
             {b.name: b.weight for b in self.config.backend_configs}
         )
+        fused = idx.fused_scores
         if idx.table.is_cuda:
             device = idx.device
-            scores = fused(hashes.to(device),
-                           chunk_off.to(torch.int32).to(device),
-                           model_name, set(pod_identifiers), weights,
-                           max_k=max_k)
-        else:
-            scores = fused(hashes, counts, model_name,
-                           set(pod_identifiers), weights, max_k=max_k)
-        return idx.scores_to_map(scores)
+            return fused(hashes.to(device),
+                         chunk_off.to(torch.int32).to(device),
+                         model_name, set(pod_identifiers), weights,
+                         max_k=max_k)
+        return fused(hashes, counts, model_name,
+                     set(pod_identifiers), weights, max_k=max_k)
 
     def _score_keys_batch(
         self, keys_per_prompt, model_name: str,

@@ -82,6 +82,10 @@ void gpu_apply_events_split(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                             int64_t, int64_t, int64_t);
 #endif
 
+namespace wire {
+void register_wirefront(pybind11::module_& m);  // wirefront.cpp
+}
+
 }  // namespace kvidx
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -115,4 +119,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
 #else
   m.attr("HAS_HIP") = false;
 #endif
+  kvidx::wire::register_wirefront(m);
 }

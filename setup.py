@@ -22,6 +22,7 @@ CSRC = os.path.join(ROOT, "llmd_kvcache_amd", "ops", "csrc")
 sources = [
     os.path.join(CSRC, "bindings.cpp"),
     os.path.join(CSRC, "cpu_ops.cpp"),
+    os.path.join(CSRC, "wirefront.cpp"),
 ]
 
 with_hip = torch.version.hip is not None

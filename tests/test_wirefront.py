@@ -1,0 +1,226 @@
+"""Native (C++) wire front tests: real sockets against the epoll server
+in ops/csrc/wirefront.cpp, CPU table backend (the identical code path
+serves the GPU table on an MI355X host)."""
+
+import json
+import socket
+
+import pytest
+
+from llmd_kvcache_amd.ops import cpu_ext
+
+pytestmark = pytest.mark.skipif(
+    cpu_ext.maybe_load() is None, reason="native extension not built"
+)
+
+MODEL = "m"
+BS = 4
+
+
+def _http_post(path, obj):
+    body = json.dumps(obj).encode()
+    return (f"POST {path} HTTP/1.1\r\nhost: x\r\n"
+            f"content-length: {len(body)}\r\n\r\n").encode() + body
+
+
+def _read_response(sock, buf=b""):
+    """Reads exactly one HTTP response; returns (status, body, rest)."""
+    while b"\r\n\r\n" not in buf:
+        chunk = sock.recv(65536)
+        assert chunk, "connection closed early"
+        buf += chunk
+    head, rest = buf.split(b"\r\n\r\n", 1)
+    status = int(head.split(b" ", 2)[1])
+    clen = 0
+    for line in head.split(b"\r\n")[1:]:
+        k, _, v = line.partition(b":")
+        if k.lower() == b"content-length":
+            clen = int(v.strip())
+    while len(rest) < clen:
+        chunk = sock.recv(65536)
+        assert chunk, "connection closed mid-body"
+        rest += chunk
+    return status, rest[:clen], rest[clen:]
+
+
+@pytest.fixture()
+def service():
+    from llmd_kvcache_amd.indexer import Config, Indexer
+    from llmd_kvcache_amd.kvblock.gpu_index import (NativeIndex,
+                                                    TableIndexConfig)
+    from llmd_kvcache_amd.kvblock.keys import PodEntry
+    from llmd_kvcache_amd.kvblock.token_processor import (
+        ChunkedTokenDatabase, TokenProcessorConfig)
+    from llmd_kvcache_amd.service.wirefront import WireIndexerService
+    from llmd_kvcache_amd.tokenization.pool import TokenizationPool
+    from llmd_kvcache_amd.tokenization.tokenizer import Tokenizer
+
+    class WordTokenizer(Tokenizer):
+        """Splits on spaces; token id = int(word)."""
+
+        def encode(self, prompt, model_name):
+            ids = [int(w) for w in prompt.split()]
+            offs = []
+            pos = 0
+            for w in prompt.split():
+                offs.append((pos, pos + len(w)))
+                pos += len(w) + 1
+            return ids, offs
+
+        def render_chat_template(self, req):  # pragma: no cover
+            raise NotImplementedError
+
+    cfg = Config(token_processor=TokenProcessorConfig(block_size=BS))
+    idx = NativeIndex(TableIndexConfig(capacity=1 << 12, pods_per_key=4))
+    pool = TokenizationPool(tokenizer=WordTokenizer())
+    indexer = Indexer(cfg, tokenization_pool=pool, kv_block_index=idx)
+
+    tp = ChunkedTokenDatabase(cfg.token_processor)
+    tokens = list(range(32))
+    keys = tp.tokens_to_kv_block_keys(None, tokens, MODEL)
+    idx.add(keys, keys, [PodEntry("pod-a", "gpu")])
+    idx.add(keys[:4], keys[:4], [PodEntry("pod-b", "cpu")])
+
+    svc = WireIndexerService(indexer)
+    port = svc.start(port=0, n_io=2)
+    yield svc, port, tokens
+    svc.stop()
+
+
+def _connect(port):
+    s = socket.create_connection(("127.0.0.1", port), timeout=10)
+    s.settimeout(10)
+    return s
+
+
+class TestWireFront:
+    def test_health(self, service):
+        _, port, _ = service
+        s = _connect(port)
+        s.sendall(b"GET /health HTTP/1.1\r\nhost: x\r\n\r\n")
+        status, body, _ = _read_response(s)
+        assert status == 200 and json.loads(body) == {"status": "ok"}
+        s.close()
+
+    def test_score_tokens(self, service):
+        _, port, tokens = service
+        s = _connect(port)
+        s.sendall(_http_post("/score", {"model": MODEL, "tokens": tokens}))
+        status, body, _ = _read_response(s)
+        assert status == 200
+        scores = json.loads(body)["scores"]
+        assert scores == {"pod-a": 8.0, "pod-b": pytest.approx(3.2)}
+        s.close()
+
+    def test_score_prompt_includes_tokenization(self, service):
+        _, port, tokens = service
+        prompt = " ".join(str(t) for t in tokens)
+        s = _connect(port)
+        s.sendall(_http_post("/score", {"model": MODEL, "prompt": prompt}))
+        status, body, _ = _read_response(s)
+        assert status == 200
+        assert json.loads(body)["scores"]["pod-a"] == 8.0
+        s.close()
+
+    def test_pod_filter(self, service):
+        _, port, tokens = service
+        s = _connect(port)
+        s.sendall(_http_post("/score", {"model": MODEL, "tokens": tokens,
+                                        "pods": ["pod-b"]}))
+        status, body, _ = _read_response(s)
+        assert json.loads(body)["scores"] == {"pod-b": pytest.approx(3.2)}
+        s.close()
+
+    def test_miss_returns_empty(self, service):
+        _, port, _ = service
+        s = _connect(port)
+        s.sendall(_http_post("/score", {"model": MODEL,
+                                        "tokens": [9999] * 8}))
+        status, body, _ = _read_response(s)
+        assert status == 200 and json.loads(body)["scores"] == {}
+        s.close()
+
+    def test_pipelined_requests_ordered(self, service):
+        """HTTP/1.1 pipelining: N requests in one write, N responses in
+        order (mixing /health inline responses with batched /score)."""
+        _, port, tokens = service
+        s = _connect(port)
+        blob = b""
+        for i in range(10):
+            if i % 3 == 2:
+                blob += b"GET /health HTTP/1.1\r\nhost: x\r\n\r\n"
+            else:
+                blob += _http_post("/score",
+                                   {"model": MODEL, "tokens": tokens})
+        s.sendall(blob)
+        rest = b""
+        for i in range(10):
+            status, body, rest = _read_response(s, rest)
+            assert status == 200
+            parsed = json.loads(body)
+            if i % 3 == 2:
+                assert parsed == {"status": "ok"}
+            else:
+                assert parsed["scores"]["pod-a"] == 8.0
+        s.close()
+
+    def test_bad_json_400(self, service):
+        _, port, _ = service
+        s = _connect(port)
+        body = b"{nonsense"
+        s.sendall((f"POST /score HTTP/1.1\r\nhost: x\r\n"
+                   f"content-length: {len(body)}\r\n\r\n").encode() + body)
+        status, _, _ = _read_response(s)
+        assert status == 400
+        s.close()
+
+    def test_unknown_path_404(self, service):
+        _, port, _ = service
+        s = _connect(port)
+        s.sendall(b"GET /nope HTTP/1.1\r\nhost: x\r\n\r\n")
+        status, _, _ = _read_response(s)
+        assert status == 404
+        s.close()
+
+    def test_concurrent_connections_coalesce(self, service):
+        import threading
+
+        svc, port, tokens = service
+        req = _http_post("/score", {"model": MODEL, "tokens": tokens})
+        n_conn, per_conn = 8, 20
+        errs = []
+
+        def worker():
+            try:
+                s = _connect(port)
+                s.sendall(req * per_conn)  # pipelined burst
+                rest = b""
+                for _ in range(per_conn):
+                    status, body, rest = _read_response(s, rest)
+                    assert status == 200
+                    assert json.loads(body)["scores"]["pod-a"] == 8.0
+                s.close()
+            except Exception as e:  # pragma: no cover
+                errs.append(e)
+
+        threads = [threading.Thread(target=worker) for _ in range(n_conn)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=30)
+        assert not errs
+        reqs, batches = svc.stats()
+        assert reqs >= n_conn * per_conn
+        assert batches < reqs  # micro-batching actually engaged
+
+    def test_unicode_prompt_escapes(self, service):
+        _, port, _ = service
+        s = _connect(port)
+        # escaped-unicode digits: 12 -> "12"; tokenizer sees "12 7"
+        body = (b'{"model":"m","prompt":"\\u0031\\u0032 7"}')
+        s.sendall((f"POST /score HTTP/1.1\r\nhost: x\r\n"
+                   f"content-length: {len(body)}\r\n\r\n").encode() + body)
+        status, rbody, _ = _read_response(s)
+        assert status == 200  # too short for a full block -> empty scores
+        assert json.loads(rbody)["scores"] == {}
+        s.close()
