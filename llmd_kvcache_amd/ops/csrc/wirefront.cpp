@@ -43,7 +43,9 @@
 #include <condition_variable>
 #include <cstring>
 #include <deque>
+#include <list>
 #include <map>
+#include <unordered_map>
 #include <memory>
 #include <mutex>
 #include <string>
@@ -268,6 +270,82 @@ static bool skip_json_value(JsonCursor& c) {
   return true;
 }
 
+// 128-bit prompt fingerprint: two independent FNV-1a streams over
+// (model \x00 prompt).  Collision odds at the cache cap are ~2^-90;
+// a collision would mis-route one scoring decision, never corrupt state.
+struct PromptKey {
+  uint64_t a, b;
+  bool operator==(const PromptKey& o) const { return a == o.a && b == o.b; }
+};
+struct PromptKeyHash {
+  size_t operator()(const PromptKey& k) const { return (size_t)(k.a ^ k.b); }
+};
+static PromptKey prompt_fingerprint(const std::string& model,
+                                    const std::string& prompt) {
+  uint64_t h1 = 14695981039346656037ull;
+  uint64_t h2 = 0x9e3779b97f4a7c15ull;
+  auto mix = [&](unsigned char c) {
+    h1 = (h1 ^ c) * 1099511628211ull;
+    h2 = (h2 ^ c) * 0xff51afd7ed558ccdull;
+    h2 ^= h2 >> 29;
+  };
+  for (unsigned char c : model) mix(c);
+  mix(0);
+  for (unsigned char c : prompt) mix(c);
+  return {h1, h2};
+}
+
+// Bounded LRU of prompt -> token ids (the C++ analog, one level up, of
+// the reference's prefix store: a warm session prompt skips Python and
+// the tokenizer entirely and rides the pre-tokenized batch path).
+class PromptCache {
+ public:
+  explicit PromptCache(size_t max_entries) : cap_(max_entries) {}
+
+  bool get(const PromptKey& k, std::vector<int64_t>& out) {
+    std::lock_guard<std::mutex> lk(mu_);
+    auto it = map_.find(k);
+    if (it == map_.end()) return false;
+    lru_.splice(lru_.begin(), lru_, it->second.first);
+    const auto& v = it->second.second;
+    out.assign(v.begin(), v.end());
+    ++hits_;
+    return true;
+  }
+
+  void put(const PromptKey& k, const int32_t* tok, size_t n) {
+    std::lock_guard<std::mutex> lk(mu_);
+    auto it = map_.find(k);
+    if (it != map_.end()) {
+      lru_.splice(lru_.begin(), lru_, it->second.first);
+      return;
+    }
+    lru_.push_front(k);
+    map_.emplace(k, std::make_pair(lru_.begin(),
+                                   std::vector<int32_t>(tok, tok + n)));
+    while (map_.size() > cap_) {
+      map_.erase(lru_.back());
+      lru_.pop_back();
+    }
+  }
+
+  uint64_t hits() const {
+    std::lock_guard<std::mutex> lk(mu_);
+    return hits_;
+  }
+
+ private:
+  size_t cap_;
+  mutable std::mutex mu_;
+  std::list<PromptKey> lru_;
+  std::unordered_map<PromptKey,
+                     std::pair<std::list<PromptKey>::iterator,
+                               std::vector<int32_t>>,
+                     PromptKeyHash>
+      map_;
+  uint64_t hits_ = 0;
+};
+
 struct ScoreRequest {
   std::string model;
   std::string prompt;           // text mode
@@ -275,6 +353,8 @@ struct ScoreRequest {
   std::vector<std::string> pods;
   bool has_tokens = false;
   bool has_prompt = false;
+  bool want_cache = false;      // miss: store tokens after tokenization
+  PromptKey cache_key{0, 0};
 };
 
 static bool parse_score_request(const char* body, size_t len,
@@ -411,6 +491,7 @@ class WireFront {
   int port() const { return bound_port_; }
   uint64_t requests() const { return n_requests_.load(); }
   uint64_t batches() const { return n_batches_.load(); }
+  uint64_t prompt_cache_hits() const { return prompt_cache_.hits(); }
 
  private:
   void stop_nogil() {
@@ -588,6 +669,19 @@ class WireFront {
       } else if (method == "POST" && path == "/score") {
         ScoreRequest req;
         if (parse_score_request(c->inbuf.data() + hdr_end + 4, clen, req)) {
+          if (req.has_prompt && !req.has_tokens) {
+            // warm-session fast path: a cached prompt becomes a
+            // pre-tokenized request right here in the io thread
+            PromptKey pk = prompt_fingerprint(req.model, req.prompt);
+            if (prompt_cache_.get(pk, req.tokens)) {
+              req.has_tokens = true;
+              req.has_prompt = false;
+              req.prompt.clear();
+            } else {
+              req.cache_key = pk;
+              req.want_cache = true;
+            }
+          }
           uint64_t slot = c->next_slot++;
           parsed.push_back(PendingReq{conn, slot, std::move(req)});
         } else {
@@ -733,6 +827,21 @@ class WireFront {
           py::list prompts;
           for (size_t i : idxs) prompts.append(batch[i].req.prompt);
           result = score_text_cb_(model, pods_t, prompts).cast<py::tuple>();
+          if (result.size() >= 4) {
+            // (scores, names, tokens_flat_i32, offsets_i64): feed the
+            // prompt cache so the NEXT request for these prompts takes
+            // the pre-tokenized path without touching Python
+            auto tf = result[2].cast<at::Tensor>().contiguous();
+            auto of = result[3].cast<at::Tensor>().contiguous();
+            const int32_t* tp = tf.data_ptr<int32_t>();
+            const int64_t* op = of.data_ptr<int64_t>();
+            for (size_t k = 0; k < idxs.size(); ++k) {
+              auto& rq = batch[idxs[k]].req;
+              if (rq.want_cache)
+                prompt_cache_.put(rq.cache_key, tp + op[k],
+                                  (size_t)(op[k + 1] - op[k]));
+            }
+          }
         } else {
           int64_t total = 0;
           for (size_t i : idxs) total += (int64_t)batch[i].req.tokens.size();
@@ -831,6 +940,7 @@ class WireFront {
 
   std::atomic<uint64_t> n_requests_{0};
   std::atomic<uint64_t> n_batches_{0};
+  PromptCache prompt_cache_{2048};
 };
 
 void register_wirefront(py::module_& m) {
@@ -843,7 +953,8 @@ void register_wirefront(py::module_& m) {
       .def("stop", &WireFront::stop)
       .def("port", &WireFront::port)
       .def("requests", &WireFront::requests)
-      .def("batches", &WireFront::batches);
+      .def("batches", &WireFront::batches)
+      .def("prompt_cache_hits", &WireFront::prompt_cache_hits);
 }
 
 }  // namespace wire

@@ -93,8 +93,12 @@ class WireIndexerService:
             flat[pos:pos + lens[i]] = np.asarray(t, dtype=np.int64)
             pos += lens[i]
             off[i + 1] = pos
-        scores = self.indexer.score_flat_tokens(
-            torch.from_numpy(flat), torch.from_numpy(off), model,
-            list(pods))
-        return scores.to("cpu", non_blocking=False).contiguous(), \
-            list(self.indexer.kv_block_index().registry.id_to_pod)
+        flat_t = torch.from_numpy(flat)
+        off_t = torch.from_numpy(off)
+        scores = self.indexer.score_flat_tokens(flat_t, off_t, model,
+                                                list(pods))
+        # 4-tuple: the C++ side feeds elements 2/3 (int32 tokens +
+        # offsets) into its prompt cache so repeats skip Python
+        return (scores.to("cpu", non_blocking=False).contiguous(),
+                list(self.indexer.kv_block_index().registry.id_to_pod),
+                flat_t.to(torch.int32), off_t)

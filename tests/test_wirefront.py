@@ -224,3 +224,21 @@ class TestWireFront:
         assert status == 200  # too short for a full block -> empty scores
         assert json.loads(rbody)["scores"] == {}
         s.close()
+
+    def test_prompt_cache_warm_hits(self, service):
+        """A repeated text prompt is served from the C++ prompt cache
+        (no Python tokenization) with identical scores."""
+        svc, port, tokens = service
+        prompt = " ".join(str(t) for t in tokens)
+        s = _connect(port)
+        bodies = []
+        rest = b""
+        for _ in range(3):
+            s.sendall(_http_post("/score",
+                                 {"model": MODEL, "prompt": prompt}))
+            status, body, rest = _read_response(s, rest)
+            assert status == 200
+            bodies.append(json.loads(body))
+        s.close()
+        assert bodies[0] == bodies[1] == bodies[2]
+        assert svc._front.prompt_cache_hits() >= 2
