@@ -55,6 +55,27 @@
 #include <torch/extension.h>
 
 namespace kvidx {
+
+// cpu_ops.cpp / hip_ops.hip (same extension)
+std::vector<at::Tensor> hash_chain_batch(at::Tensor, at::Tensor, at::Tensor,
+                                         int64_t);
+at::Tensor cpu_fused_score(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                           at::Tensor, at::Tensor, at::Tensor, int64_t,
+                           at::Tensor, at::Tensor, int64_t, at::Tensor,
+                           at::Tensor, int64_t, int64_t);
+#ifdef KVIDX_WITH_HIP
+at::Tensor gpu_fused_score(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                           at::Tensor, at::Tensor, at::Tensor, int64_t,
+                           at::Tensor, at::Tensor, int64_t, at::Tensor,
+                           at::Tensor, int64_t, int64_t, int64_t, int64_t);
+std::vector<at::Tensor> gpu_lookup(at::Tensor, at::Tensor, at::Tensor,
+                                   at::Tensor, at::Tensor, at::Tensor,
+                                   at::Tensor, int64_t, at::Tensor, int64_t,
+                                   at::Tensor, int64_t, int64_t, int64_t,
+                                   int64_t, int64_t);
+at::Tensor gpu_score_from_masks(at::Tensor, at::Tensor, at::Tensor, int64_t);
+#endif
+
 namespace wire {
 
 namespace py = pybind11;
@@ -442,10 +463,11 @@ static std::string http_response(int status, const char* status_text,
 class WireFront {
  public:
   WireFront(py::object score_tokens_cb, py::object score_text_cb,
-            int64_t max_batch)
+            int64_t max_batch, int64_t n_batchers)
       : score_tokens_cb_(std::move(score_tokens_cb)),
         score_text_cb_(std::move(score_text_cb)),
-        max_batch_(max_batch > 0 ? (size_t)max_batch : 4096) {}
+        max_batch_(max_batch > 0 ? (size_t)max_batch : 4096),
+        n_batchers_(n_batchers > 0 ? (int)n_batchers : 1) {}
 
   ~WireFront() { stop_nogil(); }
 
@@ -479,7 +501,11 @@ class WireFront {
     }
     for (int i = 0; i < n_io; ++i)
       io_threads_.emplace_back([this, i] { io_loop(i); });
-    batcher_ = std::thread([this] { batch_loop(); });
+    // Multiple batchers overlap: while one is inside the GIL-free
+    // scoring op (wire_score_flat-based callbacks release the GIL),
+    // another drains/parses the next micro-batch.
+    for (int b = 0; b < n_batchers_; ++b)
+      batchers_.emplace_back([this] { batch_loop(); });
     return bound_port_;
   }
 
@@ -504,7 +530,9 @@ class WireFront {
     }
     for (auto& t : io_threads_)
       if (t.joinable()) t.join();
-    if (batcher_.joinable()) batcher_.join();
+    for (auto& t : batchers_)
+      if (t.joinable()) t.join();
+    batchers_.clear();
     for (int fd : listeners_) close(fd);
     for (int efd : wakeups_) close(efd);
     listeners_.clear();
@@ -929,7 +957,8 @@ class WireFront {
   std::vector<int> listeners_;
   std::vector<int> wakeups_;
   std::vector<std::thread> io_threads_;
-  std::thread batcher_;
+  std::vector<std::thread> batchers_;
+  int n_batchers_ = 1;
 
   std::mutex queue_mu_;
   std::condition_variable queue_cv_;
@@ -943,11 +972,72 @@ class WireFront {
   PromptCache prompt_cache_{2048};
 };
 
+// One-call scoring path for wire batches: parallel CPU chain -> fused
+// probe/score (GPU or CPU table) -> CPU scores.  Registered with the
+// GIL RELEASED so a second batcher thread can overlap its own batch's
+// parse/launch with this one's kernel sync - and so io threads' Python
+// never stalls behind scoring.
+at::Tensor wire_score_flat(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
+                           at::Tensor pods, at::Tensor e_keys,
+                           at::Tensor e_meta, at::Tensor e_vals,
+                           int64_t pods_per_key, at::Tensor tokens_flat,
+                           at::Tensor offsets, int64_t model_id,
+                           at::Tensor filter_words, at::Tensor weights,
+                           int64_t num_pods, int64_t epoch,
+                           int64_t init_hash_bits, int64_t block_size,
+                           int64_t n_tiers) {
+  int64_t B = offsets.numel() - 1;
+  auto parents = at::full({B}, init_hash_bits, at::kLong);
+  auto chained = hash_chain_batch(tokens_flat, offsets, parents, block_size);
+  at::Tensor hashes = chained[0];
+  at::Tensor chunk_off = chained[1];  // int64 [B+1]
+  bool is_cuda = keys.is_cuda();
+  if (!is_cuda) {
+    auto counts = (chunk_off.slice(0, 1, B + 1) -
+                   chunk_off.slice(0, 0, B)).to(at::kInt);
+    return cpu_fused_score(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
+                           pods_per_key, hashes, counts, model_id,
+                           filter_words, weights, num_pods, epoch);
+  }
+#ifdef KVIDX_WITH_HIP
+  auto dev = keys.device();
+  auto h_dev = hashes.to(dev, /*non_blocking=*/false);
+  auto off_dev = chunk_off.to(at::kInt).to(dev);
+  int64_t max_k = 0;
+  {
+    const int64_t* co = chunk_off.data_ptr<int64_t>();
+    for (int64_t b = 0; b < B; ++b)
+      max_k = std::max(max_k, co[b + 1] - co[b]);
+  }
+  if (max_k == 0)
+    return at::zeros({B, num_pods}, at::kFloat);
+  int64_t W = (num_pods + 63) / 64;
+  at::Tensor scores;
+  if (max_k * n_tiers * W * 8 <= 64 * 1024) {
+    scores = gpu_fused_score(keys, meta, stamp, pods, e_keys, e_meta,
+                             e_vals, pods_per_key, h_dev, off_dev, model_id,
+                             filter_words, weights, num_pods, epoch, max_k,
+                             n_tiers);
+  } else {
+    auto fm = gpu_lookup(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
+                         pods_per_key, h_dev, model_id, filter_words,
+                         num_pods, epoch, 0, 1, n_tiers);
+    scores = gpu_score_from_masks(fm[1].contiguous(), off_dev, weights,
+                                  num_pods);
+  }
+  return scores.to(at::kCPU, /*non_blocking=*/false).contiguous();
+#else
+  TORCH_CHECK(false, "built without HIP but table is on GPU");
+#endif
+}
+
 void register_wirefront(py::module_& m) {
+  m.def("wire_score_flat", &wire_score_flat,
+        py::call_guard<py::gil_scoped_release>());
   py::class_<WireFront, std::shared_ptr<WireFront>>(m, "WireFront")
-      .def(py::init<py::object, py::object, int64_t>(),
+      .def(py::init<py::object, py::object, int64_t, int64_t>(),
            py::arg("score_tokens_cb"), py::arg("score_text_cb"),
-           py::arg("max_batch") = 4096)
+           py::arg("max_batch") = 4096, py::arg("n_batchers") = 2)
       .def("start", &WireFront::start, py::arg("host") = std::string(),
            py::arg("port") = 0, py::arg("n_io") = 2)
       .def("stop", &WireFront::stop)

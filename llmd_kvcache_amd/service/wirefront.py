@@ -37,6 +37,10 @@ class WireIndexerService:
         if not hasattr(ops, "WireFront"):
             raise RuntimeError(
                 "native extension built without the wirefront - rebuild")
+        if indexer.tokens_processor.config.hash_algo != "fnv-64a":
+            raise ValueError(
+                "wirefront requires the fnv-64a chain (the native "
+                "scoring op implements that algorithm)")
         self.indexer = indexer
         self._front = ops.WireFront(self._score_tokens_cb,
                                     self._score_text_cb, max_batch)
@@ -70,13 +74,34 @@ class WireIndexerService:
         self.stop()
         return False
 
-    # -- batch callbacks (invoked by the C++ batcher, GIL held) --------
+    # -- batch callbacks (invoked by the C++ batcher) ------------------
+    # The Python part of a batch is only id/filter/weight staging (~us);
+    # the heavy work runs inside ops.wire_score_flat, which RELEASES the
+    # GIL - so multiple C++ batcher threads overlap and io threads are
+    # never stalled behind scoring.
+    def _stage(self, model: str, pods: Sequence[str]):
+        idx = self.indexer.kv_block_index()
+        tp = self.indexer.tokens_processor
+        model_id = idx.registry.model_id(model)
+        num_pods = idx._num_pods_padded()
+        filt = idx._filter_tensor(set(pods), num_pods)
+        weights = idx.tier_weights(
+            {b.name: b.weight
+             for b in self.indexer.config.backend_configs})
+        n_tiers = max(1, len(idx.registry.id_to_tier))
+        from ..kvblock.gpu_index import _to_i64
+
+        return (idx, model_id, filt, weights, num_pods, n_tiers,
+                _to_i64(tp.config.init_hash()), tp.block_size)
+
     def _score_tokens_cb(self, model: str, pods: Sequence[str],
                          tokens_flat, offsets):
-        scores = self.indexer.score_flat_tokens(
-            tokens_flat, offsets, model, list(pods))
-        return scores.to("cpu", non_blocking=False).contiguous(), \
-            list(self.indexer.kv_block_index().registry.id_to_pod)
+        (idx, model_id, filt, weights, num_pods, n_tiers, init,
+         bs) = self._stage(model, pods)
+        scores = idx.table.ops.wire_score_flat(
+            *idx.table._t(), tokens_flat, offsets, model_id, filt,
+            weights, num_pods, idx.table.next_epoch(), init, bs, n_tiers)
+        return scores, list(idx.registry.id_to_pod)
 
     def _score_text_cb(self, model: str, pods: Sequence[str],
                        prompts: List[str]):
@@ -95,10 +120,12 @@ class WireIndexerService:
             off[i + 1] = pos
         flat_t = torch.from_numpy(flat)
         off_t = torch.from_numpy(off)
-        scores = self.indexer.score_flat_tokens(flat_t, off_t, model,
-                                                list(pods))
+        (idx, model_id, filt, weights, num_pods, n_tiers, init,
+         bs) = self._stage(model, pods)
+        scores = idx.table.ops.wire_score_flat(
+            *idx.table._t(), flat_t, off_t, model_id, filt, weights,
+            num_pods, idx.table.next_epoch(), init, bs, n_tiers)
         # 4-tuple: the C++ side feeds elements 2/3 (int32 tokens +
         # offsets) into its prompt cache so repeats skip Python
-        return (scores.to("cpu", non_blocking=False).contiguous(),
-                list(self.indexer.kv_block_index().registry.id_to_pod),
+        return (scores, list(idx.registry.id_to_pod),
                 flat_t.to(torch.int32), off_t)
