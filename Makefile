@@ -31,3 +31,13 @@ lint:
 
 clean:
 	rm -rf build llmd_kvcache_amd/ops/_kvidx_C*.so llmd_kvcache_amd/ops/csrc/*_hip.hip
+
+IMAGE ?= llmd-kvcache-amd:latest
+BASE_IMAGE ?= rocm/pytorch:latest
+
+.PHONY: image container-smoke
+image:
+	docker build -t $(IMAGE) --build-arg BASE=$(BASE_IMAGE) .
+
+container-smoke:
+	BASE=$(BASE_IMAGE) IMG=$(IMAGE) bash scripts/container_smoke.sh

@@ -190,7 +190,7 @@ def measure_wire(indexer, chains, prefix_frac, seconds_per_mode=4.0):
     from llmd_kvcache_amd.service.wirefront import WireIndexerService
 
     svc = WireIndexerService(indexer, max_batch=8192)
-    port = svc.start(port=0, n_io=6)
+    port = svc.start(port=0, n_io=8)
     out = {}
     try:
         rng = np.random.default_rng(777)
@@ -238,7 +238,7 @@ def measure_wire(indexer, chains, prefix_frac, seconds_per_mode=4.0):
                 rest += s.recv(1 << 20)
             return rest[clen:]
 
-        def measure_mode(mode, n_procs=6, depth=64):
+        def measure_mode(mode, n_procs=8, depth=96):
             import multiprocessing as _mp
             import statistics as _stats
 

@@ -301,18 +301,36 @@ struct PromptKey {
 struct PromptKeyHash {
   size_t operator()(const PromptKey& k) const { return (size_t)(k.a ^ k.b); }
 };
+static inline void fp_mix_word(uint64_t w, uint64_t& h1, uint64_t& h2) {
+  h1 = (h1 ^ w) * 0x9E3779B185EBCA87ull;
+  h1 ^= h1 >> 31;
+  h2 = (h2 ^ (w * 0xC2B2AE3D27D4EB4Full)) * 0x165667B19E3779F9ull;
+  h2 ^= h2 >> 29;
+}
+static void fp_mix_bytes(const char* p, size_t n, uint64_t& h1,
+                         uint64_t& h2) {
+  // word-at-a-time (a byte loop costs ~3 ns/B, which at 45 KB prompts
+  // was ~130 us/request - the text-mode wire bottleneck)
+  size_t i = 0;
+  for (; i + 8 <= n; i += 8) {
+    uint64_t w;
+    memcpy(&w, p + i, 8);
+    fp_mix_word(w, h1, h2);
+  }
+  if (i < n) {
+    uint64_t w = 0;
+    memcpy(&w, p + i, n - i);
+    fp_mix_word(w | ((uint64_t)(n - i) << 56), h1, h2);
+  }
+}
 static PromptKey prompt_fingerprint(const std::string& model,
                                     const std::string& prompt) {
   uint64_t h1 = 14695981039346656037ull;
   uint64_t h2 = 0x9e3779b97f4a7c15ull;
-  auto mix = [&](unsigned char c) {
-    h1 = (h1 ^ c) * 1099511628211ull;
-    h2 = (h2 ^ c) * 0xff51afd7ed558ccdull;
-    h2 ^= h2 >> 29;
-  };
-  for (unsigned char c : model) mix(c);
-  mix(0);
-  for (unsigned char c : prompt) mix(c);
+  fp_mix_bytes(model.data(), model.size(), h1, h2);
+  fp_mix_word(0xFFull ^ ((uint64_t)model.size() << 8), h1, h2);
+  fp_mix_bytes(prompt.data(), prompt.size(), h1, h2);
+  fp_mix_word((uint64_t)prompt.size(), h1, h2);
   return {h1, h2};
 }
 

@@ -94,3 +94,39 @@ def test_sharded_service_main_importable():
     mod = importlib.util.module_from_spec(spec)
     spec.loader.exec_module(mod)
     assert callable(mod.main)
+
+
+class TestPackaging:
+    """Deployment packaging parity (reference Dockerfile + deploy/)."""
+
+    ROOT = __import__("os").path.dirname(__import__("os").path.dirname(
+        __import__("os").path.abspath(__file__)))
+
+    def test_dockerfile_entrypoint_exists(self):
+        import os
+
+        df = os.path.join(self.ROOT, "Dockerfile")
+        assert os.path.exists(df)
+        content = open(df).read()
+        assert "examples/online_service.py" in content
+        assert os.path.exists(os.path.join(
+            self.ROOT, "examples", "online_service.py"))
+        assert "gfx950" in content  # builds the native extension
+
+    def test_deploy_manifests_parse(self):
+        import os
+
+        import yaml
+
+        for name in ("kustomization.yaml", "deployment.yaml",
+                     "service.yaml"):
+            path = os.path.join(self.ROOT, "deploy", name)
+            with open(path) as f:
+                doc = yaml.safe_load(f)
+            assert isinstance(doc, dict), name
+        dep = yaml.safe_load(open(os.path.join(self.ROOT, "deploy",
+                                               "deployment.yaml")))
+        container = dep["spec"]["template"]["spec"]["containers"][0]
+        ports = {p["name"]: p["containerPort"] for p in container["ports"]}
+        assert ports == {"http": 8080, "kvevents": 5557}
+        assert container["livenessProbe"]["httpGet"]["path"] == "/health"
