@@ -72,12 +72,50 @@ class TokenizerCore:
             if path:
                 tok = hf_tokenizers.Tokenizer.from_file(path)
         if tok is None:
-            tok = hf_tokenizers.Tokenizer.from_pretrained(model)
+            try:
+                tok = hf_tokenizers.Tokenizer.from_pretrained(model)
+            except Exception as hf_err:
+                # ModelScope fallback, parity with the reference sidecar
+                # (tokenizer_service/tokenizer.py:47-207): HF-unreachable
+                # deployments can still pull the tokenizer.json.
+                tok = self._load_modelscope(model, hf_tokenizers, hf_err)
         with self._lock:
             if len(self._cache) >= int(self.config.get("cache_size", 8)):
                 self._cache.pop(next(iter(self._cache)))
             self._cache[model] = tok
         return tok
+
+    @staticmethod
+    def _load_modelscope(model: str, hf_tokenizers, hf_err: Exception):
+        try:
+            from modelscope.hub.snapshot_download import snapshot_download
+        except ImportError:
+            raise RuntimeError(
+                f"HF load failed ({hf_err}) and modelscope is not "
+                "installed for fallback"
+            ) from hf_err
+        path = snapshot_download(model, allow_patterns=["tokenizer.json"])
+        return hf_tokenizers.Tokenizer.from_file(
+            os.path.join(path, "tokenizer.json"))
+
+    def preload(self, models: list, lock_path: str) -> None:
+        """Flock-guarded warmup: with N prefork workers only ONE at a
+        time loads/downloads each model into the shared HF cache - the
+        same serialization the reference's gunicorn.conf.py:20-21 does
+        around its pre-download step."""
+        import fcntl
+
+        with open(lock_path, "a+") as lf:
+            fcntl.flock(lf, fcntl.LOCK_EX)
+            try:
+                for m in models:
+                    try:
+                        self._load(m)
+                        logger.info("preloaded tokenizer %s", m)
+                    except Exception as e:
+                        logger.error("preload of %s failed: %s", m, e)
+            finally:
+                fcntl.flock(lf, fcntl.LOCK_UN)
 
     def tokenize(
         self, prompt: str, model: str, add_special_tokens: Optional[bool]
@@ -181,33 +219,123 @@ def make_app() -> web.Application:
     return app
 
 
-def main() -> None:
-    ap = argparse.ArgumentParser()
-    ap.add_argument("--socket", default=os.environ.get("UDS_SOCKET", DEFAULT_SOCKET))
-    ap.add_argument("--probe-port", type=int,
-                    default=int(os.environ.get("PROBE_PORT", "0")))
-    args = ap.parse_args()
-
-    logging.basicConfig(level=os.environ.get("LOG_LEVEL", "INFO"))
-    os.makedirs(os.path.dirname(args.socket), exist_ok=True)
-    if os.path.exists(args.socket):
-        os.unlink(args.socket)
-
+def _serve_on_socket(sock, probe_port: int, preload: list,
+                     lock_path: str) -> None:
+    """One worker: aiohttp on an inherited, already-bound UDS listener.
+    Accept races across workers are resolved by the kernel."""
     app = make_app()
-    if args.probe_port:
-        # TCP liveness probe alongside the UDS endpoint (server.py:293)
+    if preload:
+        core.preload(preload, lock_path)
+    if probe_port:
+        # TCP liveness probe alongside the UDS endpoint (server.py:293);
+        # port 0 + SO_REUSEPORT keeps N workers from colliding
         probe = web.Application()
         probe.router.add_get("/health", handle_health)
 
         async def start_probe(_app):
             runner = web.AppRunner(probe)
             await runner.setup()
-            site = web.TCPSite(runner, "0.0.0.0", args.probe_port)
+            site = web.TCPSite(runner, "0.0.0.0", probe_port,
+                               reuse_port=True)
             await site.start()
 
         app.on_startup.append(start_probe)
 
-    web.run_app(app, path=args.socket)
+    async def run():
+        runner = web.AppRunner(app)
+        await runner.setup()
+        site = web.SockSite(runner, sock)
+        await site.start()
+        while True:
+            await asyncio.sleep(3600)
+
+    try:
+        asyncio.run(run())
+    except KeyboardInterrupt:  # pragma: no cover
+        pass
+
+
+def main() -> None:
+    import signal
+    import socket as socket_mod
+    import sys
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--socket", default=os.environ.get("UDS_SOCKET", DEFAULT_SOCKET))
+    ap.add_argument("--probe-port", type=int,
+                    default=int(os.environ.get("PROBE_PORT", "0")))
+    ap.add_argument("--workers", type=int,
+                    default=int(os.environ.get("UDS_WORKERS", "1")),
+                    help="prefork worker processes sharing the socket "
+                         "(reference runs gunicorn multi-worker, "
+                         "gunicorn.conf.py)")
+    ap.add_argument("--preload-model", action="append", default=None,
+                    help="tokenizer(s) to load before serving; guarded "
+                         "by an flock so N workers initialize serially")
+    args = ap.parse_args()
+
+    logging.basicConfig(level=os.environ.get("LOG_LEVEL", "INFO"))
+    os.makedirs(os.path.dirname(args.socket) or ".", exist_ok=True)
+    if os.path.exists(args.socket):
+        os.unlink(args.socket)
+
+    sock = socket_mod.socket(socket_mod.AF_UNIX, socket_mod.SOCK_STREAM)
+    sock.bind(args.socket)
+    sock.listen(1024)
+    preload = args.preload_model or []
+    lock_path = args.socket + ".init.lock"
+
+    if args.workers <= 1:
+        _serve_on_socket(sock, args.probe_port, preload, lock_path)
+        return
+
+    # prefork supervisor: N workers accept on the shared listener; a
+    # crashed worker is restarted (failure-recovery parity with a
+    # gunicorn master), SIGTERM/SIGINT tears the set down.
+    children: Dict[int, bool] = {}
+    shutting_down = {"v": False}
+
+    def spawn() -> int:
+        pid = os.fork()
+        if pid == 0:  # child
+            # drop the supervisor's handlers: a worker must die on
+            # SIGTERM, not try to signal its siblings
+            signal.signal(signal.SIGTERM, signal.SIG_DFL)
+            signal.signal(signal.SIGINT, signal.SIG_DFL)
+            try:
+                _serve_on_socket(sock, args.probe_port, preload, lock_path)
+            finally:
+                os._exit(0)
+        return pid
+
+    def on_term(signum, frame):  # pragma: no cover - signal path
+        shutting_down["v"] = True
+        for pid in list(children):
+            try:
+                os.kill(pid, signal.SIGTERM)
+            except ProcessLookupError:
+                pass
+
+    signal.signal(signal.SIGTERM, on_term)
+    signal.signal(signal.SIGINT, on_term)
+
+    for _ in range(args.workers):
+        children[spawn()] = True
+    logger.info("uds_tokenizer: %d workers on %s", args.workers,
+                args.socket)
+    while children:
+        try:
+            pid, status = os.wait()
+        except ChildProcessError:  # pragma: no cover
+            break
+        except InterruptedError:
+            continue
+        children.pop(pid, None)
+        if not shutting_down["v"]:
+            logger.warning("worker %d exited (status %d); restarting",
+                           pid, status)
+            children[spawn()] = True
+    sys.exit(0)
 
 
 if __name__ == "__main__":

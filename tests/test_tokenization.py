@@ -433,3 +433,125 @@ class TestInlineModeConcurrency:
         assert all(not t.is_alive() for t in threads)  # nobody deadlocked
         for i in range(8):
             assert results[i] == [i + 1] * 3
+
+
+class TestUdsMultiWorker:
+    """Prefork multi-worker sidecar (reference runs gunicorn N workers
+    with flock-guarded init; here: shared-listener prefork supervisor
+    with worker restart)."""
+
+    def _spawn(self, sock_path, tokenizer_dir, workers=2):
+        import subprocess
+        import sys
+        import time as _time
+
+        root = os.path.join(os.path.dirname(__file__), "..")
+        env = dict(os.environ, LOCAL_TOKENIZER_DIR=tokenizer_dir,
+                   PYTHONPATH=os.path.abspath(root))
+        proc = subprocess.Popen(
+            [sys.executable,
+             os.path.join(root, "services", "uds_tokenizer", "server.py"),
+             "--socket", sock_path, "--workers", str(workers)],
+            env=env, stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+        for _ in range(100):
+            if os.path.exists(sock_path):
+                break
+            _time.sleep(0.1)
+        else:
+            proc.terminate()
+            raise RuntimeError("sidecar did not come up")
+        return proc
+
+    def test_prefork_serves_and_restarts(self, tokenizer_fixture_dir,
+                                         tmp_path):
+        import signal
+        import subprocess
+        import threading
+        import time as _time
+
+        from llmd_kvcache_amd.tokenization.uds import (UdsTokenizer,
+                                                       UdsTokenizerConfig)
+
+        sock_path = str(tmp_path / "mw.socket")
+        proc = self._spawn(sock_path, tokenizer_fixture_dir, workers=2)
+        try:
+            client = UdsTokenizer(UdsTokenizerConfig(
+                socket_path=sock_path, timeout_s=10.0))
+            results, errs = [], []
+
+            def call():
+                try:
+                    ids, offs = client.encode("hello world", "test-model")
+                    results.append(ids)
+                except Exception as e:  # pragma: no cover
+                    errs.append(e)
+
+            threads = [threading.Thread(target=call) for _ in range(8)]
+            for t in threads:
+                t.start()
+            for t in threads:
+                t.join(timeout=30)
+            assert not errs
+            assert len(results) == 8 and all(len(r) == 2 for r in results)
+
+            # kill one worker: the supervisor must restart it and the
+            # service keeps answering
+            out = subprocess.run(
+                ["pgrep", "-P", str(proc.pid)], capture_output=True,
+                text=True)
+            kids = [int(x) for x in out.stdout.split()]
+            assert len(kids) == 2
+            os.kill(kids[0], signal.SIGKILL)
+            deadline = _time.monotonic() + 15
+            while _time.monotonic() < deadline:
+                out = subprocess.run(["pgrep", "-P", str(proc.pid)],
+                                     capture_output=True, text=True)
+                now = [int(x) for x in out.stdout.split()]
+                if len(now) == 2 and set(now) != set(kids):
+                    break
+                _time.sleep(0.2)
+            ids, _ = client.encode("hello world", "test-model")
+            assert len(ids) == 2
+        finally:
+            proc.terminate()
+            proc.wait(timeout=10)
+
+    def test_flock_preload_serializes(self, tokenizer_fixture_dir,
+                                      tmp_path):
+        """Two cores preloading under the same lock never overlap."""
+        import threading
+        import time as _time
+
+        sys_path_root = os.path.join(os.path.dirname(__file__), "..")
+        import sys
+
+        sys.path.insert(0, sys_path_root)
+        from services.uds_tokenizer.server import TokenizerCore
+
+        lock_path = str(tmp_path / "init.lock")
+        active = []
+        overlaps = []
+
+        class SlowCore(TokenizerCore):
+            def _load(self, model):
+                active.append(1)
+                if len(active) - len(overlaps) > 1:  # pragma: no cover
+                    overlaps.append(1)
+                _time.sleep(0.2)
+                active.pop()
+                return object()
+
+        cores = [SlowCore({"local_dir": tokenizer_fixture_dir})
+                 for _ in range(3)]
+        threads = [
+            threading.Thread(target=c.preload,
+                             args=(["test-model"], lock_path))
+            for c in cores
+        ]
+        t0 = _time.monotonic()
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=10)
+        assert not overlaps
+        assert _time.monotonic() - t0 >= 0.55  # serialized, not parallel
