@@ -80,6 +80,12 @@ void gpu_apply_events_split(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                             at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                             at::Tensor, at::Tensor, int64_t, int64_t, int64_t,
                             int64_t, int64_t, int64_t);
+void gpu_apply_events_split_tr(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                            at::Tensor, at::Tensor, at::Tensor, int64_t,
+                            at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                            at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                            at::Tensor, int64_t, int64_t, int64_t,
+                            int64_t, int64_t, int64_t);
 #endif
 
 namespace wire {
@@ -116,6 +122,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("row_major") = 0);
   m.def("gpu_apply_events", &kvidx::gpu_apply_events);
   m.def("gpu_apply_events_split", &kvidx::gpu_apply_events_split);
+  m.def("gpu_apply_events_split_tr", &kvidx::gpu_apply_events_split_tr);
 #else
   m.attr("HAS_HIP") = false;
 #endif

@@ -734,6 +734,43 @@ __global__ void k_event_chains(
   }
 }
 
+// Phase A variant for batches with NO batch-local parent references
+// (the host checks parents against the batch's engine hashes): chains
+// are then independent across EVENTS, not just pod groups, so one LANE
+// per event runs them - 8x the parallelism of lane-per-group at the
+// bench shape - and the tokens are staged TRANSPOSED [token_pos][event]
+// in int32 like the read path's chain kernel, so each position is one
+// coalesced 256 B line across the wave instead of 64 scattered 8 B
+// reads (ROADMAP round-1 #2 follow-up; write path was 9.9M blocks/s
+// with the serial chain phase dominating).
+__global__ void k_event_chains_ev(
+    DevTable v, const int32_t* __restrict__ tokens_t,  // [maxT, E]
+    const int32_t* __restrict__ tok_off, const uint64_t* __restrict__ ehashes,
+    const int32_t* __restrict__ eh_off, const uint64_t* __restrict__ parents,
+    const uint8_t* __restrict__ has_parent, const uint8_t* __restrict__ ev_type,
+    int64_t E, uint32_t model, uint64_t init_hash, int block_size,
+    uint64_t* __restrict__ req_scratch) {
+  const int64_t e = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (e >= E) return;
+  if (ev_type[e] == 1) return;  // removals have no chain
+  const int nh = eh_off[e + 1] - eh_off[e];
+  const int n_chunks = (tok_off[e + 1] - tok_off[e]) / block_size;
+  if (n_chunks != nh) return;  // drop mismatched event (see k_apply_events)
+  uint64_t parent = init_hash;
+  if (has_parent[e]) {
+    int64_t ei = dev_emap_find(v, parents[e], model);
+    if (ei >= 0) parent = v.e_vals[ei];
+  }
+  uint64_t h = parent;
+  uint32_t tok[64];  // block_size <= 64
+  for (int c = 0; c < n_chunks; ++c) {
+    for (int j = 0; j < block_size; ++j)
+      tok[j] = (uint32_t)tokens_t[(int64_t)(c * block_size + j) * E + e];
+    h = chunk_hash_fast(h, tok, block_size);
+    req_scratch[eh_off[e] + c] = h;
+  }
+}
+
 // Phase B: one thread per block applies its insert/removal.
 __global__ void k_event_inserts(
     DevTable v, const uint64_t* __restrict__ ehashes,
@@ -1086,6 +1123,53 @@ void gpu_apply_events(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
       (uint64_t)init_hash_bits, (int)block_size, (int32_t)epoch,
       (int)shard_id, (int)num_shards,
       reinterpret_cast<uint64_t*>(req_scratch.data_ptr<int64_t>()));
+}
+
+// Transposed-chain phase-split apply: NO batch-local parents (host
+// verified), so chains run one lane per EVENT from transposed int32
+// tokens; inserts thread-per-block as usual.  No bmap is needed -
+// parents resolve via the persistent engine map or the chain root.
+void gpu_apply_events_split_tr(
+    at::Tensor keys, at::Tensor meta, at::Tensor stamp, at::Tensor pods,
+    at::Tensor e_keys, at::Tensor e_meta, at::Tensor e_vals,
+    int64_t pods_per_key, at::Tensor tokens_t, at::Tensor tok_off,
+    at::Tensor ehashes, at::Tensor eh_off, at::Tensor parents,
+    at::Tensor has_parent, at::Tensor ev_type, at::Tensor pod_entry,
+    at::Tensor ev_of, int64_t model_id, int64_t init_hash_bits,
+    int64_t block_size, int64_t epoch, int64_t shard_id,
+    int64_t num_shards) {
+  auto v = dev_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
+                    pods_per_key);
+  TORCH_CHECK(tokens_t.dtype() == at::kInt && tokens_t.dim() == 2,
+              "tokens_t must be int32 [maxT, E]");
+  TORCH_CHECK(block_size <= 64, "event chain kernel supports block_size<=64");
+  int64_t E = ev_type.numel();
+  int64_t n = ehashes.numel();
+  if (E == 0 || n == 0) return;
+  TORCH_CHECK(tokens_t.size(1) == E, "tokens_t second dim must be E");
+  auto req_scratch = at::zeros({n}, ehashes.options());
+  int threads = 256;
+  hipLaunchKernelGGL(k_event_chains_ev,
+                     dim3((int)((E + threads - 1) / threads)), dim3(threads),
+                     0, STREAM, v, tokens_t.data_ptr<int32_t>(),
+                     tok_off.data_ptr<int32_t>(), U64P(ehashes),
+                     eh_off.data_ptr<int32_t>(), U64P(parents),
+                     has_parent.data_ptr<uint8_t>(),
+                     ev_type.data_ptr<uint8_t>(), E, (uint32_t)model_id,
+                     (uint64_t)init_hash_bits, (int)block_size,
+                     reinterpret_cast<uint64_t*>(
+                         req_scratch.data_ptr<int64_t>()));
+  hipLaunchKernelGGL(k_event_inserts,
+                     dim3((int)((n + threads - 1) / threads)), dim3(threads),
+                     0, STREAM, v, U64P(ehashes), eh_off.data_ptr<int32_t>(),
+                     ev_of.data_ptr<int32_t>(), ev_type.data_ptr<uint8_t>(),
+                     tok_off.data_ptr<int32_t>(),
+                     reinterpret_cast<const uint32_t*>(
+                         pod_entry.data_ptr<int32_t>()),
+                     n, (uint32_t)model_id, (int)block_size, (int32_t)epoch,
+                     (int)shard_id, (int)num_shards,
+                     reinterpret_cast<uint64_t*>(
+                         req_scratch.data_ptr<int64_t>()));
 }
 
 // Phase-split apply: bmap build -> chains (lane per group) -> inserts
