@@ -85,7 +85,7 @@ void gpu_apply_events_split_tr(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                             at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                             at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                             at::Tensor, int64_t, int64_t, int64_t,
-                            int64_t, int64_t, int64_t);
+                            int64_t, int64_t, int64_t, int64_t);
 #endif
 
 namespace wire {

@@ -743,6 +743,21 @@ __global__ void k_event_chains(
 // coalesced 256 B line across the wave instead of 64 scattered 8 B
 // reads (ROADMAP round-1 #2 follow-up; write path was 9.9M blocks/s
 // with the serial chain phase dominating).
+// Device-side staging for k_event_chains_ev: flat per-event int32
+// tokens -> transposed [token_pos][event].  Host-side transpose of the
+// ragged batch measured ~2-5 ms/batch in numpy (it briefly halved
+// ingest); here it is one trivial bandwidth-bound kernel over ~2 MB.
+__global__ void k_transpose_ev_tokens(const int32_t* __restrict__ flat,
+                                      const int32_t* __restrict__ tok_off,
+                                      int64_t E,
+                                      int32_t* __restrict__ out) {  // [maxT,E]
+  const int e = blockIdx.x;
+  const int len = tok_off[e + 1] - tok_off[e];
+  for (int i = blockIdx.y * blockDim.x + threadIdx.x; i < len;
+       i += gridDim.y * blockDim.x)
+    out[(int64_t)i * E + e] = flat[tok_off[e] + i];
+}
+
 __global__ void k_event_chains_ev(
     DevTable v, const int32_t* __restrict__ tokens_t,  // [maxT, E]
     const int32_t* __restrict__ tok_off, const uint64_t* __restrict__ ehashes,
@@ -1132,23 +1147,34 @@ void gpu_apply_events(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
 void gpu_apply_events_split_tr(
     at::Tensor keys, at::Tensor meta, at::Tensor stamp, at::Tensor pods,
     at::Tensor e_keys, at::Tensor e_meta, at::Tensor e_vals,
-    int64_t pods_per_key, at::Tensor tokens_t, at::Tensor tok_off,
+    int64_t pods_per_key, at::Tensor tokens_flat, at::Tensor tok_off,
     at::Tensor ehashes, at::Tensor eh_off, at::Tensor parents,
     at::Tensor has_parent, at::Tensor ev_type, at::Tensor pod_entry,
     at::Tensor ev_of, int64_t model_id, int64_t init_hash_bits,
     int64_t block_size, int64_t epoch, int64_t shard_id,
-    int64_t num_shards) {
+    int64_t num_shards, int64_t max_tokens) {
   auto v = dev_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
                     pods_per_key);
-  TORCH_CHECK(tokens_t.dtype() == at::kInt && tokens_t.dim() == 2,
-              "tokens_t must be int32 [maxT, E]");
+  TORCH_CHECK(tokens_flat.dtype() == at::kInt,
+              "event tokens must be int32 (uint32 bit pattern)");
   TORCH_CHECK(block_size <= 64, "event chain kernel supports block_size<=64");
   int64_t E = ev_type.numel();
   int64_t n = ehashes.numel();
   if (E == 0 || n == 0) return;
-  TORCH_CHECK(tokens_t.size(1) == E, "tokens_t second dim must be E");
+  TORCH_CHECK(max_tokens > 0, "max_tokens must be positive");
+  auto tokens_t = at::empty({max_tokens, E},
+                            tokens_flat.options());
   auto req_scratch = at::zeros({n}, ehashes.options());
   int threads = 256;
+  {
+    int ygrid = (int)std::min<int64_t>((max_tokens + threads - 1) / threads,
+                                       64);
+    hipLaunchKernelGGL(k_transpose_ev_tokens, dim3((int)E, ygrid),
+                       dim3(threads), 0, STREAM,
+                       tokens_flat.data_ptr<int32_t>(),
+                       tok_off.data_ptr<int32_t>(), E,
+                       tokens_t.data_ptr<int32_t>());
+  }
   hipLaunchKernelGGL(k_event_chains_ev,
                      dim3((int)((E + threads - 1) / threads)), dim3(threads),
                      0, STREAM, v, tokens_t.data_ptr<int32_t>(),
