@@ -276,6 +276,46 @@ class TestApplyEventsKernel:
             assert set(c.get(rk, [])) == set(g.get(rk, [])), rk
 
 
+    def test_transposed_event_path_ragged_matches_cpu(self):
+        """Lane-per-event transposed chain path (parentless batch,
+        RAGGED event sizes - exercises the on-device transpose padding)
+        produces the same state as the CPU digest."""
+        from llmd_kvcache_amd.kvevents.pool import digest_events
+
+        rng = random.Random(21)
+        tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=4))
+        cpu_idx = NativeIndex(TableIndexConfig(capacity=1 << 12,
+                                               pods_per_key=10))
+        gpu_idx = GpuIndex(GpuIndexConfig(capacity=1 << 12, pods_per_key=10))
+
+        batches = []
+        next_hash = 5000
+        for i in range(40):
+            pod = f"pod-{rng.randrange(6)}"
+            n_blocks = rng.randrange(1, 9)  # ragged: 4..32 tokens
+            toks = [rng.randrange(0, 1 << 31) for _ in range(n_blocks * 4)]
+            hs = list(range(next_hash, next_hash + n_blocks))
+            next_hash += n_blocks
+            batches.append((pod, MODEL, [BlockStored(hs, None, toks, 4)]))
+
+        for pod, model, events in batches:
+            digest_events(cpu_idx, tp, pod, model, events)
+        gpu_idx.apply_event_batches(batches, tp)
+        torch.cuda.synchronize()
+
+        for pod, model, events in batches:
+            ev = events[0]
+            req = tp.tokens_to_kv_block_keys(None, ev.token_ids, model)
+            c = cpu_idx.lookup(req, set())
+            g = gpu_idx.lookup(req, set())
+            assert {k: set(v) for k, v in c.items()} == \
+                {k: set(v) for k, v in g.items()}, (pod, ev.block_hashes)
+            for h in ev.block_hashes:
+                ck = Key(model, h)
+                assert (cpu_idx.get_request_key(ck) ==
+                        gpu_idx.get_request_key(ck)), ck
+
+
 class TestConcurrentGpuInserts:
     def test_many_duplicate_inserts_converge(self):
         """Thousands of threads inserting the same keys concurrently must
