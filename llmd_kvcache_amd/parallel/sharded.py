@@ -74,6 +74,12 @@ class ShardedIndex:
         self.local = TableIndex(cfg)
         self.device = self.local.device
         self.registry = self.local.registry
+        # Testing hook: force the reduce_scatter merge strategy even on
+        # small payloads / non-NCCL backends (gloo lacks
+        # reduce_scatter_tensor; _merge_reduce_scatter then emulates it
+        # with all_reduce+slice so the partition/flag/gather logic is
+        # CPU-testable).  Must be set identically on every rank.
+        self.force_reduce_scatter = False
 
     # -- write path ----------------------------------------------------
     def add(self, engine_keys, request_keys, entries) -> None:
@@ -135,11 +141,10 @@ class ShardedIndex:
 
         offs_cpu = offsets.to(dtype=torch.int32, device="cpu")
         payload = K * n_tiers * W * 8
-        use_rs = (
-            self.world_size > 1
-            and B >= self.world_size
-            and payload >= REDUCE_SCATTER_MIN_BYTES
-            and dist.get_backend(self.group) == "nccl"
+        use_rs = self.world_size > 1 and B >= self.world_size and (
+            self.force_reduce_scatter
+            or (payload >= REDUCE_SCATTER_MIN_BYTES
+                and dist.get_backend(self.group) == "nccl")
         )
         if use_rs:
             scores = self._merge_reduce_scatter(
@@ -207,8 +212,14 @@ class ShardedIndex:
                     bview[r, : (k_hi - k_lo) * TW].copy_(
                         flat[k_lo:k_hi].reshape(-1))
         out = torch.empty(seg, dtype=torch.int64, device=dev)
-        dist.reduce_scatter_tensor(out, buf, op=dist.ReduceOp.SUM,
-                                   group=self.group)
+        if dist.get_backend(self.group) == "nccl":
+            dist.reduce_scatter_tensor(out, buf, op=dist.ReduceOp.SUM,
+                                       group=self.group)
+        else:
+            # gloo has no reduce_scatter_tensor: emulate (same result,
+            # more traffic) so CPU tests exercise this exact code path
+            dist.all_reduce(buf, op=dist.ReduceOp.SUM, group=self.group)
+            out.copy_(buf.view(self.world_size, seg)[self.rank])
 
         # score rows for this rank's prompt group (+1 flag row)
         P = num_pods

@@ -348,3 +348,106 @@ def test_follower_failure_keeps_collective_alive():
     assert results[1] == "served"
     assert outcomes == ["raised", "raised"]  # explicit error, not bad data
     assert recovered == healthy              # follower recovered fully
+
+
+def _body_reduce_scatter_differential(rank, world_size):
+    """Forces the reduce_scatter merge strategy (prompt partitioning,
+    error-flag rows, score all_gather reassembly) and differentials it
+    against single-process scoring - same checks as the all_reduce path
+    test, covering the large-batch RCCL strategy's logic on CPU."""
+    from llmd_kvcache_amd.kvblock.gpu_index import (
+        NativeIndex,
+        TableIndexConfig,
+        _to_i64,
+    )
+    from llmd_kvcache_amd.kvblock.keys import Key, PodEntry
+    from llmd_kvcache_amd.parallel.sharded import ShardedIndex
+    from llmd_kvcache_amd.scorer import new_kv_block_scorer
+
+    rng = random.Random(55)
+    sharded = ShardedIndex(TableIndexConfig(capacity=1 << 12, pods_per_key=10))
+    sharded.force_reduce_scatter = True
+    single = NativeIndex(TableIndexConfig(capacity=1 << 12, pods_per_key=10))
+    scorer = new_kv_block_scorer()
+
+    for _ in range(80):
+        start = rng.randrange(50)
+        n = rng.randrange(1, 8)
+        keys = [Key(MODEL, 7000 + start + i) for i in range(n)]
+        pod = f"pod-{rng.randrange(8)}"
+        tier = rng.choice(["gpu", "cpu"])
+        sharded.add(keys, keys, [PodEntry(pod, tier)])
+        single.add(keys, keys, [PodEntry(pod, tier)])
+
+    mismatches = []
+    for trial in range(10):
+        prompts = []
+        for _ in range(rng.randrange(world_size, world_size * 3 + 1)):
+            start = rng.randrange(50)
+            n = rng.randrange(1, 10)
+            prompts.append([Key(MODEL, 7000 + start + i) for i in range(n)])
+        flat = [_to_i64(k.chunk_hash) for p in prompts for k in p]
+        offsets = [0]
+        for p in prompts:
+            offsets.append(offsets[-1] + len(p))
+        scores = sharded.sharded_scores(
+            torch.tensor(flat, dtype=torch.int64),
+            torch.tensor(offsets, dtype=torch.int32), MODEL, set())
+        maps = sharded.local.scores_to_map(scores)
+        for p, got in zip(prompts, maps):
+            expected = {
+                pod: s for pod, s in
+                scorer.score(p, single.lookup(p, set())).items() if s != 0
+            }
+            if set(got) != set(expected) or any(
+                abs(got[k] - expected[k]) > 1e-4 for k in expected
+            ):
+                mismatches.append((trial, got, expected))
+    return mismatches
+
+
+def test_reduce_scatter_merge_matches_single_process():
+    for ws in (2, 4):
+        results = run_distributed("_body_reduce_scatter_differential",
+                                  world_size=ws)
+        for rank, mismatches in results.items():
+            assert mismatches == [], f"ws={ws} rank {rank}: {mismatches[:3]}"
+
+
+def _body_reduce_scatter_follower_failure(rank, world_size):
+    """Error-flag propagation through the reduce_scatter strategy."""
+    from llmd_kvcache_amd.kvblock.gpu_index import TableIndexConfig, _to_i64
+    from llmd_kvcache_amd.kvblock.keys import Key, PodEntry
+    from llmd_kvcache_amd.parallel.sharded import ShardedIndex
+
+    sharded = ShardedIndex(TableIndexConfig(capacity=1 << 10, pods_per_key=4))
+    sharded.force_reduce_scatter = True
+    keys = [Key(MODEL, 40 + i) for i in range(8)]
+    sharded.add(keys, keys, [PodEntry("pod-a", "gpu")])
+
+    flat = torch.tensor([_to_i64(k.chunk_hash) for k in keys for _ in [0]],
+                        dtype=torch.int64)
+    offsets = torch.tensor([0, 4, 8], dtype=torch.int32)  # 2 prompts
+
+    if rank == 1:
+        real = sharded.local.table.lookup
+        sharded.local.table.lookup = lambda *a, **kw: (_ for _ in ()).throw(
+            RuntimeError("injected"))
+    outcome = None
+    try:
+        sharded.sharded_scores(flat, offsets, MODEL, set())
+        outcome = "no-raise"
+    except RuntimeError as e:
+        outcome = "peer" if "peer rank" in str(e) else "local"
+    if rank == 1:
+        sharded.local.table.lookup = real
+    # a follow-up call must still work on every rank (no stuck collective)
+    scores = sharded.sharded_scores(flat, offsets, MODEL, set())
+    ok = abs(float(scores[0].max().item()) - 4.0) < 1e-5
+    return (outcome, ok)
+
+
+def test_reduce_scatter_error_flag():
+    results = run_distributed("_body_reduce_scatter_follower_failure")
+    assert results[0] == ("peer", True)
+    assert results[1] == ("local", True)
