@@ -61,24 +61,31 @@ def main():
     idx.apply_event_batches(batches[0], tp)
     torch.cuda.synchronize()
 
-    # end-to-end
+    # end-to-end pipelined (fresh inserts; host staging of batch i+1
+    # overlaps batch i's kernels - the events pool's deployment shape)
     t0 = time.monotonic()
     for b in batches[1:]:
         idx.apply_event_batches(b, tp)
     torch.cuda.synchronize()
     wall = time.monotonic() - t0
     blocks = (len(batches) - 1) * EVENTS_PER_BATCH * BLOCKS_PER_EVENT
-    print(f"end-to-end: {blocks / wall / 1e6:.2f}M blocks/s "
+    print(f"end-to-end pipelined: {blocks / wall / 1e6:.2f}M blocks/s "
           f"({wall * 1000 / (len(batches) - 1):.2f} ms/batch)")
 
-    # host-only estimate: time with a sync after every batch (kernel
-    # time exposed) vs sync-free pipeline above
+    # per-batch synced on a FRESH index (re-inserting existing keys is
+    # cheaper and cache-warm, so reusing the table above would flatter
+    # this number)
+    idx2 = GpuIndex(GpuIndexConfig(capacity=1 << 22, pods_per_key=10))
+    for p in range(NUM_PODS):
+        idx2.registry.pod_id(f"pod-{p}")
+    idx2.apply_event_batches(batches[0], tp)
+    torch.cuda.synchronize()
     t0 = time.monotonic()
     for b in batches[1:]:
-        idx.apply_event_batches(b, tp)
+        idx2.apply_event_batches(b, tp)
         torch.cuda.synchronize()
     synced = time.monotonic() - t0
-    print(f"per-batch synced: {blocks / synced / 1e6:.2f}M blocks/s "
+    print(f"per-batch synced (fresh): {blocks / synced / 1e6:.2f}M blocks/s "
           f"({synced * 1000 / (len(batches) - 1):.2f} ms/batch)")
 
 
