@@ -81,12 +81,15 @@ def populate_index(index, device, rank):
 
     # Each "chain" is a shared prefix owned by a subset of pods; prompts
     # later re-walk these chains. Store chain token ids for prompt gen.
+    # Phase 1 (untimed): build every event batch up front so the timed
+    # apply loop measures the PIPELINED deployment shape - the events
+    # pool applies batch i+1's host staging while batch i's kernels run;
+    # a sync per batch would serialize them (and read ~11-12M instead of
+    # the ~22M the pool actually sustains; profiles/r02_kernel_stats.md).
     chains = []
-    t_total = 0.0
-    blocks_done = 0
     engine_hash = 1
     batch = []
-    batches_applied = 0
+    batches = []
     for e in range(n_events):
         tokens = rng.integers(0, VOCAB, size=BLOCKS_PER_EVENT * BLOCK_SIZE,
                               dtype=np.int64)
@@ -99,19 +102,29 @@ def populate_index(index, device, rank):
         if e < 2048:  # keep a sample of chains for the read workload
             chains.append(tokens)
         if len(batch) >= events_per_batch or e == n_events - 1:
-            t0 = time.monotonic()
-            if hasattr(index, "apply_event_batches") and index.table.is_cuda:
-                index.apply_event_batches(batch, tp)
-                torch.cuda.synchronize()
-            else:
-                _apply_cpu(index, batch, tp)
-            batches_applied += 1
-            if batches_applied > 1:  # first batch pays kernel compilation
-                t_total += time.monotonic() - t0
-                blocks_done += sum(len(ev.block_hashes)
-                                   for _, _, evs in batch for ev in evs)
+            batches.append(batch)
             batch = []
-    rate = blocks_done / t_total if t_total > 0 else 0.0
+
+    gpu_path = hasattr(index, "apply_event_batches") and index.table.is_cuda
+
+    def apply(b):
+        if gpu_path:
+            index.apply_event_batches(b, tp)
+        else:
+            _apply_cpu(index, b, tp)
+
+    apply(batches[0])  # warmup batch pays kernel compilation (untimed)
+    if gpu_path:
+        torch.cuda.synchronize()
+    blocks_done = sum(len(ev.block_hashes)
+                      for b in batches[1:] for _, _, evs in b for ev in evs)
+    t0 = time.monotonic()
+    for b in batches[1:]:
+        apply(b)
+    if gpu_path:
+        torch.cuda.synchronize()
+    t_total = time.monotonic() - t0
+    rate = blocks_done / t_total if t_total > 0 and len(batches) > 1 else 0.0
     return chains, rate
 
 
