@@ -73,7 +73,8 @@ def populate_index(index, device, rank):
     rng = np.random.default_rng(1234)  # same stream on every rank
     tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=BLOCK_SIZE))
     n_events = NUM_BLOCKS // BLOCKS_PER_EVENT
-    events_per_batch = 512
+    # keep >=2 batches so small --blocks runs still measure ingest
+    events_per_batch = min(512, max(1, n_events // 4))
     pods = [f"pod-{i}" for i in range(NUM_PODS)]
     # register every pod up front so the mask width is stable
     for p in pods:
