@@ -127,3 +127,69 @@ def test_full_service_stack_on_gpu(tokenizer_dir):
         http.stop()
         events.shutdown()
         indexer.shutdown()
+
+
+def test_wirefront_on_gpu(tokenizer_dir):
+    """Native wire front over the HBM table: text request (tokenization
+    included), token request, prompt-cache warm repeat."""
+    import socket
+
+    from llmd_kvcache_amd.indexer import Config, Indexer
+    from llmd_kvcache_amd.kvblock.gpu_index import GpuIndex, GpuIndexConfig
+    from llmd_kvcache_amd.kvblock.keys import PodEntry
+    from llmd_kvcache_amd.kvblock.token_processor import (
+        ChunkedTokenDatabase, TokenProcessorConfig)
+    from llmd_kvcache_amd.service.wirefront import WireIndexerService
+    from llmd_kvcache_amd.tokenization.pool import TokenizationPool
+    from llmd_kvcache_amd.tokenization.prefixstore import LRUTokenStore
+    from llmd_kvcache_amd.tokenization.tokenizer import (
+        LocalTokenizerConfig,
+        new_cached_local_tokenizer,
+    )
+
+    cfg = Config()
+    cfg.token_processor = TokenProcessorConfig(block_size=BLOCK_SIZE)
+    index = GpuIndex(GpuIndexConfig(capacity=1 << 16, pods_per_key=10))
+    tokenizer = new_cached_local_tokenizer(
+        LocalTokenizerConfig(auto_discover_dir=tokenizer_dir))
+    pool = TokenizationPool(cfg.tokenizers_pool, indexer=LRUTokenStore(),
+                            tokenizer=tokenizer)
+    pool.run()
+    indexer = Indexer(cfg, tokenization_pool=pool, kv_block_index=index)
+
+    prompt = " ".join(f"w{i % 100}" for i in range(64))
+    tokens, _ = tokenizer.encode(prompt, MODEL)
+    tp = ChunkedTokenDatabase(cfg.token_processor)
+    keys = tp.tokens_to_kv_block_keys(None, tokens, MODEL)
+    index.add(keys, keys, [PodEntry("gpu-pod", "gpu")])
+
+    svc = WireIndexerService(indexer)
+    port = svc.start(port=0, n_io=2)
+    try:
+        def post(obj):
+            body = json.dumps(obj).encode()
+            s = socket.create_connection(("127.0.0.1", port), timeout=10)
+            s.sendall((f"POST /score HTTP/1.1\r\nhost: x\r\n"
+                       f"content-length: {len(body)}\r\n\r\n").encode()
+                      + body)
+            buf = b""
+            while b"\r\n\r\n" not in buf:
+                buf += s.recv(65536)
+            head, rest = buf.split(b"\r\n\r\n", 1)
+            clen = int([ln.split(b":")[1] for ln in head.split(b"\r\n")
+                        if ln.lower().startswith(b"content-length")][0])
+            while len(rest) < clen:
+                rest += s.recv(65536)
+            s.close()
+            return int(head.split(b" ")[1]), json.loads(rest[:clen])
+
+        status, out = post({"model": MODEL, "prompt": prompt})
+        assert status == 200 and out["scores"] == {"gpu-pod": 4.0}
+        status, out = post({"model": MODEL, "tokens": tokens})
+        assert status == 200 and out["scores"] == {"gpu-pod": 4.0}
+        status, out = post({"model": MODEL, "prompt": prompt})
+        assert status == 200 and out["scores"] == {"gpu-pod": 4.0}
+        assert svc._front.prompt_cache_hits() >= 1
+    finally:
+        svc.stop()
+        pool.shutdown()
