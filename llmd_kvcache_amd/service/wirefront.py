@@ -94,12 +94,22 @@ class WireIndexerService:
         return (idx, model_id, filt, weights, num_pods, n_tiers,
                 _to_i64(tp.config.init_hash()), tp.block_size)
 
+    @staticmethod
+    def _run_flat(idx, *args):
+        if idx.table.is_cuda:
+            return idx.table.ops.wire_score_flat(*args)
+        # CPU tables are single-writer/locked-reader (gpu_index.py
+        # fused_scores takes the same lock); the op releases the GIL so
+        # the lock is what serializes against concurrent event applies
+        with idx._write_lock:
+            return idx.table.ops.wire_score_flat(*args)
+
     def _score_tokens_cb(self, model: str, pods: Sequence[str],
                          tokens_flat, offsets):
         (idx, model_id, filt, weights, num_pods, n_tiers, init,
          bs) = self._stage(model, pods)
-        scores = idx.table.ops.wire_score_flat(
-            *idx.table._t(), tokens_flat, offsets, model_id, filt,
+        scores = self._run_flat(
+            idx, *idx.table._t(), tokens_flat, offsets, model_id, filt,
             weights, num_pods, idx.table.next_epoch(), init, bs, n_tiers)
         return scores, list(idx.registry.id_to_pod)
 
@@ -122,8 +132,8 @@ class WireIndexerService:
         off_t = torch.from_numpy(off)
         (idx, model_id, filt, weights, num_pods, n_tiers, init,
          bs) = self._stage(model, pods)
-        scores = idx.table.ops.wire_score_flat(
-            *idx.table._t(), flat_t, off_t, model_id, filt, weights,
+        scores = self._run_flat(
+            idx, *idx.table._t(), flat_t, off_t, model_id, filt, weights,
             num_pods, idx.table.next_epoch(), init, bs, n_tiers)
         # 4-tuple: the C++ side feeds elements 2/3 (int32 tokens +
         # offsets) into its prompt cache so repeats skip Python
