@@ -194,6 +194,80 @@ def _wire_client_proc(port, blob, n_blobs, depth, q):
     q.put((n_blobs * depth, dt))
 
 
+def _grpc_client_proc(target, prompt, n_reqs, q):
+    """gRPC throughput client (subprocess, real grpcio channel)."""
+    import time as _time
+
+    from llmd_kvcache_amd.service.grpc_server import IndexerClient
+
+    client = IndexerClient(target, timeout_s=30.0)
+    client.get_pod_scores(prompt, MODEL)  # warm
+    t0 = _time.monotonic()
+    for _ in range(n_reqs):
+        client.get_pod_scores(prompt, MODEL)
+    dt = _time.monotonic() - t0
+    client.close()
+    q.put((n_reqs, dt))
+
+
+def measure_grpc(indexer, chains, prefix_frac, n_procs=8, per_proc=400):
+    """Score() QPS + p50 through the real gRPC surface (grpcio server,
+    hand-written proto3 codec, coalesced handlers) - the reference's
+    IndexerService API measured end to end with tokenization."""
+    import multiprocessing as _mp
+    import statistics as _stats
+
+    import numpy as np
+
+    from llmd_kvcache_amd.service.grpc_server import (IndexerClient,
+                                                      serve)
+
+    server = serve(indexer, address="127.0.0.1:0", max_workers=32)
+    target = f"127.0.0.1:{server._kvidx_port}"
+    out = {}
+    try:
+        rng = np.random.default_rng(555)
+        chain = chains[rng.integers(len(chains))]
+        reuse = int(PROMPT_TOKENS * prefix_frac)
+        toks = np.empty(PROMPT_TOKENS, dtype=np.int64)
+        toks[:reuse] = np.tile(chain, reuse // len(chain) + 1)[:reuse]
+        toks[reuse:] = rng.integers(0, VOCAB, size=PROMPT_TOKENS - reuse)
+        prompt = " ".join(str(x) for x in toks)
+
+        client = IndexerClient(target, timeout_s=30.0)
+        lat = []
+        for _ in range(100):
+            t0 = time.monotonic()
+            client.get_pod_scores(prompt, MODEL)
+            lat.append(time.monotonic() - t0)
+        client.close()
+        out["grpc_p50_ms"] = _stats.median(lat) * 1000.0
+
+        ctx = _mp.get_context("spawn")
+        q = ctx.SimpleQueue()
+        procs = [
+            ctx.Process(target=_grpc_client_proc,
+                        args=(target, prompt, per_proc, q))
+            for _ in range(n_procs)
+        ]
+        for p in procs:
+            p.start()
+        total, worst = 0, 0.0
+        for _ in procs:
+            n, dt = q.get()
+            total += n
+            worst = max(worst, dt)
+        for p in procs:
+            p.join(timeout=60)
+        out["grpc_qps"] = total / worst if worst else 0.0
+    finally:
+        server.stop(None)
+        co = getattr(server, "_kvidx_coalescer", None)
+        if co is not None:
+            co.stop()
+    return out
+
+
 def measure_wire(indexer, chains, prefix_frac, seconds_per_mode=4.0):
     """Wire-level Score() benchmark against the native front: real
     sockets, HTTP parse, micro-batch scoring, JSON responses.  Returns
@@ -255,7 +329,7 @@ def measure_wire(indexer, chains, prefix_frac, seconds_per_mode=4.0):
                 rest += s.recv(1 << 20)
             return rest[clen:]
 
-        def measure_mode(mode, n_procs=8, depth=96):
+        def measure_mode(mode, n_procs=8, depth=192):
             import multiprocessing as _mp
             import statistics as _stats
 
@@ -771,6 +845,8 @@ def main():
             wire_indexer = build_wire_indexer(index)
             wire_stats = measure_wire(wire_indexer, chains,
                                       args.prefix_frac)
+            wire_stats.update(measure_grpc(wire_indexer, chains,
+                                           args.prefix_frac))
             log(rank, f"# wire: {wire_stats}")
         except Exception as e:  # pragma: no cover - keep headline alive
             log(rank, f"# wire bench failed: {e}")
