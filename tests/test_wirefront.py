@@ -266,3 +266,61 @@ class TestWireServiceExample:
         assert cfg_idx.native is not None
         monkeypatch.setenv("KVCACHE_INDEX_BACKEND", "tiered")
         assert mod.build_index_config().tiered is not None
+
+
+class TestWireFrontRobustness:
+    def test_garbage_bytes_do_not_wedge_server(self, service):
+        """Random garbage, truncated requests, oversized headers and
+        abrupt disconnects must leave the server serving."""
+        import random
+
+        _, port, tokens = service
+        rng = random.Random(3)
+        for i in range(30):
+            s = _connect(port)
+            kind = i % 5
+            try:
+                if kind == 0:
+                    s.sendall(bytes(rng.randrange(256)
+                                    for _ in range(rng.randrange(1, 400))))
+                elif kind == 1:  # truncated body, then hang up
+                    s.sendall(b"POST /score HTTP/1.1\r\n"
+                              b"content-length: 5000\r\n\r\n{\"mo")
+                elif kind == 2:  # header larger than the guard
+                    s.sendall(b"GET / HTTP/1.1\r\n" +
+                              b"x-pad: " + b"a" * 5000 + b"\r\n")
+                elif kind == 3:  # pipelined valid + garbage
+                    s.sendall(_http_post("/score", {"model": MODEL,
+                                                    "tokens": tokens}) +
+                              b"\x00\xff\x01garbage")
+                    _read_response(s)
+                else:  # content-length lies (smaller than body)
+                    body = b'{"model":"m","tokens":[1,2,3]}'
+                    s.sendall(b"POST /score HTTP/1.1\r\n"
+                              b"content-length: 10\r\n\r\n" + body)
+            except (BrokenPipeError, ConnectionResetError):
+                pass
+            finally:
+                s.close()
+        # server still healthy and scoring
+        s = _connect(port)
+        s.sendall(_http_post("/score", {"model": MODEL, "tokens": tokens}))
+        status, body, _ = _read_response(s)
+        assert status == 200
+        assert json.loads(body)["scores"]["pod-a"] == 8.0
+        s.close()
+
+    def test_disconnect_before_response(self, service):
+        """A client that hangs up mid-batch must not take the batcher
+        or other connections down."""
+        _, port, tokens = service
+        req = _http_post("/score", {"model": MODEL, "tokens": tokens})
+        for _ in range(5):
+            s = _connect(port)
+            s.sendall(req * 4)
+            s.close()  # responses will hit a dead socket
+        s = _connect(port)
+        s.sendall(req)
+        status, body, _ = _read_response(s)
+        assert status == 200 and json.loads(body)["scores"]["pod-a"] == 8.0
+        s.close()
