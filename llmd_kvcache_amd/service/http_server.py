@@ -27,7 +27,7 @@ from ..preprocessing import chat_completions as cc
 logger = logging.getLogger("llmd_kvcache_amd.http")
 
 
-def make_handler(indexer: Indexer):
+def make_handler(indexer: Indexer, coalescer=None):
     class Handler(BaseHTTPRequestHandler):
         protocol_version = "HTTP/1.1"
 
@@ -96,7 +96,12 @@ def make_handler(indexer: Indexer):
             if not prompt:
                 self._send_error(400, "field 'prompt' required")
                 return
-            pods = indexer.get_pod_scores(None, prompt, model, [])
+            if coalescer is not None:
+                tokens = indexer.tokenizers_pool.tokenize(
+                    None, prompt, model)
+                pods = coalescer.score(tokens, model, req.get("pods", []))
+            else:
+                pods = indexer.get_pod_scores(None, prompt, model, [])
             self._send_json(200, pods or {})
 
         def _score_batch(self):
@@ -151,8 +156,20 @@ def make_handler(indexer: Indexer):
 
 
 class HttpService:
-    def __init__(self, indexer: Indexer, host: str = "0.0.0.0", port: int = 8080):
-        self.server = ThreadingHTTPServer((host, port), make_handler(indexer))
+    def __init__(self, indexer: Indexer, host: str = "0.0.0.0",
+                 port: int = 8080, coalesce: bool = True):
+        """coalesce=True batches concurrent /score_completions requests
+        from the threaded handlers into shared fused kernel launches
+        (service/coalesce.py) - same default as the gRPC front."""
+        self._coalescer = None
+        if (coalesce and hasattr(indexer, "tokenizers_pool")
+                and hasattr(indexer, "score_tokens_batch")):
+            from .coalesce import CoalescingScorer
+
+            self._coalescer = CoalescingScorer(indexer)
+            self._coalescer.start()
+        self.server = ThreadingHTTPServer(
+            (host, port), make_handler(indexer, self._coalescer))
         self.port = self.server.server_address[1]
         self._thread: Optional[threading.Thread] = None
 
@@ -166,5 +183,7 @@ class HttpService:
     def stop(self) -> None:
         self.server.shutdown()
         self.server.server_close()
+        if self._coalescer is not None:
+            self._coalescer.stop()
         if self._thread:
             self._thread.join(timeout=2.0)
