@@ -614,18 +614,24 @@ class WireFront {
         int fd = events[i].data.fd;
         auto it = conns.find(fd);
         if (it == conns.end()) continue;
-        Conn* c = it->second.get();
+        // hold a strong ref: flush() may close_conn(), which erases the
+        // map entry (invalidating `it`) - the local ConnPtr keeps the
+        // object alive and `closed` gates further use
+        ConnPtr cp = it->second;
         if (events[i].events & (EPOLLHUP | EPOLLERR)) {
           close_conn(ep, conns, fd);
           continue;
         }
-        if (events[i].events & EPOLLOUT) flush(ep, conns, c);
+        if (events[i].events & EPOLLOUT) {
+          flush(ep, conns, cp.get());
+          if (cp->closed.load()) continue;
+        }
         if (events[i].events & EPOLLIN) {
-          if (!read_and_parse(it->second)) {
+          if (!read_and_parse(cp)) {
             close_conn(ep, conns, fd);
             continue;
           }
-          flush(ep, conns, c);
+          flush(ep, conns, cp.get());
         }
       }
       if (!running_.load()) break;
