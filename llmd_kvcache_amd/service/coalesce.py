@@ -79,6 +79,17 @@ class CoalescingScorer:
             self._thread = None
         if t is not None:
             t.join(timeout=2.0)
+        # A caller that passed the _running check just before stop() may
+        # have enqueued AFTER the sentinel; fail those explicitly so no
+        # caller blocks forever on its event.
+        while True:
+            try:
+                p = self._queue.get_nowait()
+            except queue.Empty:
+                break
+            if p is not None:
+                p.error = RuntimeError("coalescing scorer stopped")
+                p.event.set()
 
     # -- API -----------------------------------------------------------
     def score(self, tokens: Sequence[int], model: str,
