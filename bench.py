@@ -280,8 +280,8 @@ def measure_wire(indexer, chains, prefix_frac, seconds_per_mode=4.0):
 
     from llmd_kvcache_amd.service.wirefront import WireIndexerService
 
-    svc = WireIndexerService(indexer, max_batch=8192)
-    port = svc.start(port=0, n_io=8)
+    svc = WireIndexerService(indexer, max_batch=8192, n_batchers=4)
+    port = svc.start(port=0, n_io=16)
     out = {}
     try:
         rng = np.random.default_rng(777)

@@ -30,7 +30,8 @@ from ..indexer import Indexer
 
 
 class WireIndexerService:
-    def __init__(self, indexer: Indexer, max_batch: int = 4096):
+    def __init__(self, indexer: Indexer, max_batch: int = 4096,
+                 n_batchers: int = 2):
         from ..ops import cpu_ext
 
         ops = cpu_ext.require()
@@ -43,7 +44,8 @@ class WireIndexerService:
                 "scoring op implements that algorithm)")
         self.indexer = indexer
         self._front = ops.WireFront(self._score_tokens_cb,
-                                    self._score_text_cb, max_batch)
+                                    self._score_text_cb, max_batch,
+                                    n_batchers)
         self._running = False
 
     # -- lifecycle -----------------------------------------------------

@@ -25,9 +25,8 @@ def main():
     indexer = bench.build_wire_indexer(idx)
 
     for n_io, n_batchers in [(8, 2), (12, 2), (8, 3), (12, 3), (16, 4)]:
-        svc = WireIndexerService(indexer, max_batch=8192)
-        svc._front = type(svc._front)(svc._score_tokens_cb,
-                                      svc._score_text_cb, 8192, n_batchers)
+        svc = WireIndexerService(indexer, max_batch=8192,
+                                 n_batchers=n_batchers)
         port = svc.start(port=0, n_io=n_io)
         try:
             # reuse bench's measure_mode machinery via measure_wire with a
