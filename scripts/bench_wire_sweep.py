@@ -24,7 +24,7 @@ def main():
     chains, _ = bench.populate_index(idx, torch.device("cuda:0"), 0)
     indexer = bench.build_wire_indexer(idx)
 
-    for n_io, n_batchers in [(8, 2), (12, 2), (8, 3), (12, 3), (16, 4)]:
+    for n_io, n_batchers, n_procs in [(16, 4, 8), (16, 4, 12), (16, 4, 16), (20, 5, 12)]:
         svc = WireIndexerService(indexer, max_batch=8192,
                                  n_batchers=n_batchers)
         port = svc.start(port=0, n_io=n_io)
@@ -55,7 +55,7 @@ def main():
             blob = b"".join((reqs * (depth // len(reqs) + 1))[:depth])
             ctx = _mp.get_context("spawn")
             q = ctx.SimpleQueue()
-            n_procs, n_blobs = 8, 24
+            n_blobs = 24
             procs = [ctx.Process(target=bench._wire_client_proc,
                                  args=(port, blob, n_blobs, depth, q))
                      for _ in range(n_procs)]
@@ -69,7 +69,7 @@ def main():
             for p in procs:
                 p.join(timeout=60)
             reqs_srv, batches = svc.stats()
-            print(f"io={n_io} batchers={n_batchers}: "
+            print(f"io={n_io} batchers={n_batchers} clients={n_procs}: "
                   f"{total / worst / 1000:.1f}k req/s "
                   f"(avg batch {reqs_srv / max(batches, 1):.1f})",
                   flush=True)
