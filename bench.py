@@ -329,7 +329,7 @@ def measure_wire(indexer, chains, prefix_frac, seconds_per_mode=4.0):
                 rest += s.recv(1 << 20)
             return rest[clen:]
 
-        def measure_mode(mode, n_procs=8, depth=96):
+        def measure_mode(mode, n_procs=16, depth=96):
             import multiprocessing as _mp
             import statistics as _stats
 

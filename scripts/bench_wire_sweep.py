@@ -24,7 +24,7 @@ def main():
     chains, _ = bench.populate_index(idx, torch.device("cuda:0"), 0)
     indexer = bench.build_wire_indexer(idx)
 
-    for n_io, n_batchers, n_procs in [(16, 4, 8), (16, 4, 12), (16, 4, 16), (20, 5, 12)]:
+    for n_io, n_batchers, n_procs in [(16, 4, 16), (16, 4, 24), (16, 4, 32), (24, 6, 32)]:
         svc = WireIndexerService(indexer, max_batch=8192,
                                  n_batchers=n_batchers)
         port = svc.start(port=0, n_io=n_io)
