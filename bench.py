@@ -164,19 +164,22 @@ def build_prompts(chains, n_prompts, device, seed, prefix_frac=0.5):
     return t, offsets
 
 
-def _wire_client_proc(port, blob, n_blobs, depth, q):
+def _wire_client_proc(port, blob, n_blobs, depth, q, barrier):
     """Throughput client (subprocess): pipelined bursts over one
     connection; counts completed responses.  Runs in a separate process
     so client-side Python work does not steal GIL time from the server's
-    per-batch scoring callback."""
+    per-batch scoring callback.  All clients rendezvous at the barrier
+    AFTER interpreter boot + connect + one warm blob, so the measured
+    windows truly overlap (CLOCK_MONOTONIC is system-wide on Linux; the
+    caller aggregates total / (max end - min start))."""
     import socket as _socket
     import time as _time
 
     PAT = b"HTTP/1.1 200"
     s = _socket.create_connection(("127.0.0.1", port), timeout=30)
     s.settimeout(30)
-    t0 = _time.monotonic()
-    for _ in range(n_blobs):
+
+    def run_blob():
         s.sendall(blob)
         got = 0
         tail = b""
@@ -184,14 +187,18 @@ def _wire_client_proc(port, blob, n_blobs, depth, q):
             chunk = s.recv(1 << 20)
             if not chunk:
                 raise RuntimeError("server closed connection")
-            # count response heads; the carried tail is shorter than the
-            # pattern so a boundary-spanning match is counted exactly once
             work = tail + chunk
             got += work.count(PAT)
             tail = work[-(len(PAT) - 1):]
-    dt = _time.monotonic() - t0
+
+    run_blob()  # warm (connection, server caches, allocator)
+    barrier.wait()
+    t0 = _time.monotonic()
+    for _ in range(n_blobs):
+        run_blob()
+    t1 = _time.monotonic()
     s.close()
-    q.put((n_blobs * depth, dt))
+    q.put((n_blobs * depth, t0, t1))
 
 
 def _grpc_client_proc(target, prompt, n_reqs, q):
@@ -363,23 +370,25 @@ def measure_wire(indexer, chains, prefix_frac, seconds_per_mode=4.0):
             blob = b"".join(blob_src)
             ctx = _mp.get_context("spawn")
             q = ctx.SimpleQueue()
+            barrier = ctx.Barrier(n_procs)
             procs = [
                 ctx.Process(target=_wire_client_proc,
-                            args=(port, blob, n_blobs, depth, q))
+                            args=(port, blob, n_blobs, depth, q, barrier))
                 for _ in range(n_procs)
             ]
-            t0 = time.monotonic()
             for p in procs:
                 p.start()
             total = 0
-            worst = 0.0
+            starts, ends = [], []
             for _ in procs:
-                n, dt = q.get()
+                n, ts, te = q.get()
                 total += n
-                worst = max(worst, dt)
+                starts.append(ts)
+                ends.append(te)
             for p in procs:
                 p.join(timeout=30)
-            qps = total / worst if worst > 0 else 0.0
+            wall = max(ends) - min(starts)
+            qps = total / wall if wall > 0 else 0.0
             return qps, p50_ms
 
         out["wire_qps_tokens"], out["wire_p50_tokens_ms"] = (

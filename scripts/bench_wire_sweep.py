@@ -55,22 +55,26 @@ def main():
             blob = b"".join((reqs * (depth // len(reqs) + 1))[:depth])
             ctx = _mp.get_context("spawn")
             q = ctx.SimpleQueue()
-            n_blobs = 24
+            n_blobs = 48
+            barrier = ctx.Barrier(n_procs)
             procs = [ctx.Process(target=bench._wire_client_proc,
-                                 args=(port, blob, n_blobs, depth, q))
+                                 args=(port, blob, n_blobs, depth, q,
+                                       barrier))
                      for _ in range(n_procs)]
             for p in procs:
                 p.start()
-            total, worst = 0, 0.0
+            total, starts, ends = 0, [], []
             for _ in procs:
-                n, dt = q.get()
+                n, ts, te = q.get()
                 total += n
-                worst = max(worst, dt)
+                starts.append(ts)
+                ends.append(te)
             for p in procs:
                 p.join(timeout=60)
+            wall = max(ends) - min(starts)
             reqs_srv, batches = svc.stats()
             print(f"io={n_io} batchers={n_batchers} clients={n_procs}: "
-                  f"{total / worst / 1000:.1f}k req/s "
+                  f"{total / wall / 1000:.1f}k req/s "
                   f"(avg batch {reqs_srv / max(batches, 1):.1f})",
                   flush=True)
         finally:
