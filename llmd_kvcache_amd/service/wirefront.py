@@ -121,7 +121,10 @@ class WireIndexerService:
         import torch
 
         pool = self.indexer.tokenizers_pool
-        token_lists = [pool.tokenize(None, p, model) for p in prompts]
+        if hasattr(pool, "tokenize_batch"):
+            token_lists = pool.tokenize_batch(list(prompts), model)
+        else:  # custom pool injected without the batch API
+            token_lists = [pool.tokenize(None, p, model) for p in prompts]
         lens = [len(t) for t in token_lists]
         flat = np.empty(sum(lens), dtype=np.int64)
         off = np.zeros(len(token_lists) + 1, dtype=np.int64)

@@ -45,6 +45,16 @@ class Tokenizer:
     ) -> Tuple[List[int], List[Offset]]:
         raise NotImplementedError
 
+    def encode_batch(
+        self, prompts: List[str], model_name: str,
+        add_special_tokens: bool = True,
+    ) -> List[Tuple[List[int], List[Offset]]]:
+        """Batch encode; backends with a parallel core (HF tokenizers'
+        Rust encode_batch releases the GIL and uses a rayon pool)
+        override this - the default is a sequential loop."""
+        return [self.encode(p, model_name, add_special_tokens)
+                for p in prompts]
+
     def render_chat_template(self, req) -> str:
         raise NotImplementedError
 
@@ -190,6 +200,16 @@ class CachedTokenizer(Tokenizer):
         enc = tok.encode(prompt, add_special_tokens=add_special_tokens)
         return list(enc.ids), [tuple(o) for o in enc.offsets]
 
+    def encode_batch(
+        self, prompts: List[str], model_name: str,
+        add_special_tokens: bool = True,
+    ) -> List[Tuple[List[int], List[Offset]]]:
+        tok = self._get(model_name)
+        encs = tok.encode_batch(prompts,
+                                add_special_tokens=add_special_tokens)
+        return [(list(e.ids), [tuple(o) for o in e.offsets])
+                for e in encs]
+
     def render_chat_template(self, req) -> str:
         from ..preprocessing import chat_completions
 
@@ -273,6 +293,29 @@ class CompositeTokenizer(Tokenizer):
                 errors.append(f"{tok.type}: {e}")
         raise TokenizationError(
             f"all tokenizer backends failed for {model_name!r}: {'; '.join(errors)}"
+        )
+
+    def encode_batch(
+        self, prompts: List[str], model_name: str,
+        add_special_tokens: bool = True,
+    ) -> List[Tuple[List[int], List[Offset]]]:
+        from ..metrics import collector
+
+        errors = []
+        for tok in self.chain:
+            try:
+                t0 = collector.monotonic()
+                result = tok.encode_batch(prompts, model_name,
+                                          add_special_tokens)
+                collector.observe_tokenization(
+                    tok.type, collector.monotonic() - t0,
+                    sum(len(r[0]) for r in result))
+                return result
+            except Exception as e:
+                errors.append(f"{tok.type}: {e}")
+        raise TokenizationError(
+            f"all tokenizer backends failed for {model_name!r}: "
+            f"{'; '.join(errors)}"
         )
 
     def render_chat_template(self, req) -> str:

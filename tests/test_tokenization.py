@@ -555,3 +555,59 @@ class TestUdsMultiWorker:
             t.join(timeout=10)
         assert not overlaps
         assert _time.monotonic() - t0 >= 0.55  # serialized, not parallel
+
+
+class TestTokenizeBatch:
+    def test_batch_matches_sequential_and_uses_store(
+            self, tokenizer_fixture_dir):
+        from llmd_kvcache_amd.tokenization.pool import (TokenizationConfig,
+                                                        TokenizationPool)
+        from llmd_kvcache_amd.tokenization.tokenizer import (
+            LocalTokenizerConfig, new_cached_local_tokenizer)
+
+        tok = new_cached_local_tokenizer(
+            LocalTokenizerConfig(auto_discover_dir=tokenizer_fixture_dir))
+        pool = TokenizationPool(TokenizationConfig(), tokenizer=tok)
+        # prompts must span full 256-char prefix-store blocks to be
+        # cacheable (lru_store semantics: partial tail chunks drop)
+        prompts = ["hello world " * 60, "hello there " * 55,
+                   "hello world " * 60]
+        batch = pool.tokenize_batch(prompts, "test-model")
+        seq = [pool.tokenize(None, p, "test-model") for p in prompts]
+        assert batch == seq
+        # warm repeat: all served from the prefix store (no encode)
+        calls = {"n": 0}
+        orig = tok.encode_batch
+
+        def counting(*a, **kw):
+            calls["n"] += 1
+            return orig(*a, **kw)
+
+        tok.encode_batch = counting
+        again = pool.tokenize_batch(prompts, "test-model")
+        assert again == batch
+        assert calls["n"] == 0
+
+    def test_composite_encode_batch_falls_back(self):
+        from llmd_kvcache_amd.tokenization.tokenizer import (
+            CompositeTokenizer, Tokenizer)
+
+        class Bad(Tokenizer):
+            @property
+            def type(self):
+                return "bad"
+
+            def encode(self, *a, **kw):
+                raise RuntimeError("nope")
+
+        class Good(Tokenizer):
+            @property
+            def type(self):
+                return "good"
+
+            def encode(self, prompt, model, add_special_tokens=True):
+                return [len(prompt)], [(0, len(prompt))]
+
+        comp = CompositeTokenizer([Bad(), Good()])
+        out = comp.encode_batch(["ab", "abc"], "m")
+        assert out == [([2], [(0, 2)]), ([3], [(0, 3)])]
