@@ -570,8 +570,8 @@ class TestTokenizeBatch:
         pool = TokenizationPool(TokenizationConfig(), tokenizer=tok)
         # prompts must span full 256-char prefix-store blocks to be
         # cacheable (lru_store semantics: partial tail chunks drop)
-        prompts = ["hello world " * 60, "hello there " * 55,
-                   "hello world " * 60]
+        prompts = ["hello world " * 300, "hello there " * 280,
+                   "hello world " * 300]
         batch = pool.tokenize_batch(prompts, "test-model")
         seq = [pool.tokenize(None, p, "test-model") for p in prompts]
         assert batch == seq
