@@ -51,7 +51,11 @@ class Tokenizer:
     ) -> List[Tuple[List[int], List[Offset]]]:
         """Batch encode; backends with a parallel core (HF tokenizers'
         Rust encode_batch releases the GIL and uses a rayon pool)
-        override this - the default is a sequential loop."""
+        override this - the default is a sequential loop.  The default
+        flag value calls the 2-arg form so duck-typed tokenizers that
+        omit add_special_tokens keep working."""
+        if add_special_tokens:
+            return [self.encode(p, model_name) for p in prompts]
         return [self.encode(p, model_name, add_special_tokens)
                 for p in prompts]
 
