@@ -350,3 +350,60 @@ class TestTierRegistryOverflow:
         ok = BlockStored([11, 12], None, tokens, 4, medium="gpu")
         digest_events(idx, tp, "pod-x", "m", [ok])
         assert len(idx.lookup(keys, set())) == 2
+
+
+class TestWireScoreFlatOp:
+    def test_matches_python_paths(self):
+        """ops.wire_score_flat (the GIL-released whole-batch wire op)
+        must equal the Python-orchestrated batch path on random data."""
+        import random
+
+        import torch
+
+        from llmd_kvcache_amd.indexer import Config, Indexer
+        from llmd_kvcache_amd.kvblock.gpu_index import (NativeIndex,
+                                                        TableIndexConfig,
+                                                        _to_i64)
+        from llmd_kvcache_amd.kvblock.keys import PodEntry
+        from llmd_kvcache_amd.kvblock.token_processor import (
+            ChunkedTokenDatabase, TokenProcessorConfig)
+
+        rng = random.Random(17)
+        cfg = Config(token_processor=TokenProcessorConfig(block_size=4))
+        idx = NativeIndex(TableIndexConfig(capacity=1 << 12, pods_per_key=6))
+        indexer = Indexer(cfg, kv_block_index=idx)
+        tp = ChunkedTokenDatabase(cfg.token_processor)
+
+        stored = []
+        for _ in range(30):
+            toks = [rng.randrange(1 << 30) for _ in range(rng.randrange(1, 6) * 4)]
+            keys = tp.tokens_to_kv_block_keys(None, toks, "m")
+            pod = f"pod-{rng.randrange(5)}"
+            idx.add(keys, keys, [PodEntry(pod, rng.choice(["gpu", "cpu"]))])
+            stored.append(toks)
+
+        prompts = []
+        for _ in range(20):
+            base = list(stored[rng.randrange(len(stored))])
+            if rng.random() < 0.5:
+                base += [rng.randrange(1 << 30) for _ in range(8)]
+            prompts.append(base)
+        prompts.append([1, 2, 3])  # sub-block: zero chunks
+
+        lens = [len(p) for p in prompts]
+        flat = torch.tensor([t for p in prompts for t in p],
+                            dtype=torch.int64)
+        offs = torch.tensor([0] + list(__import__("itertools").accumulate(lens)),
+                            dtype=torch.int64)
+        model_id = idx.registry.model_id("m")
+        num_pods = idx._num_pods_padded()
+        weights = idx.tier_weights()
+        n_tiers = max(1, len(idx.registry.id_to_tier))
+        init = _to_i64(tp.config.init_hash())
+        scores = idx.table.ops.wire_score_flat(
+            *idx.table._t(), flat, offs, model_id,
+            torch.zeros(0, dtype=torch.int64), weights, num_pods,
+            idx.table.next_epoch(), init, 4, n_tiers)
+        got = idx.scores_to_map(scores)
+        want = indexer.score_tokens_batch(prompts, "m", [])
+        assert got == want
