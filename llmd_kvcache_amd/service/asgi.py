@@ -88,10 +88,11 @@ class AsgiIndexerApp:
         if not isinstance(prompts, list) or not prompts:
             raise _BadRequest("field 'prompts' (non-empty list) required")
         model = body.get("model", "")
-        token_lists = [
-            self.indexer.tokenizers_pool.tokenize(None, p, model)
-            for p in prompts
-        ]
+        pool = self.indexer.tokenizers_pool
+        if hasattr(pool, "tokenize_batch"):
+            token_lists = pool.tokenize_batch(list(prompts), model)
+        else:
+            token_lists = [pool.tokenize(None, p, model) for p in prompts]
         scores = self.indexer.score_tokens_batch(
             token_lists, model, body.get("pods", []))
         await self._json(send, 200, {"scores": scores})

@@ -114,10 +114,12 @@ def make_handler(indexer: Indexer, coalescer=None):
                 self._send_error(400, "field 'prompts' (non-empty list) "
                                       "required")
                 return
-            token_lists = [
-                indexer.tokenizers_pool.tokenize(None, p, model)
-                for p in prompts
-            ]
+            pool = indexer.tokenizers_pool
+            if hasattr(pool, "tokenize_batch"):
+                token_lists = pool.tokenize_batch(list(prompts), model)
+            else:
+                token_lists = [pool.tokenize(None, p, model)
+                               for p in prompts]
             scores = indexer.score_tokens_batch(
                 token_lists, model, req.get("pods", []))
             self._send_json(200, {"scores": scores})
