@@ -52,6 +52,9 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--seconds", type=int, default=180)
     ap.add_argument("--publishers", type=int, default=4)
+    ap.add_argument("--front", choices=["http", "wire"], default="http",
+                    help="scoring surface: stdlib HTTP service or the "
+                         "native C++ wirefront")
     args = ap.parse_args()
     assert torch.cuda.is_available()
 
@@ -66,8 +69,17 @@ def main():
                                      concurrency=4),
                         index, indexer.tokens_processor)
     events.start(with_subscriber=True)
-    http = HttpService(indexer, host="127.0.0.1", port=0)
-    http.start()
+    if args.front == "wire":
+        from llmd_kvcache_amd.service.wirefront import WireIndexerService
+
+        front = WireIndexerService(indexer)
+        front_port = front.start(port=0, n_io=4)
+        score_path = "/score"
+    else:
+        front = HttpService(indexer, host="127.0.0.1", port=0)
+        front.start()
+        front_port = front.port
+        score_path = "/score_completions"
     while events._subscriber.port is None:
         time.sleep(0.05)
 
@@ -126,11 +138,13 @@ def main():
             t0 = time.monotonic()
             try:
                 req = urllib.request.Request(
-                    f"http://127.0.0.1:{http.port}/score_completions",
+                    f"http://127.0.0.1:{front_port}{score_path}",
                     data=body,
                     headers={"Content-Type": "application/json"})
                 with urllib.request.urlopen(req, timeout=5) as resp:
                     scores = json.loads(resp.read())
+                if "scores" in scores:  # wirefront response envelope
+                    scores = scores["scores"]
                 with lock:
                     stats["scores"] += 1
                     if scores.get(pod, 0) > 0:
@@ -164,7 +178,7 @@ def main():
     stop.set()
     time.sleep(1)
     events.shutdown()
-    http.stop()
+    front.stop()
     indexer.shutdown()
     mem1 = torch.cuda.memory_allocated()
     with lock:
