@@ -41,3 +41,9 @@ image:
 
 container-smoke:
 	BASE=$(BASE_IMAGE) IMG=$(IMAGE) bash scripts/container_smoke.sh
+
+.PHONY: asan
+asan:
+	g++ -std=c++17 -O1 -g -fsanitize=address,undefined \
+	    -fno-sanitize-recover=all -o build/asan_check scripts/asan_check.cc \
+	    && ./build/asan_check
