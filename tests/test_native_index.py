@@ -407,3 +407,31 @@ class TestWireScoreFlatOp:
         got = idx.scores_to_map(scores)
         want = indexer.score_tokens_batch(prompts, "m", [])
         assert got == want
+
+
+def test_asan_harness_clean():
+    """Compiles and runs the standalone ASan/UBSan harness over
+    kvidx_common.h (scripts/asan_check.cc) - sanitizer-clean hashing/
+    packing core is a CI invariant, not a manual step."""
+    import os
+    import subprocess
+    import sys
+    import tempfile
+
+    root = os.path.join(os.path.dirname(__file__), "..")
+    with tempfile.TemporaryDirectory() as td:
+        exe = os.path.join(td, "asan_check")
+        build = subprocess.run(
+            ["g++", "-std=c++17", "-O1", "-g",
+             "-fsanitize=address,undefined", "-fno-sanitize-recover=all",
+             "-o", exe, os.path.join(root, "scripts", "asan_check.cc")],
+            capture_output=True, text=True, timeout=120)
+        if build.returncode != 0:
+            import pytest
+
+            pytest.skip(f"sanitizer toolchain unavailable: "
+                        f"{build.stderr[:200]}")
+        run = subprocess.run([exe], capture_output=True, text=True,
+                             timeout=120)
+        assert run.returncode == 0, run.stderr[-2000:]
+        assert "asan_check OK" in run.stdout
