@@ -123,6 +123,8 @@ DEV int64_t dev_table_put(const DevTable& v, uint64_t h, uint32_t model,
 
 DEV void dev_pod_set_add(const DevTable& v, int64_t slot, uint32_t entry,
                          int32_t epoch) {
+  KVIDX_ASSERT(slot >= 0 && (uint64_t)slot <= v.cap_mask);
+  KVIDX_ASSERT(entry != 0);
   uint32_t* p = v.pods + slot * v.pods_per_key;
   for (int j = 0; j < v.pods_per_key; ++j)
     if (p[j] == entry) return;
@@ -706,6 +708,7 @@ __global__ void k_event_chains(
   if (g >= G) return;
   const int e_begin = grp_off[g];
   const int e_end = grp_off[g + 1];
+  KVIDX_ASSERT(e_begin >= 0 && e_begin <= e_end);
   const int grp_first_block = eh_off[e_begin];
   for (int e = e_begin; e < e_end; ++e) {
     if (ev_type[e] == 1) continue;  // removals have no chain
@@ -769,6 +772,7 @@ __global__ void k_event_chains_ev(
   if (e >= E) return;
   if (ev_type[e] == 1) return;  // removals have no chain
   const int nh = eh_off[e + 1] - eh_off[e];
+  KVIDX_ASSERT(nh >= 0 && tok_off[e + 1] >= tok_off[e]);
   const int n_chunks = (tok_off[e + 1] - tok_off[e]) / block_size;
   if (n_chunks != nh) return;  // drop mismatched event (see k_apply_events)
   uint64_t parent = init_hash;

@@ -27,6 +27,24 @@
 
 #include <cstdint>
 
+// Debug asserts (ROADMAP r1 #10): compile with KVIDX_DEBUG=1 in the
+// environment (setup.py adds -DKVIDX_DEBUG_ASSERTS) to trap on
+// violated kernel/table invariants - a hipMemcheck-style guard with
+// zero cost in release builds.
+#ifdef KVIDX_DEBUG_ASSERTS
+#if defined(__HIP_DEVICE_COMPILE__)
+#define KVIDX_ASSERT(c) \
+  do {                  \
+    if (!(c)) __builtin_trap(); \
+  } while (0)
+#else
+#include <cassert>
+#define KVIDX_ASSERT(c) assert(c)
+#endif
+#else
+#define KVIDX_ASSERT(c) ((void)0)
+#endif
+
 #if defined(__HIPCC__)
 #define KVIDX_HD __host__ __device__ __forceinline__
 #else
@@ -46,6 +64,11 @@ static constexpr int MAX_TIERS = 4;
 static constexpr int PROBE_MAX = 128;
 
 KVIDX_HD uint64_t remap_hash(uint64_t h) { return h == 0 ? 1ull : h; }
+
+KVIDX_HD uint64_t kvidx_checked_slot(uint64_t i, uint64_t cap_mask) {
+  KVIDX_ASSERT(i <= cap_mask);
+  return i;
+}
 
 KVIDX_HD uint64_t fnv1a_64_byte(uint64_t h, uint8_t b) {
   return (h ^ (uint64_t)b) * FNV64_PRIME;

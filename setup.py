@@ -25,6 +25,11 @@ sources = [
     os.path.join(CSRC, "wirefront.cpp"),
 ]
 
+extra_defs = []
+if os.environ.get("KVIDX_DEBUG", "0") == "1":
+    # hipMemcheck-style kernel/table invariant traps (kvidx_common.h)
+    extra_defs.append("-DKVIDX_DEBUG_ASSERTS")
+
 with_hip = torch.version.hip is not None
 if with_hip:
     sources.append(os.path.join(CSRC, "hip_ops.hip"))
@@ -32,15 +37,15 @@ if with_hip:
         name="llmd_kvcache_amd.ops._kvidx_C",
         sources=sources,
         extra_compile_args={
-            "cxx": ["-O3", "-std=c++17", "-DKVIDX_WITH_HIP"],
-            "nvcc": ["-O3", "-std=c++17", "-DKVIDX_WITH_HIP"],
+            "cxx": ["-O3", "-std=c++17", "-DKVIDX_WITH_HIP"] + extra_defs,
+            "nvcc": ["-O3", "-std=c++17", "-DKVIDX_WITH_HIP"] + extra_defs,
         },
     )
 else:  # CPU-only fallback (non-ROCm torch)
     ext = cpp_extension.CppExtension(
         name="llmd_kvcache_amd.ops._kvidx_C",
         sources=sources,
-        extra_compile_args=["-O3", "-std=c++17"],
+        extra_compile_args=["-O3", "-std=c++17"] + extra_defs,
     )
 
 setup(
