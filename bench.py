@@ -901,6 +901,11 @@ def main():
     qps = total_prompts / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
     p50_ms = statistics.median(latencies) * 1000.0
+    lat_sorted = sorted(latencies)
+
+    def pct(p):
+        return lat_sorted[min(len(lat_sorted) - 1,
+                              int(p * len(lat_sorted)))] * 1000.0
 
     if rank == 0:
         result = {
@@ -933,6 +938,9 @@ def main():
                 "block_size": BLOCK_SIZE,
                 "keys_per_prompt": KEYS_PER_PROMPT,
                 "p50_batch_latency_ms": round(p50_ms, 3),
+                "p90_batch_latency_ms": round(pct(0.90), 3),
+                "p99_batch_latency_ms": round(pct(0.99), 3),
+                "max_batch_latency_ms": round(lat_sorted[-1] * 1000.0, 3),
                 "p50_single_prompt_ms": (round(single_ms, 3)
                                          if single_ms is not None else None),
                 "p50_single_prompt_cold_ms": (
