@@ -487,7 +487,17 @@ class WireFront {
         max_batch_(max_batch > 0 ? (size_t)max_batch : 4096),
         n_batchers_(n_batchers > 0 ? (int)n_batchers : 1) {}
 
-  ~WireFront() { stop_nogil(); }
+  ~WireFront() {
+    // The batcher threads acquire the GIL per batch; joining them while
+    // holding it would deadlock.  stop() releases it explicitly, but a
+    // destructor reached via garbage collection runs WITH the GIL held.
+    if (PyGILState_Check()) {
+      py::gil_scoped_release rel;
+      stop_nogil();
+    } else {
+      stop_nogil();
+    }
+  }
 
   int start(const std::string& host, int port, int n_io) {
     TORCH_CHECK(!running_.load(), "wirefront already running");
