@@ -463,6 +463,30 @@ static void set_nonblock(int fd) {
   fcntl(fd, F_SETFL, fl | O_NONBLOCK);
 }
 
+// Pod identifiers are normally DNS-label-safe, but the event stream is
+// external input - escape so a hostile name cannot break the JSON.
+static void json_escape_into(std::string& out, const std::string& s) {
+  for (unsigned char c : s) {
+    switch (c) {
+      case '"': out += "\\\""; break;
+      case '\\': out += "\\\\"; break;
+      case '\b': out += "\\b"; break;
+      case '\f': out += "\\f"; break;
+      case '\n': out += "\\n"; break;
+      case '\r': out += "\\r"; break;
+      case '\t': out += "\\t"; break;
+      default:
+        if (c < 0x20) {
+          char buf[8];
+          snprintf(buf, sizeof(buf), "\\u%04x", c);
+          out += buf;
+        } else {
+          out += (char)c;
+        }
+    }
+  }
+}
+
 static std::string http_response(int status, const char* status_text,
                                  const std::string& body) {
   std::string r;
@@ -923,8 +947,14 @@ class WireFront {
               score_tokens_cb_(model, pods_t, tokens, offsets)
                   .cast<py::tuple>();
         }
-        scores = result[0].cast<at::Tensor>().contiguous();
+        scores = result[0].cast<at::Tensor>()
+                     .to(at::kFloat).contiguous();
         names = result[1].cast<std::vector<std::string>>();
+        // validate INSIDE the try: a malformed callback return must
+        // become a 500, never an uncaught throw in the batcher thread
+        if (scores.dim() != 2 ||
+            scores.size(0) != (int64_t)idxs.size())
+          throw std::runtime_error("scorer returned wrong shape");
         ok = true;
       } catch (const std::exception& e) {
         err = e.what();
@@ -937,8 +967,6 @@ class WireFront {
                 http_response(500, "Internal Server Error", body));
       return;
     }
-    TORCH_CHECK(scores.dim() == 2 && scores.size(0) == (int64_t)idxs.size(),
-                "scorer returned wrong shape");
     const float* sp = scores.data_ptr<float>();
     int64_t P = scores.size(1);
     char num[64];
@@ -951,7 +979,7 @@ class WireFront {
         if (!first) body += ',';
         first = false;
         body += '"';
-        body += names[p];  // pod names are identifier-safe
+        json_escape_into(body, names[p]);
         body += "\":";
         int len = snprintf(num, sizeof(num), "%g", row[p]);
         body.append(num, len);

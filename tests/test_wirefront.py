@@ -324,3 +324,51 @@ class TestWireFrontRobustness:
         status, body, _ = _read_response(s)
         assert status == 200 and json.loads(body)["scores"]["pod-a"] == 8.0
         s.close()
+
+    def test_hostile_pod_name_escaped(self, service):
+        """A pod identifier containing JSON metacharacters (it arrives
+        from the external event stream) must not break responses."""
+        from llmd_kvcache_amd.kvblock.keys import Key, PodEntry
+        from llmd_kvcache_amd.kvblock.token_processor import (
+            ChunkedTokenDatabase, TokenProcessorConfig)
+
+        svc, port, tokens = service
+        idx = svc.indexer.kv_block_index()
+        tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=BS))
+        evil = 'pod"\\inject\n'
+        hostile_tokens = [77000 + i for i in range(8)]
+        keys = tp.tokens_to_kv_block_keys(None, hostile_tokens, MODEL)
+        idx.add(keys, keys, [PodEntry(evil, "gpu")])
+        s = _connect(port)
+        s.sendall(_http_post("/score", {"model": MODEL,
+                                        "tokens": hostile_tokens}))
+        status, body, _ = _read_response(s)
+        assert status == 200
+        scores = json.loads(body)["scores"]  # must parse cleanly
+        assert scores == {evil: 2.0}
+        s.close()
+
+    def test_bad_scorer_return_is_500_not_crash(self, service):
+        """A callback returning garbage must produce 500s and leave the
+        server serving (validation inside the batcher's try)."""
+        svc, port, tokens = service
+        orig = svc._score_tokens_cb
+        svc._front.stop()  # restart front with a broken callback
+        from llmd_kvcache_amd.ops import cpu_ext
+
+        ops = cpu_ext.require()
+        svc._front = ops.WireFront(lambda *a: ("nonsense", []),
+                                   svc._score_text_cb, 4096, 2)
+        port2 = svc._front.start("", 0, 2)
+        s = _connect(port2)
+        s.sendall(_http_post("/score", {"model": MODEL, "tokens": tokens}))
+        status, body, _ = _read_response(s)
+        assert status == 500
+        # still alive
+        s.sendall(b"GET /health HTTP/1.1\r\nhost: x\r\n\r\n")
+        status, body, _ = _read_response(s)
+        assert status == 200
+        s.close()
+        svc._front.stop()
+        svc._front = ops.WireFront(orig, svc._score_text_cb, 4096, 2)
+        svc._running = False
