@@ -435,3 +435,53 @@ def test_asan_harness_clean():
                              timeout=120)
         assert run.returncode == 0, run.stderr[-2000:]
         assert "asan_check OK" in run.stdout
+
+
+def test_long_mixed_differential_vs_python_reference():
+    """30k randomized add/evict/lookup ops: the C++ table matches the
+    pure-Python InMemoryIndex exactly while per-key pod sets stay within
+    pods_per_key (beyond that only the documented overflow-victim policy
+    differs - docs/architecture.md divergences table)."""
+    import random
+
+    from llmd_kvcache_amd.kvblock.gpu_index import (NativeIndex,
+                                                    TableIndexConfig)
+    from llmd_kvcache_amd.kvblock.in_memory import (InMemoryIndex,
+                                                    InMemoryIndexConfig)
+    from llmd_kvcache_amd.kvblock.keys import Key, PodEntry
+
+    rng = random.Random(2024)
+    nat = NativeIndex(TableIndexConfig(capacity=1 << 14, pods_per_key=8))
+    ref = InMemoryIndex(InMemoryIndexConfig(size=100000, pod_cache_size=8))
+    for step in range(30000):
+        op = rng.random()
+        if op < 0.55:
+            n = rng.randrange(1, 6)
+            base = rng.randrange(4000)
+            eks = [Key("m", 100000 + base + i) for i in range(n)]
+            rks = [Key("m", 200000 + base + i) for i in range(n)]
+            # 4 pods x 2 tiers = 8 distinct entries -> never overflows
+            ent = [PodEntry(f"p{rng.randrange(4)}",
+                            rng.choice(["gpu", "cpu"]))]
+            nat.add(eks, rks, ent)
+            ref.add(eks, rks, ent)
+        elif op < 0.75:
+            ek = Key("m", 100000 + rng.randrange(4000))
+            ent = [PodEntry(f"p{rng.randrange(4)}",
+                            rng.choice(["gpu", "cpu"]))]
+            for idx in (nat, ref):
+                try:
+                    idx.evict(ek, ent)
+                except Exception:
+                    pass
+        else:
+            base = rng.randrange(4000)
+            n = rng.randrange(1, 8)
+            q = [Key("m", 200000 + base + i) for i in range(n)]
+            na = {k: sorted((e.pod_identifier, e.device_tier)
+                            for e in v)
+                  for k, v in nat.lookup(q, set()).items()}
+            nb = {k: sorted((e.pod_identifier, e.device_tier)
+                            for e in v)
+                  for k, v in ref.lookup(q, set()).items()}
+            assert na == nb, (step, q)
