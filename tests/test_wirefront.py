@@ -372,3 +372,40 @@ class TestWireFrontRobustness:
         svc._front.stop()
         svc._front = ops.WireFront(orig, svc._score_text_cb, 4096, 2)
         svc._running = False
+
+
+class TestWireParserDifferential:
+    def test_random_requests_match_direct_scoring(self, service):
+        """Randomized JSON bodies (weird-but-valid: unicode escapes,
+        extra unknown keys, whitespace, negative/huge ints in unknown
+        fields) through the C++ parser produce exactly the scores the
+        direct Python path computes."""
+        import random
+
+        svc, port, _ = service
+        rng = random.Random(8)
+        s = _connect(port)
+        rest = b""
+        for trial in range(40):
+            tokens = [rng.randrange(0, 1 << 31)
+                      for _ in range(rng.randrange(0, 40))]
+            obj = {"model": MODEL, "tokens": tokens}
+            if rng.random() < 0.4:
+                obj["pods"] = ["pod-a", "pod-b"][: rng.randrange(3)]
+            if rng.random() < 0.5:
+                obj["ignored_" + str(trial)] = rng.choice(
+                    [None, True, -12345678901234, {"x": [1, "é"]},
+                     "mixed ☃ text\n"])
+            body = json.dumps(obj, ensure_ascii=rng.random() < 0.5,
+                              indent=rng.choice([None, 1]))
+            s.sendall((f"POST /score HTTP/1.1\r\nhost: x\r\n"
+                       f"content-length: {len(body.encode())}\r\n\r\n"
+                       ).encode() + body.encode())
+            status, rbody, rest = _read_response(s, rest)
+            assert status == 200, body
+            got = json.loads(rbody)["scores"]
+            want = svc.indexer.score_tokens(
+                tokens, MODEL, obj.get("pods", []))
+            want = {k: v for k, v in want.items() if v}
+            assert got == want, (trial, body)
+        s.close()
