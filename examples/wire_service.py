@@ -9,6 +9,9 @@ never touch Python, and micro-batches run one fused kernel each
 Environment:
   WIRE_PORT              (default 8080) scoring endpoint port
   WIRE_IO_THREADS        (default 4)
+  METRICS_PORT           optional: also serve the full HTTP surface
+                         (/metrics, /score_batch, chat endpoints) on
+                         this port via the stdlib service
   ZMQ_ENDPOINT           (default tcp://*:5557)
   ZMQ_TOPIC              (default kv@)
   POOL_CONCURRENCY       (default 4)
@@ -90,6 +93,15 @@ def main() -> None:
                      n_io=int(os.environ.get("WIRE_IO_THREADS", "4")))
     logger.info("wirefront serving on :%d", port)
 
+    http = None
+    metrics_port = int(os.environ.get("METRICS_PORT", "0"))
+    if metrics_port:
+        from llmd_kvcache_amd.service.http_server import HttpService
+
+        http = HttpService(indexer, host="0.0.0.0", port=metrics_port)
+        http.start()
+        logger.info("metrics/full HTTP surface on :%d", http.port)
+
     stop = {"flag": False}
 
     def on_term(signum, frame):
@@ -102,6 +114,8 @@ def main() -> None:
             time.sleep(0.5)
     finally:
         svc.stop()
+        if http is not None:
+            http.stop()
         events.shutdown()
         indexer.shutdown()
         reqs, batches = svc.stats()
