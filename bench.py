@@ -91,9 +91,18 @@ def populate_index(index, device, rank):
     engine_hash = 1
     batch = []
     batches = []
+    ev_tokens = BLOCKS_PER_EVENT * BLOCK_SIZE
+    tokens_block = None
     for e in range(n_events):
-        tokens = rng.integers(0, VOCAB, size=BLOCKS_PER_EVENT * BLOCK_SIZE,
-                              dtype=np.int64)
+        # one rng draw per BATCH of events (a per-event draw costs real
+        # CPU when 8 ranks generate concurrently on one host)
+        j = e % events_per_batch
+        if j == 0:
+            n_in = min(events_per_batch, n_events - e)
+            tokens_block = rng.integers(0, VOCAB,
+                                        size=(n_in, ev_tokens),
+                                        dtype=np.int64)
+        tokens = tokens_block[j]
         hashes = np.arange(engine_hash, engine_hash + BLOCKS_PER_EVENT,
                            dtype=np.uint64)
         engine_hash += BLOCKS_PER_EVENT
