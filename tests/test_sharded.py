@@ -451,3 +451,32 @@ def test_reduce_scatter_error_flag():
     results = run_distributed("_body_reduce_scatter_follower_failure")
     assert results[0] == ("peer", True)
     assert results[1] == ("local", True)
+
+
+def test_partition_prompts_properties():
+    """Unit properties of the reduce_scatter prompt partitioner:
+    monotone bounds, full coverage, deterministic, reasonable balance."""
+    import random
+
+    from llmd_kvcache_amd.parallel.sharded import _partition_prompts
+
+    rng = random.Random(5)
+    for _ in range(200):
+        world = rng.choice([2, 3, 4, 8])
+        B = rng.randrange(world, 200)
+        counts = [rng.randrange(0, 64) for _ in range(B)]
+        offs = [0]
+        for c in counts:
+            offs.append(offs[-1] + c)
+        offs_t = torch.tensor(offs, dtype=torch.int32)
+        bounds = _partition_prompts(offs_t, world)
+        assert bounds[0] == 0 and bounds[-1] == B
+        assert all(bounds[i] <= bounds[i + 1] for i in range(world))
+        assert bounds == _partition_prompts(offs_t, world)  # deterministic
+        total = offs[-1]
+        if total:
+            # no rank should carry more than ~(1/world + one prompt)'s
+            # worth of keys beyond the ideal share
+            max_keys = max(offs[bounds[r + 1]] - offs[bounds[r]]
+                           for r in range(world))
+            assert max_keys <= total // world + max(counts)
