@@ -614,8 +614,8 @@ class GpuIndex(TableIndex):
         ev_type: List[int] = []
         pod_entry: List[int] = []
         grp_off = [0]
-        model_ids: Set[int] = set()
-        model_id = 0
+        model_of: List[int] = []  # model id per event (mixed-model
+        # batches apply in ONE launch; the kernels read model_of[e])
 
         by_pod: Dict[Tuple[str, str], list] = {}
         order: List[Tuple[str, str]] = []
@@ -628,7 +628,6 @@ class GpuIndex(TableIndex):
 
         for pod, model in order:
             model_id = self.registry.model_id(model)
-            model_ids.add(model_id)
             pod_id = self.registry.pod_id(pod)
             n_events_in_group = 0
             for ev in by_pod[(pod, model)]:
@@ -665,6 +664,7 @@ class GpuIndex(TableIndex):
                     parents.append(par)
                     has_parent.append(hp)
                     ev_type.append(0)
+                    model_of.append(model_id)
                     pod_entry.append((tier << 24) | (pod_id + 1))
                     n_events_in_group += 1
                 elif isinstance(ev, BlockRemoved):
@@ -683,6 +683,7 @@ class GpuIndex(TableIndex):
                     parents.append(0)
                     has_parent.append(0)
                     ev_type.append(1)
+                    model_of.append(model_id)
                     pod_entry.append((tier << 24) | (pod_id + 1))
                     n_events_in_group += 1
                 elif isinstance(ev, AllBlocksCleared):
@@ -691,15 +692,6 @@ class GpuIndex(TableIndex):
                 grp_off.append(grp_off[-1] + n_events_in_group)
 
         if len(grp_off) == 1:
-            return
-        if len(model_ids) > 1:
-            # mixed-model batch: split and recurse per model (the kernel
-            # takes one model id per launch)
-            by_model: Dict[str, list] = {}
-            for pod, model, events in batches:
-                by_model.setdefault(model, []).append((pod, model, events))
-            for group in by_model.values():
-                self.apply_event_batches(group, token_processor)
             return
 
         d = self.device
@@ -768,6 +760,7 @@ class GpuIndex(TableIndex):
                 up.up(ev_type, "evt", d, dtype=np.uint8),
                 up.up(pod_entry, "pe", d, dtype=np.int32),
                 up.up(grp_off, "grp", d, dtype=np.int32),
+                up.up(model_of, "mdl", d, dtype=np.int32),
             )
         else:
             i32 = torch.int32
@@ -781,6 +774,7 @@ class GpuIndex(TableIndex):
                 torch.tensor(ev_type, dtype=torch.uint8, device=d),
                 torch.tensor(pod_entry, dtype=i32, device=d),
                 torch.tensor(grp_off, dtype=i32, device=d),
+                torch.tensor(model_of, dtype=i32, device=d),
             )
         if use_split:
             counts = np.diff(np.asarray(eh_off, dtype=np.int64))
@@ -790,21 +784,21 @@ class GpuIndex(TableIndex):
             if use_tr:
                 self.table.ops.gpu_apply_events_split_tr(
                     *self.table._t(), tok_t, *common[:7],
-                    ev_of_t, model_id, _to_i64(init_hash), block_size,
+                    ev_of_t, common[8], _to_i64(init_hash), block_size,
                     self.table.next_epoch(), self.cfg.shard_id,
                     self.cfg.num_shards, max_tokens,
                 )
             else:
                 self.table.ops.gpu_apply_events_split(
-                    *self.table._t(), tok_t, *common, ev_of_t,
-                    model_id, _to_i64(init_hash), block_size,
+                    *self.table._t(), tok_t, *common[:8], ev_of_t,
+                    common[8], _to_i64(init_hash), block_size,
                     self.table.next_epoch(), self.cfg.shard_id,
                     self.cfg.num_shards,
                 )
         else:
             self.table.ops.gpu_apply_events(
                 *self.table._t(), tok_t, *common,
-                model_id, _to_i64(init_hash), block_size,
+                _to_i64(init_hash), block_size,
                 self.table.next_epoch(), self.cfg.shard_id,
                 self.cfg.num_shards,
             )

@@ -173,21 +173,18 @@ class EventsPool:
             except DecodeError as e:
                 logger.debug("dropping poison-pill message: %s", e)
                 continue
-            # split by model: the on-device kernel takes one model per call
             batches.append((msg.pod_identifier, msg.model_name, batch.events))
         if not batches:
             return
-        by_model = {}
-        for pod, model, events in batches:
-            by_model.setdefault(model, []).append((pod, model, events))
-        for model, group in by_model.items():
-            try:
-                self.index.apply_event_batches(group, self.token_processor)
-            except Exception:
-                logger.exception("on-device event application failed; "
-                                 "falling back to per-event path")
-                for pod, model_name, events in group:
-                    self.digest_events(pod, model_name, events)
+        # mixed-model bursts apply in ONE launch: the kernels take a
+        # per-event model id (model_of), so no host-side model split
+        try:
+            self.index.apply_event_batches(batches, self.token_processor)
+        except Exception:
+            logger.exception("on-device event application failed; "
+                             "falling back to per-event path")
+            for pod, model_name, events in batches:
+                self.digest_events(pod, model_name, events)
 
     def process_event(self, msg: Message) -> None:
         try:

@@ -72,19 +72,20 @@ at::Tensor gpu_hash_chain_tr(at::Tensor, at::Tensor, at::Tensor, int64_t,
 void gpu_apply_events(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, int64_t, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, at::Tensor,
-                      at::Tensor, at::Tensor, at::Tensor, at::Tensor, int64_t,
-                      int64_t, int64_t, int64_t, int64_t, int64_t);
+                      at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                      at::Tensor, int64_t, int64_t, int64_t, int64_t,
+                      int64_t);
 void gpu_apply_events_split(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                             at::Tensor, at::Tensor, at::Tensor, int64_t,
                             at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                             at::Tensor, at::Tensor, at::Tensor, at::Tensor,
-                            at::Tensor, at::Tensor, int64_t, int64_t, int64_t,
-                            int64_t, int64_t, int64_t);
+                            at::Tensor, at::Tensor, at::Tensor, int64_t,
+                            int64_t, int64_t, int64_t, int64_t);
 void gpu_apply_events_split_tr(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                             at::Tensor, at::Tensor, at::Tensor, int64_t,
                             at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                             at::Tensor, at::Tensor, at::Tensor, at::Tensor,
-                            at::Tensor, int64_t, int64_t, int64_t,
+                            at::Tensor, at::Tensor, int64_t, int64_t,
                             int64_t, int64_t, int64_t, int64_t);
 #endif
 

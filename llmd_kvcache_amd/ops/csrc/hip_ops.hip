@@ -577,7 +577,8 @@ __global__ void k_apply_events(
     const uint8_t* __restrict__ ev_type,     // [E] 0=stored 1=removed
     const uint32_t* __restrict__ pod_entry,  // [E]
     const int32_t* __restrict__ grp_off,     // [G+1] event groups by pod
-    int64_t G, uint32_t model, uint64_t init_hash, int block_size,
+    const int32_t* __restrict__ model_of,    // [E] model id per event
+    int64_t G, uint64_t init_hash, int block_size,
     int32_t epoch, int shard_id, int num_shards,
     uint64_t* __restrict__ req_scratch) {    // [total engine hashes]
   const int g = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
@@ -589,6 +590,7 @@ __global__ void k_apply_events(
   for (int e = e_begin; e < e_end; ++e) {
     const int nh = eh_off[e + 1] - eh_off[e];
     const uint64_t* eh = ehashes + eh_off[e];
+    const uint32_t model = (uint32_t)model_of[e];
     if (ev_type[e] == 1) {  // BlockRemoved
       for (int i = lane; i < nh; i += 64) {
         int64_t ei = dev_emap_find(v, eh[i], model);
@@ -700,8 +702,8 @@ __global__ void k_event_chains(
     const int32_t* __restrict__ tok_off, const uint64_t* __restrict__ ehashes,
     const int32_t* __restrict__ eh_off, const uint64_t* __restrict__ parents,
     const uint8_t* __restrict__ has_parent, const uint8_t* __restrict__ ev_type,
-    const int32_t* __restrict__ grp_off, int64_t G, uint32_t model,
-    uint64_t init_hash, int block_size,
+    const int32_t* __restrict__ grp_off, const int32_t* __restrict__ model_of,
+    int64_t G, uint64_t init_hash, int block_size,
     const uint64_t* __restrict__ bkeys, const int32_t* __restrict__ bvals,
     int64_t bmask, uint64_t* __restrict__ req_scratch) {
   int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -723,7 +725,8 @@ __global__ void k_event_chains(
       if (bi >= grp_first_block && bi < eh_off[e]) {
         parent = req_scratch[bi];  // earlier event of this group
       } else {
-        int64_t ei = dev_emap_find(v, parents[e], model);
+        int64_t ei = dev_emap_find(v, parents[e],
+                                   (uint32_t)model_of[e]);
         if (ei >= 0) parent = v.e_vals[ei];
       }
     }
@@ -766,8 +769,8 @@ __global__ void k_event_chains_ev(
     const int32_t* __restrict__ tok_off, const uint64_t* __restrict__ ehashes,
     const int32_t* __restrict__ eh_off, const uint64_t* __restrict__ parents,
     const uint8_t* __restrict__ has_parent, const uint8_t* __restrict__ ev_type,
-    int64_t E, uint32_t model, uint64_t init_hash, int block_size,
-    uint64_t* __restrict__ req_scratch) {
+    const int32_t* __restrict__ model_of, int64_t E, uint64_t init_hash,
+    int block_size, uint64_t* __restrict__ req_scratch) {
   const int64_t e = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (e >= E) return;
   if (ev_type[e] == 1) return;  // removals have no chain
@@ -777,7 +780,7 @@ __global__ void k_event_chains_ev(
   if (n_chunks != nh) return;  // drop mismatched event (see k_apply_events)
   uint64_t parent = init_hash;
   if (has_parent[e]) {
-    int64_t ei = dev_emap_find(v, parents[e], model);
+    int64_t ei = dev_emap_find(v, parents[e], (uint32_t)model_of[e]);
     if (ei >= 0) parent = v.e_vals[ei];
   }
   uint64_t h = parent;
@@ -795,12 +798,14 @@ __global__ void k_event_inserts(
     DevTable v, const uint64_t* __restrict__ ehashes,
     const int32_t* __restrict__ eh_off, const int32_t* __restrict__ ev_of,
     const uint8_t* __restrict__ ev_type, const int32_t* __restrict__ tok_off,
-    const uint32_t* __restrict__ pod_entry, int64_t n_blocks, uint32_t model,
+    const uint32_t* __restrict__ pod_entry,
+    const int32_t* __restrict__ model_of, int64_t n_blocks,
     int block_size, int32_t epoch, int shard_id, int num_shards,
     const uint64_t* __restrict__ req_scratch) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n_blocks) return;
   const int e = ev_of[i];
+  const uint32_t model = (uint32_t)model_of[e];
   if (ev_type[e] == 1) {  // BlockRemoved
     int64_t ei = dev_emap_find(v, ehashes[i], model);
     if (ei < 0) return;
@@ -1121,13 +1126,15 @@ void gpu_apply_events(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
                       at::Tensor ehashes, at::Tensor eh_off,
                       at::Tensor parents, at::Tensor has_parent,
                       at::Tensor ev_type, at::Tensor pod_entry,
-                      at::Tensor grp_off, int64_t model_id,
+                      at::Tensor grp_off, at::Tensor model_of,
                       int64_t init_hash_bits, int64_t block_size,
                       int64_t epoch, int64_t shard_id, int64_t num_shards) {
   auto v = dev_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
                     pods_per_key);
   int64_t G = grp_off.numel() - 1;
   if (G == 0) return;
+  TORCH_CHECK(model_of.numel() == ev_type.numel(),
+              "model_of must have one id per event");
   auto req_scratch = at::empty({std::max<int64_t>(ehashes.numel(), 1)},
                                ehashes.options());
   const int waves_per_block = 4;
@@ -1138,7 +1145,7 @@ void gpu_apply_events(at::Tensor keys, at::Tensor meta, at::Tensor stamp,
       eh_off.data_ptr<int32_t>(), U64P(parents),
       has_parent.data_ptr<uint8_t>(), ev_type.data_ptr<uint8_t>(),
       reinterpret_cast<const uint32_t*>(pod_entry.data_ptr<int32_t>()),
-      grp_off.data_ptr<int32_t>(), G, (uint32_t)model_id,
+      grp_off.data_ptr<int32_t>(), model_of.data_ptr<int32_t>(), G,
       (uint64_t)init_hash_bits, (int)block_size, (int32_t)epoch,
       (int)shard_id, (int)num_shards,
       reinterpret_cast<uint64_t*>(req_scratch.data_ptr<int64_t>()));
@@ -1154,7 +1161,7 @@ void gpu_apply_events_split_tr(
     int64_t pods_per_key, at::Tensor tokens_flat, at::Tensor tok_off,
     at::Tensor ehashes, at::Tensor eh_off, at::Tensor parents,
     at::Tensor has_parent, at::Tensor ev_type, at::Tensor pod_entry,
-    at::Tensor ev_of, int64_t model_id, int64_t init_hash_bits,
+    at::Tensor ev_of, at::Tensor model_of, int64_t init_hash_bits,
     int64_t block_size, int64_t epoch, int64_t shard_id,
     int64_t num_shards, int64_t max_tokens) {
   auto v = dev_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
@@ -1179,13 +1186,15 @@ void gpu_apply_events_split_tr(
                        tok_off.data_ptr<int32_t>(), E,
                        tokens_t.data_ptr<int32_t>());
   }
+  TORCH_CHECK(model_of.numel() == E, "model_of must have one id per event");
   hipLaunchKernelGGL(k_event_chains_ev,
                      dim3((int)((E + threads - 1) / threads)), dim3(threads),
                      0, STREAM, v, tokens_t.data_ptr<int32_t>(),
                      tok_off.data_ptr<int32_t>(), U64P(ehashes),
                      eh_off.data_ptr<int32_t>(), U64P(parents),
                      has_parent.data_ptr<uint8_t>(),
-                     ev_type.data_ptr<uint8_t>(), E, (uint32_t)model_id,
+                     ev_type.data_ptr<uint8_t>(),
+                     model_of.data_ptr<int32_t>(), E,
                      (uint64_t)init_hash_bits, (int)block_size,
                      reinterpret_cast<uint64_t*>(
                          req_scratch.data_ptr<int64_t>()));
@@ -1196,7 +1205,8 @@ void gpu_apply_events_split_tr(
                      tok_off.data_ptr<int32_t>(),
                      reinterpret_cast<const uint32_t*>(
                          pod_entry.data_ptr<int32_t>()),
-                     n, (uint32_t)model_id, (int)block_size, (int32_t)epoch,
+                     model_of.data_ptr<int32_t>(), n,
+                     (int)block_size, (int32_t)epoch,
                      (int)shard_id, (int)num_shards,
                      reinterpret_cast<uint64_t*>(
                          req_scratch.data_ptr<int64_t>()));
@@ -1210,7 +1220,7 @@ void gpu_apply_events_split(
     int64_t pods_per_key, at::Tensor tokens, at::Tensor tok_off,
     at::Tensor ehashes, at::Tensor eh_off, at::Tensor parents,
     at::Tensor has_parent, at::Tensor ev_type, at::Tensor pod_entry,
-    at::Tensor grp_off, at::Tensor ev_of, int64_t model_id,
+    at::Tensor grp_off, at::Tensor ev_of, at::Tensor model_of,
     int64_t init_hash_bits, int64_t block_size, int64_t epoch,
     int64_t shard_id, int64_t num_shards) {
   auto v = dev_view(keys, meta, stamp, pods, e_keys, e_meta, e_vals,
@@ -1230,6 +1240,8 @@ void gpu_apply_events_split(
                      0, STREAM, U64P(ehashes), n,
                      reinterpret_cast<uint64_t*>(bkeys.data_ptr<int64_t>()),
                      bvals.data_ptr<int32_t>(), bcap - 1);
+  TORCH_CHECK(model_of.numel() == ev_type.numel(),
+              "model_of must have one id per event");
   hipLaunchKernelGGL(k_event_chains,
                      dim3((int)((G + threads - 1) / threads)), dim3(threads),
                      0, STREAM, v, tokens.data_ptr<int64_t>(),
@@ -1237,7 +1249,8 @@ void gpu_apply_events_split(
                      eh_off.data_ptr<int32_t>(), U64P(parents),
                      has_parent.data_ptr<uint8_t>(),
                      ev_type.data_ptr<uint8_t>(), grp_off.data_ptr<int32_t>(),
-                     G, (uint32_t)model_id, (uint64_t)init_hash_bits,
+                     model_of.data_ptr<int32_t>(),
+                     G, (uint64_t)init_hash_bits,
                      (int)block_size,
                      reinterpret_cast<uint64_t*>(bkeys.data_ptr<int64_t>()),
                      bvals.data_ptr<int32_t>(), bcap - 1,
@@ -1250,7 +1263,8 @@ void gpu_apply_events_split(
                      tok_off.data_ptr<int32_t>(),
                      reinterpret_cast<const uint32_t*>(
                          pod_entry.data_ptr<int32_t>()),
-                     n, (uint32_t)model_id, (int)block_size, (int32_t)epoch,
+                     model_of.data_ptr<int32_t>(), n,
+                     (int)block_size, (int32_t)epoch,
                      (int)shard_id, (int)num_shards,
                      reinterpret_cast<uint64_t*>(
                          req_scratch.data_ptr<int64_t>()));

@@ -130,10 +130,10 @@ class TestNestedNoneOverlay:
 
 
 class TestMultiModelBurst:
-    def test_mixed_model_batches_split(self):
-        """The pool's GPU burst path hands apply_event_batches
-        single-model groups; gpu_index additionally self-splits mixed
-        batches (one model id per kernel launch)."""
+    def test_mixed_model_burst_single_apply(self):
+        """Round 2: the event kernels take a per-event model id
+        (model_of), so a mixed-model burst reaches apply_event_batches
+        as ONE call - no host-side model split, one launch set."""
         from llmd_kvcache_amd.kvevents.events import BlockStored, EventBatch
         from llmd_kvcache_amd.kvevents.pool import (
             EventsConfig,
@@ -153,11 +153,10 @@ class TestMultiModelBurst:
         pool.start(with_subscriber=False)
         pool.drain()
         pool.shutdown()
-        models_seen = [
-            {m for _, m, _ in call} for call in index.applied
-        ]
-        for mset in models_seen:
-            assert len(mset) == 1  # each apply call is single-model
+        # all three messages (two models) landed in one apply call
+        assert len(index.applied) == 1
+        assert {m for _, m, _ in index.applied[0]} == {"model-a",
+                                                       "model-b"}
 
 
 class TestTieredConfig:

@@ -316,6 +316,58 @@ class TestApplyEventsKernel:
                         gpu_idx.get_request_key(ck)), ck
 
 
+    def test_mixed_model_single_launch_matches_cpu(self):
+        """Two models in ONE apply call (round-2 per-event model ids):
+        state must equal the CPU digest, and model scoping must hold
+        (same token chains under different models never alias)."""
+        from llmd_kvcache_amd.kvevents.pool import digest_events
+
+        rng = random.Random(31)
+        tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=4))
+        cpu_idx = NativeIndex(TableIndexConfig(capacity=1 << 12,
+                                               pods_per_key=10))
+        gpu_idx = GpuIndex(GpuIndexConfig(capacity=1 << 12, pods_per_key=10))
+
+        batches = []
+        next_hash = 9000
+        shared_tokens = list(range(16))  # same tokens under BOTH models
+        for model in ("model-a", "model-b"):
+            hs = list(range(next_hash, next_hash + 4))
+            next_hash += 4
+            batches.append((f"pod-{model}", model,
+                            [BlockStored(hs, None, shared_tokens, 4)]))
+        for i in range(20):
+            model = rng.choice(["model-a", "model-b"])
+            toks = [rng.randrange(1 << 30) for _ in range(rng.randrange(1, 5) * 4)]
+            hs = list(range(next_hash, next_hash + len(toks) // 4))
+            next_hash += len(toks) // 4
+            batches.append((f"pod-{rng.randrange(4)}", model,
+                            [BlockStored(hs, None, toks, 4)]))
+
+        for pod, model, events in batches:
+            digest_events(cpu_idx, tp, pod, model, events)
+        gpu_idx.apply_event_batches(batches, tp)  # ONE call, two models
+        torch.cuda.synchronize()
+
+        for pod, model, events in batches:
+            ev = events[0]
+            req = tp.tokens_to_kv_block_keys(None, ev.token_ids, model)
+            c = cpu_idx.lookup(req, set())
+            g = gpu_idx.lookup(req, set())
+            assert {k: set(v) for k, v in c.items()} == \
+                {k: set(v) for k, v in g.items()}, (pod, model)
+        # model scoping: identical chains under different models resolve
+        # to their own pods only
+        ka = tp.tokens_to_kv_block_keys(None, shared_tokens, "model-a")
+        kb = tp.tokens_to_kv_block_keys(None, shared_tokens, "model-b")
+        ga = gpu_idx.lookup(ka, set())
+        gb = gpu_idx.lookup(kb, set())
+        assert {e.pod_identifier for v in ga.values() for e in v} == \
+            {"pod-model-a"}
+        assert {e.pod_identifier for v in gb.values() for e in v} == \
+            {"pod-model-b"}
+
+
 class TestConcurrentGpuInserts:
     def test_many_duplicate_inserts_converge(self):
         """Thousands of threads inserting the same keys concurrently must
