@@ -24,7 +24,7 @@ equivalent with the request path in C++ instead of Go.
 
 from __future__ import annotations
 
-from typing import List, Optional, Sequence, Tuple
+from typing import List, Sequence, Tuple
 
 from ..indexer import Indexer
 
