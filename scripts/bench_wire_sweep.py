@@ -15,7 +15,13 @@ from llmd_kvcache_amd.service.wirefront import WireIndexerService
 
 
 def main():
+    import os
+
     assert torch.cuda.is_available()
+    # KVIDX_SWEEP_TOKENS overrides the request length (A/B: is the wire
+    # ceiling the per-request CPU chain work - 512 chunks at 8192 - or
+    # the parse/socket path?)
+    bench.PROMPT_TOKENS = int(os.environ.get("KVIDX_SWEEP_TOKENS", "8192"))
     bench.NUM_BLOCKS = 1 << 18  # 256k blocks: faster populate, same probe shape
     idx = GpuIndex(GpuIndexConfig(capacity=1 << 20, pods_per_key=10))
     from llmd_kvcache_amd.kvblock.gpu_index import GpuIndex as _G
