@@ -28,14 +28,19 @@ def main():
     index = InMemoryIndex(InMemoryIndexConfig(size=100_000, pod_cache_size=10))
     tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=block_size))
     pool = EventsPool(
-        EventsConfig(zmq_endpoint="tcp://127.0.0.1:5557", concurrency=4),
+        EventsConfig(zmq_endpoint="tcp://127.0.0.1:0", concurrency=4),
         index,
         tp,
     )
     pool.start(with_subscriber=True)
-    time.sleep(0.5)  # let the SUB bind
+    # ephemeral port: wait for the SUB to bind and read the port back
+    # (a fixed port collides with concurrent runs/tests)
+    deadline = time.monotonic() + 10
+    while pool._subscriber.port is None and time.monotonic() < deadline:
+        time.sleep(0.05)
+    assert pool._subscriber.port, "subscriber did not bind"
 
-    pub = Publisher("tcp://127.0.0.1:5557")
+    pub = Publisher(f"tcp://127.0.0.1:{pool._subscriber.port}")
     assert pub.wait_ready(), "subscriber did not come up"
     print("publishing synthetic vLLM KV events for pod-0 ...")
     simulate_events(pub, "pod-0", "demo-model", n_batches=20,

@@ -9,6 +9,17 @@ import sys
 
 import pytest
 
+
+def _free_port():
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 TINY = ["--device", "cpu", "--batch", "32", "--blocks", "2048",
         "--steps", "2", "--warmup", "1", "--calls-per-step", "1"]
@@ -20,9 +31,10 @@ REQUIRED_CONFIG = {"model", "global_batch", "seq_len", "parallelism"}
 
 
 def run_bench(extra):
+    env = dict(os.environ, MASTER_PORT=str(_free_port()))
     proc = subprocess.run(
         [sys.executable, os.path.join(ROOT, "bench.py")] + TINY + extra,
-        capture_output=True, text=True, timeout=300, cwd=ROOT)
+        capture_output=True, text=True, timeout=300, cwd=ROOT, env=env)
     assert proc.returncode == 0, proc.stderr[-800:]
     line = proc.stdout.strip().splitlines()[-1]
     return json.loads(line)
@@ -74,12 +86,12 @@ class TestBenchTorchrun:
         return json.loads(json_lines[0])
 
     def test_replicated_world2(self):
-        d = self.run_torchrun([], 29541)
+        d = self.run_torchrun([], _free_port())
         assert d["config"]["parallelism"] == "replicated2"
         assert d["scaling"] == "weak"
         assert d["config"]["global_batch"] == 32  # whole-job aggregate
 
     def test_sharded_world2(self):
-        d = self.run_torchrun(["--sharded"], 29542)
+        d = self.run_torchrun(["--sharded"], _free_port())
         assert d["config"]["parallelism"] == "shard2"
         assert d["scaling"] == "strong"
