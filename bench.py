@@ -490,7 +490,9 @@ def main():
     global NUM_BLOCKS, NUM_PODS
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--steps", type=int, default=25,
+                    help="timed steps; the default gives a >=1s timed "
+                         "region at the default batch/calls")
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--batch", type=int, default=32768,
                     help="prompts per scoring call (sub-batch). The chain "
