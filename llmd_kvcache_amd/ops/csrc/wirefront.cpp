@@ -448,6 +448,7 @@ struct Conn {
   std::map<uint64_t, std::string> ready;  // completed out-of-order
   std::atomic<bool> closed{false};
   bool want_writable = false;
+  bool sent_100 = false;  // io thread only
 };
 
 using ConnPtr = std::shared_ptr<Conn>;
@@ -733,8 +734,9 @@ class WireFront {
       if (sp1 == std::string::npos || sp2 == std::string::npos) return false;
       std::string method = line.substr(0, sp1);
       std::string path = line.substr(sp1 + 1, sp2 - sp1 - 1);
-      // content-length
+      // content-length (+ curl-style Expect: 100-continue)
       size_t clen = 0;
+      bool expect_100 = false;
       {
         const char* h = c->inbuf.c_str() + line_end + 2;
         const char* hend = c->inbuf.c_str() + hdr_end + 2;
@@ -743,12 +745,35 @@ class WireFront {
           if (!eol || eol > hend) break;
           if ((eol - h) > 15 && strncasecmp(h, "content-length:", 15) == 0) {
             clen = (size_t)strtoull(h + 15, nullptr, 10);
+          } else if ((eol - h) > 7 &&
+                     strncasecmp(h, "expect:", 7) == 0 &&
+                     strcasestr(std::string(h, eol - h).c_str(),
+                                "100-continue")) {
+            expect_100 = true;
           }
           h = eol + 2;
         }
       }
       size_t total = hdr_end + 4 + clen;
-      if (c->inbuf.size() < total) return true;  // body incomplete
+      if (c->inbuf.size() < total) {
+        // interim 100 unblocks curl-style clients that wait before
+        // sending the body; only safe to write directly when no earlier
+        // response can still be pending (otherwise order would break -
+        // the client's 1s fallback covers that rare mix)
+        if (expect_100 && !c->sent_100) {
+          bool idle;
+          {
+            std::lock_guard<std::mutex> lk(c->mu);
+            idle = c->ready.empty() && c->flushed_slot == c->next_slot;
+          }
+          if (idle && c->outbuf.empty()) {
+            c->outbuf += "HTTP/1.1 100 Continue\r\n\r\n";
+            c->sent_100 = true;
+          }
+        }
+        return true;  // body incomplete
+      }
+      c->sent_100 = false;
 
       if (method == "GET" && path == "/health") {
         enqueue_inline(conn, http_response(200, "OK", "{\"status\":\"ok\"}"));

@@ -409,3 +409,24 @@ class TestWireParserDifferential:
             want = {k: v for k, v in want.items() if v}
             assert got == want, (trial, body)
         s.close()
+
+    def test_expect_100_continue(self, service):
+        """curl-style two-phase POST: headers with Expect: 100-continue
+        first, body after the interim 100."""
+        import time as _time
+
+        _, port, tokens = service
+        body = json.dumps({"model": MODEL, "tokens": tokens}).encode()
+        s = _connect(port)
+        s.sendall((f"POST /score HTTP/1.1\r\nhost: x\r\n"
+                   f"expect: 100-continue\r\n"
+                   f"content-length: {len(body)}\r\n\r\n").encode())
+        interim = s.recv(4096)
+        assert interim.startswith(b"HTTP/1.1 100")
+        s.sendall(body)
+        # read final response (strip any leftover interim bytes)
+        buf = interim[len(b"HTTP/1.1 100 Continue\r\n\r\n"):]
+        status, rbody, _ = _read_response(s, buf)
+        assert status == 200
+        assert json.loads(rbody)["scores"]["pod-a"] == 8.0
+        s.close()
