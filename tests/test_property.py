@@ -209,3 +209,32 @@ def test_compact_is_lookup_invariant(seed, n_ops):
     after = {k: sorted(map(tuple, v))
              for k, v in nat.lookup(keys, set()).items()}
     assert after == before
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.data())
+def test_chain_cache_transparent(data):
+    """Property: chunk_hashes with the session chain cache enabled is
+    bit-identical to the uncached chain for ANY interleaving of prompts
+    that share prefixes (the cache must be a pure accelerator)."""
+    from llmd_kvcache_amd.kvblock.token_processor import (
+        ChunkedTokenDatabase, TokenProcessorConfig)
+
+    bs = data.draw(st.sampled_from([4, 16]))
+    seg = data.draw(st.sampled_from([2, 4]))
+    cached = ChunkedTokenDatabase(TokenProcessorConfig(
+        block_size=bs, chain_cache_seg_chunks=seg,
+        chain_cache_entries=32))
+    plain = ChunkedTokenDatabase(TokenProcessorConfig(
+        block_size=bs, chain_cache_entries=0))
+    base = data.draw(st.lists(st.integers(0, 2**31 - 1),
+                              min_size=bs * seg, max_size=bs * seg * 6))
+    prompts = []
+    for _ in range(data.draw(st.integers(1, 6))):
+        cut = data.draw(st.integers(0, len(base)))
+        tail = data.draw(st.lists(st.integers(0, 2**31 - 1),
+                                  min_size=0, max_size=bs * seg * 2))
+        prompts.append(base[:cut] + tail)
+    init = cached.config.init_hash()
+    for p in prompts:
+        assert cached.chunk_hashes(init, p) == plain.chunk_hashes(init, p)
