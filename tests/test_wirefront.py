@@ -430,3 +430,34 @@ class TestWireParserDifferential:
         assert status == 200
         assert json.loads(rbody)["scores"]["pod-a"] == 8.0
         s.close()
+
+    def test_two_models_in_one_burst(self, service):
+        """Requests for different models in one pipelined burst group
+        correctly (model is part of the micro-batch group key)."""
+        from llmd_kvcache_amd.kvblock.keys import Key, PodEntry
+        from llmd_kvcache_amd.kvblock.token_processor import (
+            ChunkedTokenDatabase, TokenProcessorConfig)
+
+        svc, port, tokens = service
+        idx = svc.indexer.kv_block_index()
+        tp = ChunkedTokenDatabase(TokenProcessorConfig(block_size=BS))
+        other_tokens = [55000 + i for i in range(8)]
+        keys = tp.tokens_to_kv_block_keys(None, other_tokens, "other-model")
+        idx.add(keys, keys, [PodEntry("pod-other", "gpu")])
+
+        s = _connect(port)
+        blob = (_http_post("/score", {"model": MODEL, "tokens": tokens}) +
+                _http_post("/score", {"model": "other-model",
+                                      "tokens": other_tokens}) +
+                _http_post("/score", {"model": MODEL,
+                                      "tokens": other_tokens}))
+        s.sendall(blob)
+        rest = b""
+        status, b1, rest = _read_response(s, rest)
+        status2, b2, rest = _read_response(s, rest)
+        status3, b3, rest = _read_response(s, rest)
+        assert (status, status2, status3) == (200, 200, 200)
+        assert json.loads(b1)["scores"]["pod-a"] == 8.0
+        assert json.loads(b2)["scores"] == {"pod-other": 2.0}
+        assert json.loads(b3)["scores"] == {}  # other-model keys, MODEL ns
+        s.close()
