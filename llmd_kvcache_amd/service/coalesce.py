@@ -81,15 +81,9 @@ class CoalescingScorer:
             t.join(timeout=2.0)
         # A caller that passed the _running check just before stop() may
         # have enqueued AFTER the sentinel; fail those explicitly so no
-        # caller blocks forever on its event.
-        while True:
-            try:
-                p = self._queue.get_nowait()
-            except queue.Empty:
-                break
-            if p is not None:
-                p.error = RuntimeError("coalescing scorer stopped")
-                p.event.set()
+        # caller blocks forever on its event (score() re-checks after its
+        # put and self-drains for the symmetric window).
+        self._fail_drain()
 
     # -- API -----------------------------------------------------------
     def score(self, tokens: Sequence[int], model: str,
@@ -100,10 +94,26 @@ class CoalescingScorer:
             return self.indexer.score_tokens(tokens, model, pods)
         p = _Pending(tokens, model, pods)
         self._queue.put(p)
+        if not self._running:
+            # stop() may have finished its drain between our check and
+            # the put; no dispatcher will ever serve the queue again, so
+            # fail-drain whatever is left ourselves (possibly including
+            # our own entry) - nobody may block forever
+            self._fail_drain()
         p.event.wait()
         if p.error is not None:
             raise p.error
         return p.result
+
+    def _fail_drain(self) -> None:
+        while True:
+            try:
+                q = self._queue.get_nowait()
+            except queue.Empty:
+                return
+            if q is not None:
+                q.error = RuntimeError("coalescing scorer stopped")
+                q.event.set()
 
     # -- dispatcher ----------------------------------------------------
     def _dispatch_loop(self) -> None:

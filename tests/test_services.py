@@ -466,3 +466,42 @@ class TestCoalescingScorer:
             assert sc.score(tokens, "m", ["nobody"]) == {}
         finally:
             sc.stop()
+
+    def test_stop_during_inflight_requests(self):
+        """stop() must never strand a caller: in-flight and
+        enqueued-after-sentinel requests either complete or fail with a
+        clear error; nobody blocks forever."""
+        import threading
+        import time as _time
+
+        from llmd_kvcache_amd.service.coalesce import CoalescingScorer
+
+        indexer, tokens = self._indexer()
+        sc = CoalescingScorer(indexer)
+        sc.start()
+        outcomes = []
+        lock = threading.Lock()
+
+        def caller():
+            try:
+                r = sc.score(tokens, "m", [])
+                with lock:
+                    outcomes.append(("ok", r))
+            except RuntimeError as e:
+                with lock:
+                    outcomes.append(("err", str(e)))
+
+        threads = [threading.Thread(target=caller) for _ in range(24)]
+        for i, t in enumerate(threads):
+            t.start()
+            if i == 10:
+                threading.Thread(target=sc.stop).start()
+        for t in threads:
+            t.join(timeout=15)
+        assert all(not t.is_alive() for t in threads)  # no stranded caller
+        assert len(outcomes) == 24
+        for kind, payload in outcomes:
+            if kind == "ok":
+                assert payload == {"pod-a": 8.0}
+            else:
+                assert "stopped" in payload
