@@ -745,6 +745,12 @@ class WireFront {
           if (!eol || eol > hend) break;
           if ((eol - h) > 15 && strncasecmp(h, "content-length:", 15) == 0) {
             clen = (size_t)strtoull(h + 15, nullptr, 10);
+            // clamp BEFORE any arithmetic: a hostile value near
+            // SIZE_MAX would wrap hdr_end+4+clen and hand the JSON
+            // cursor an out-of-bounds end pointer
+            if (clen > (64u << 20)) {
+              return false;  // reject oversized/overflowing bodies
+            }
           } else if ((eol - h) > 7 &&
                      strncasecmp(h, "expect:", 7) == 0 &&
                      strcasestr(std::string(h, eol - h).c_str(),

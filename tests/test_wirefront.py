@@ -461,3 +461,19 @@ class TestWireParserDifferential:
         assert json.loads(b2)["scores"] == {"pod-other": 2.0}
         assert json.loads(b3)["scores"] == {}  # other-model keys, MODEL ns
         s.close()
+
+    def test_hostile_content_length_rejected(self, service):
+        """content-length near SIZE_MAX must not wrap the body-offset
+        arithmetic (out-of-bounds parse); the connection is dropped and
+        the server keeps serving."""
+        _, port, tokens = service
+        s = _connect(port)
+        s.sendall(b"POST /score HTTP/1.1\r\nhost: x\r\n"
+                  b"content-length: 18446744073709551615\r\n\r\n{}")
+        got = s.recv(4096)  # server closes (b"" ) or errors; never OOB
+        s.close()
+        s = _connect(port)
+        s.sendall(_http_post("/score", {"model": MODEL, "tokens": tokens}))
+        status, body, _ = _read_response(s)
+        assert status == 200 and json.loads(body)["scores"]["pod-a"] == 8.0
+        s.close()
