@@ -840,6 +840,12 @@ class WireFront {
 
   void flush(int ep, std::map<int, ConnPtr>& conns, Conn* c) {
     collect_ready(*c);
+    if (c->outbuf.size() > (64u << 20)) {
+      // slow-reader guard: a client that pipelines requests without
+      // ever reading responses would grow outbuf unboundedly
+      close_conn(ep, conns, c->fd);
+      return;
+    }
     while (!c->outbuf.empty()) {
       ssize_t w = send(c->fd, c->outbuf.data(), c->outbuf.size(),
                        MSG_NOSIGNAL);
