@@ -480,3 +480,67 @@ def test_partition_prompts_properties():
             max_keys = max(offs[bounds[r + 1]] - offs[bounds[r]]
                            for r in range(world))
             assert max_keys <= total // world + max(counts)
+
+
+def _body_rs_ragged_edges(rank, world_size):
+    """reduce_scatter merge with deliberately nasty shapes: zero-key
+    prompts, heavy skew (one giant prompt), and a prompt count equal to
+    world_size (minimum allowed)."""
+    from llmd_kvcache_amd.kvblock.gpu_index import (
+        NativeIndex,
+        TableIndexConfig,
+        _to_i64,
+    )
+    from llmd_kvcache_amd.kvblock.keys import Key, PodEntry
+    from llmd_kvcache_amd.parallel.sharded import ShardedIndex
+    from llmd_kvcache_amd.scorer import new_kv_block_scorer
+
+    sharded = ShardedIndex(TableIndexConfig(capacity=1 << 10, pods_per_key=4))
+    sharded.force_reduce_scatter = True
+    single = NativeIndex(TableIndexConfig(capacity=1 << 10, pods_per_key=4))
+    scorer = new_kv_block_scorer()
+    keys = [Key(MODEL, 800 + i) for i in range(40)]
+    for idx in (sharded, single):
+        idx.add(keys, keys, [PodEntry("pod-a", "gpu")])
+        idx.add(keys[:5], keys[:5], [PodEntry("pod-b", "cpu")])
+
+    cases = [
+        # [prompt key-counts]: zero-key prompts + skew
+        [0, 40, 0],
+        [1, 1, 1],
+        [0, 0, 38],
+        [12, 0, 12, 0, 12],
+    ]
+    mismatches = []
+    for counts in cases:
+        if len(counts) < world_size:
+            continue
+        prompts, pos = [], 0
+        for c in counts:
+            prompts.append(keys[pos:pos + c] if c else [])
+            pos += c
+        flat = [_to_i64(k.chunk_hash) for p in prompts for k in p]
+        offsets = [0]
+        for p in prompts:
+            offsets.append(offsets[-1] + len(p))
+        scores = sharded.sharded_scores(
+            torch.tensor(flat, dtype=torch.int64),
+            torch.tensor(offsets, dtype=torch.int32), MODEL, set())
+        maps = sharded.local.scores_to_map(scores)
+        for p, got in zip(prompts, maps):
+            expected = ({} if not p else {
+                pod: s for pod, s in
+                scorer.score(p, single.lookup(p, set())).items() if s != 0
+            })
+            if set(got) != set(expected) or any(
+                abs(got[k] - expected[k]) > 1e-4 for k in expected
+            ):
+                mismatches.append((counts, got, expected))
+    return mismatches
+
+
+def test_reduce_scatter_ragged_edges():
+    for ws in (2, 3, 4):
+        results = run_distributed("_body_rs_ragged_edges", world_size=ws)
+        for rank, mm in results.items():
+            assert mm == [], f"ws={ws} rank {rank}: {mm[:2]}"
